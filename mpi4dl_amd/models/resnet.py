@@ -1,0 +1,194 @@
+"""ResNet v1 (6n+2) and v2 (9n+2, pre-activation bottleneck) builders.
+
+Reference parity: src/models/resnet.py (get_resnet_v1 :145,
+get_resnet_v2 :270). Built as a flat ``nn.Sequential`` of coarse cells
+(one residual block per cell) so the pipeline partitioner can split at
+cell granularity. Residual connections live *inside* a cell, so every
+inter-cell activation is a single tensor — the simple pipeline case.
+
+Deliberate deviations from the reference:
+* no softmax before CrossEntropyLoss (reference double-softmaxes,
+  resnet.py:140,265);
+* the classifier head is one cell (pool + flatten + linear).
+
+``device='meta'`` builds the model with meta parameters — used for
+shape inference and for ranks that never materialise remote stages.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+
+def conv_bn_act(
+    in_ch: int,
+    out_ch: int,
+    kernel: int = 3,
+    stride: int = 1,
+    act: bool = True,
+    bn: bool = True,
+):
+    pad = kernel // 2
+    layers = [nn.Conv2d(in_ch, out_ch, kernel, stride=stride, padding=pad, bias=not bn)]
+    if bn:
+        layers.append(nn.BatchNorm2d(out_ch))
+    if act:
+        layers.append(nn.ReLU(inplace=True))
+    return nn.Sequential(*layers)
+
+
+class BasicBlockV1(nn.Module):
+    """v1 cell: conv-bn-relu, conv-bn, (+ projection), relu."""
+
+    def __init__(self, in_ch: int, out_ch: int, stride: int = 1):
+        super().__init__()
+        self.body = nn.Sequential(
+            conv_bn_act(in_ch, out_ch, 3, stride),
+            conv_bn_act(out_ch, out_ch, 3, 1, act=False),
+        )
+        self.proj = None
+        if stride != 1 or in_ch != out_ch:
+            self.proj = conv_bn_act(in_ch, out_ch, 1, stride, act=False, bn=False)
+        self.act = nn.ReLU(inplace=True)
+
+    def forward(self, x):
+        s = x if self.proj is None else self.proj(x)
+        return self.act(self.body(x) + s)
+
+
+class BottleneckV2(nn.Module):
+    """v2 cell: pre-activation bottleneck (BN-ReLU-conv ×3) + skip."""
+
+    expansion = 4
+
+    def __init__(self, in_ch: int, mid_ch: int, stride: int = 1):
+        super().__init__()
+        out_ch = mid_ch * self.expansion
+        self.pre = nn.Sequential(nn.BatchNorm2d(in_ch), nn.ReLU(inplace=True))
+        self.body = nn.Sequential(
+            nn.Conv2d(in_ch, mid_ch, 1, bias=False),
+            nn.BatchNorm2d(mid_ch),
+            nn.ReLU(inplace=True),
+            nn.Conv2d(mid_ch, mid_ch, 3, stride=stride, padding=1, bias=False),
+            nn.BatchNorm2d(mid_ch),
+            nn.ReLU(inplace=True),
+            nn.Conv2d(mid_ch, out_ch, 1, bias=False),
+        )
+        self.proj = None
+        if stride != 1 or in_ch != out_ch:
+            self.proj = nn.Conv2d(in_ch, out_ch, 1, stride=stride, bias=False)
+
+    def forward(self, x):
+        h = self.pre(x)
+        s = x if self.proj is None else self.proj(h)
+        return self.body(h) + s
+
+
+class Head(nn.Module):
+    """Global average pool + linear classifier (one cell)."""
+
+    def __init__(self, in_ch: int, num_classes: int, final_bn: bool = False):
+        super().__init__()
+        self.final = (
+            nn.Sequential(nn.BatchNorm2d(in_ch), nn.ReLU(inplace=True))
+            if final_bn
+            else nn.Identity()
+        )
+        self.pool = nn.AdaptiveAvgPool2d(1)
+        self.fc = nn.Linear(in_ch, num_classes)
+
+    def forward(self, x):
+        x = self.pool(self.final(x))
+        return self.fc(torch.flatten(x, 1))
+
+
+def _stem(in_ch: int, filters: int, image_size: int) -> nn.Module:
+    """Stem sized to the image: big images get a stride-2 7x7 + maxpool so
+    activations stay tractable (the reference keeps a 3x3 stride-1 stem for
+    CIFAR-scale and relies on SP for big images; we keep stride-1 below 128)."""
+    if image_size >= 128:
+        return nn.Sequential(
+            nn.Conv2d(in_ch, filters, 7, stride=2, padding=3, bias=False),
+            nn.BatchNorm2d(filters),
+            nn.ReLU(inplace=True),
+            nn.MaxPool2d(3, stride=2, padding=1),
+        )
+    return conv_bn_act(in_ch, filters, 3, 1)
+
+
+def get_resnet_v1(
+    input_shape,
+    num_classes: int = 10,
+    n: int = 3,
+    num_filters: int = 16,
+    device: Optional[str] = None,
+) -> nn.Sequential:
+    """6n+2-layer v1 ResNet as a flat Sequential of cells
+    (reference resnet.py:145-178)."""
+    _, in_ch, H, W = input_shape
+    cells = [_stem(in_ch, num_filters, min(H, W))]
+    ch = num_filters
+    for group in range(3):
+        out_ch = num_filters * (2**group)
+        for block in range(n):
+            stride = 2 if (group > 0 and block == 0) else 1
+            cells.append(BasicBlockV1(ch, out_ch, stride))
+            ch = out_ch
+    cells.append(Head(ch, num_classes))
+    model = nn.Sequential(*cells)
+    if device is not None:
+        model = model.to(device)
+    return model
+
+
+def get_resnet_v2(
+    input_shape,
+    num_classes: int = 10,
+    n: int = 12,
+    num_filters: int = 16,
+    device: Optional[str] = None,
+) -> nn.Sequential:
+    """9n+2-layer v2 (pre-activation bottleneck) ResNet
+    (reference resnet.py:270-326)."""
+    _, in_ch, H, W = input_shape
+    cells = [_stem(in_ch, num_filters, min(H, W))]
+    ch = num_filters
+    for group in range(3):
+        mid = num_filters * (2**group)
+        for block in range(n):
+            stride = 2 if (group > 0 and block == 0) else 1
+            cells.append(BottleneckV2(ch, mid, stride))
+            ch = mid * BottleneckV2.expansion
+    cells.append(Head(ch, num_classes, final_bn=True))
+    model = nn.Sequential(*cells)
+    if device is not None:
+        model = model.to(device)
+    return model
+
+
+def get_resnet101_cells(
+    input_shape,
+    num_classes: int = 1000,
+    width: int = 64,
+    device: Optional[str] = None,
+) -> nn.Sequential:
+    """ImageNet-style ResNet-101 as flat cells (for BASELINE config 4:
+    ResNet-101 SP+PP at 2048²). Bottleneck counts 3-4-23-3."""
+    _, in_ch, H, W = input_shape
+    cells = [_stem(in_ch, width, min(H, W))]
+    ch = width
+    for group, blocks in enumerate([3, 4, 23, 3]):
+        mid = width * (2**group)
+        for b in range(blocks):
+            stride = 2 if (group > 0 and b == 0) else 1
+            cells.append(BottleneckV2(ch, mid, stride))
+            ch = mid * BottleneckV2.expansion
+    cells.append(Head(ch, num_classes, final_bn=True))
+    model = nn.Sequential(*cells)
+    if device is not None:
+        model = model.to(device)
+    return model

@@ -1,0 +1,248 @@
+"""LP (layer parallelism) + PP (pipeline) training engine.
+
+Reference parity: torchgems ``train_model`` (src/torchgems/
+mp_pipeline.py:171-538): per-rank forward/backward over ``parts``
+micro-batches with point-to-point activation/grad transfer between
+stages, GPipe-style all-forwards-then-all-backwards schedule, tuple
+(multi-tensor) activations, GEMS_INVERSE mirrored peer maps
+(mp_pipeline.py:238-248).
+
+MI355X-native design:
+* P2P via grouped RCCL send/recv over xGMI (p2p.py), stream-ordered —
+  no host-side synchronize fences;
+* persistent preallocated recv buffers per micro-batch
+  (mp_pipeline.py:250-292 keeps the same structure);
+* optional bf16 autocast compute with a fixed boundary ``act_dtype`` so
+  activation messages are half the bytes of the reference's fp32;
+* loss scaled by 1/parts at backward so gradients are the mean over the
+  full batch (the reference instead rescales grads in the allreduce,
+  comm.py:440-458).
+"""
+
+from __future__ import annotations
+
+import logging
+from typing import List, Optional
+
+import torch
+import torch.nn as nn
+
+from .. import p2p
+from ..comm import Communicator, current_device
+
+log = logging.getLogger(__name__)
+
+
+def _as_list(spec):
+    """Shape spec -> list of shape tuples (single tensor = 1-list)."""
+    if isinstance(spec, list):
+        return [tuple(s) for s in spec]
+    return [tuple(spec)]
+
+
+class train_model:
+    """One pipeline-stage engine living on one rank/GPU.
+
+    Parameters mirror the reference where meaningful; ``comm`` supplies
+    all topology. ``local_rank`` is the engine's in-clique position
+    (NOT the global rank); for a GEMS-inverse engine it is the mirrored
+    position and peers are mapped through ``comm.engine_peer``.
+    """
+
+    def __init__(
+        self,
+        model_gen,
+        local_rank: int,
+        batch_size: int,
+        parts: int,
+        comm: Communicator,
+        criterion: Optional[nn.Module] = None,
+        optimizer: Optional[torch.optim.Optimizer] = None,
+        lr: float = 0.001,
+        GEMS_INVERSE: bool = False,
+        act_dtype: Optional[torch.dtype] = None,
+        autocast_dtype: Optional[torch.dtype] = None,
+        device: Optional[torch.device] = None,
+    ):
+        self.model_gen = model_gen
+        self.comm = comm
+        self.local_rank = local_rank
+        self.split_size = model_gen.split_size
+        self.split_rank = comm.get_split_rank(local_rank)
+        self.batch_size = batch_size
+        self.parts = parts
+        assert batch_size % parts == 0, "batch_size must divide into parts"
+        self.mb = batch_size // parts  # micro-batch size
+        self.GEMS_INVERSE = GEMS_INVERSE
+        self.device = device or current_device()
+        self.autocast_dtype = autocast_dtype
+        self.act_dtype = act_dtype or torch.float32
+
+        self.models = model_gen.models  # local stage (after ready_model)
+        assert self.models is not None, "call model_gen.ready_model first"
+        self.criterion = criterion or nn.CrossEntropyLoss()
+        self.optimizer = optimizer or torch.optim.SGD(
+            self.models.parameters(), lr=lr, momentum=0.9
+        )
+
+        self.shape_list = model_gen.shape_list
+        assert self.shape_list, "model_gen must have run get_output_shapes"
+
+        self._init_peers()
+        self._init_buffers()
+
+        # per-step state
+        self.inputs: List = [None] * parts   # leaf inputs per part
+        self.outputs: List = [None] * parts  # stage outputs per part
+        self.loss_sum = 0.0
+        self.correct_sum = 0
+        self.seen = 0
+        self._pending: List[p2p.Transfer] = []
+
+    # ------------------------------------------------------------------
+    # topology
+    # ------------------------------------------------------------------
+
+    def _init_peers(self):
+        """Prev/next stage peers (reference mp_pipeline.py:238-248)."""
+        self.first_stage = self.split_rank == 0
+        self.last_stage = self.split_rank == self.split_size - 1
+        self.prev_rank = None
+        self.next_rank = None
+        if not self.first_stage:
+            self.prev_rank = self.comm.engine_peer(self.local_rank - 1, self.GEMS_INVERSE)
+        if not self.last_stage:
+            self.next_rank = self.comm.engine_peer(self.local_rank + 1, self.GEMS_INVERSE)
+
+    def _recv_shapes(self) -> List[tuple]:
+        """Shapes of the activations this stage receives (= prev stage's
+        outputs), with batch dim set to the micro-batch size."""
+        spec = _as_list(self.shape_list[self.split_rank - 1])
+        return [(self.mb,) + tuple(s[1:]) for s in spec]
+
+    def _init_buffers(self):
+        """Preallocated per-part recv buffers (mp_pipeline.py:250-292)."""
+        self.input_buffers = None
+        self.grad_buffers = None
+        if not self.first_stage:
+            self.input_buffers = [
+                [
+                    torch.zeros(s, device=self.device, dtype=self.act_dtype)
+                    for s in self._recv_shapes()
+                ]
+                for _ in range(self.parts)
+            ]
+        if not self.last_stage:
+            out_spec = _as_list(self.shape_list[self.split_rank])
+            self.grad_buffers = [
+                [
+                    torch.zeros((self.mb,) + tuple(s[1:]), device=self.device, dtype=self.act_dtype)
+                    for s in out_spec
+                ]
+                for _ in range(self.parts)
+            ]
+
+    # ------------------------------------------------------------------
+    # forward / backward per micro-batch
+    # ------------------------------------------------------------------
+
+    def _run_stage(self, x):
+        if self.autocast_dtype is not None and self.device.type == "cuda":
+            with torch.autocast("cuda", dtype=self.autocast_dtype):
+                return self.models(x)
+        return self.models(x)
+
+    def receive_input(self, part: int):
+        bufs = self.input_buffers[part]
+        p2p.recv_tensors(bufs, self.prev_rank, tag_base=1000 + part * 16)
+        leaves = []
+        for b in bufs:
+            t = b.clone().requires_grad_(True)
+            leaves.append(t)
+        return leaves[0] if len(leaves) == 1 else tuple(leaves)
+
+    def send_output(self, y, part: int):
+        ts = [t.to(self.act_dtype) for t in (y if isinstance(y, tuple) else (y,))]
+        tr = p2p.isend_tensors(ts, self.next_rank, tag_base=1000 + part * 16)
+        self._pending.append(tr)
+
+    def forward_pass(self, data_x, data_y, part: int):
+        """One micro-batch forward on this stage (mp_pipeline.py:434-473)."""
+        if self.first_stage:
+            x = data_x.to(self.device, non_blocking=True)
+        else:
+            x = self.receive_input(part)
+        self.inputs[part] = x
+        y = self._run_stage(x)
+        self.outputs[part] = y
+        if self.last_stage:
+            yl = data_y.to(self.device, non_blocking=True)
+            logits = y[0] if isinstance(y, tuple) else y
+            loss = self.criterion(logits.float(), yl)
+            self.loss_sum += float(loss.detach())
+            with torch.no_grad():
+                self.correct_sum += int((logits.argmax(dim=1) == yl).sum())
+                self.seen += yl.numel()
+            # keep scaled loss for backward
+            self.outputs[part] = loss * (1.0 / self.parts)
+        else:
+            self.send_output(y, part)
+        return self.outputs[part]
+
+    def backward_pass(self, part: int):
+        """One micro-batch backward (mp_pipeline.py:475-507)."""
+        y = self.outputs[part]
+        if self.last_stage:
+            y.backward()
+        else:
+            gbufs = self.grad_buffers[part]
+            p2p.recv_tensors(gbufs, self.next_rank, tag_base=3000 + part * 16)
+            ys = list(y) if isinstance(y, tuple) else [y]
+            grads = [g.to(t.dtype) for g, t in zip(gbufs, ys)]
+            torch.autograd.backward(ys, grads)
+        if not self.first_stage:
+            x = self.inputs[part]
+            xs = list(x) if isinstance(x, tuple) else [x]
+            gsend = [
+                (t.grad if t.grad is not None else torch.zeros_like(t)).to(self.act_dtype)
+                for t in xs
+            ]
+            tr = p2p.isend_tensors(gsend, self.prev_rank, tag_base=3000 + part * 16)
+            self._pending.append(tr)
+        # free graph state for this part
+        self.outputs[part] = None
+        self.inputs[part] = None
+
+    # ------------------------------------------------------------------
+    # step / update
+    # ------------------------------------------------------------------
+
+    def run_step(self, inputs: Optional[torch.Tensor], labels: Optional[torch.Tensor]):
+        """One training step over the whole batch (mp_pipeline.py:509-534):
+        all forwards, then all backwards (GPipe fill-drain)."""
+        self.loss_sum = 0.0
+        self.correct_sum = 0
+        self.seen = 0
+        parts_x = [None] * self.parts
+        parts_y = [None] * self.parts
+        if inputs is not None and self.first_stage:
+            parts_x = list(inputs.chunk(self.parts, dim=0))
+        if labels is not None and self.last_stage:
+            parts_y = list(labels.chunk(self.parts, dim=0))
+        for part in range(self.parts):
+            self.forward_pass(parts_x[part], parts_y[part], part)
+        for part in range(self.parts):
+            self.backward_pass(part)
+        self._drain()
+        return self.loss_sum / max(self.parts, 1), self.correct_sum, self.seen
+
+    def _drain(self):
+        for tr in self._pending:
+            tr.wait()
+        self._pending = []
+
+    def update(self):
+        """Optimizer step + grad reset (mp_pipeline.py:536). Grads may be
+        views into a FlatGrads buffer — never set_to_none."""
+        self.optimizer.step()
+        self.optimizer.zero_grad(set_to_none=False)
