@@ -1,0 +1,207 @@
+"""Distributed spatial nn.Modules: halo-aware conv / pool / exchange layer.
+
+Reference parity: ``conv_spatial`` (src/torchgems/spatial.py:25-1029),
+``halo_exchange_layer`` (:1032-1413), ``Pool`` (:1416-1509).
+
+Design notes (MI355X-first):
+* A module owns a HaloExchanger; communication is one grouped RCCL call
+  per exchange (see ops/halo.py), stream-ordered — none of the
+  reference's tag bookkeeping or host fences.
+* Compute runs on halo-padded tiles with padding=0. On gfx950 the conv
+  itself is either MIOpen (via F.conv2d) or the gemscore implicit-GEMM
+  MFMA kernel (ops/conv_native.py), selected per-layer.
+* ``grad_mode='exact'`` (default) gives bit-parity with single-GPU conv
+  by doing a transposed halo exchange in backward; 'drop' reproduces the
+  reference's halo-gradient-dropping semantics.
+* Meta tensors: shape-only path, no comm (partitioner shape inference).
+* D2 support: ``halo_len=0`` convs with asymmetric outer-edge padding by
+  tile position (reference spatial.py:67-111), and a standalone
+  HaloExchangeLayer with a large halo every ``fused_layers`` blocks
+  (reference resnet_spatial_d2.py).
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from .halo import HaloExchanger, TileLayout, halo_pad
+
+
+def outer_pad_only(x: torch.Tensor, layout: TileLayout, tile: int, pad: int):
+    """Asymmetric zero-pad: pad only the sides of the tile that lie on the
+    image boundary (D2 design — reference spatial.py:67-111). Interior
+    sides are covered by earlier large-halo exchanges."""
+    if pad == 0:
+        return x
+    r, c = layout.pos(tile)
+    left = pad if c == 0 else 0
+    right = pad if c == layout.cols - 1 else 0
+    top = pad if r == 0 else 0
+    bottom = pad if r == layout.rows - 1 else 0
+    if left or right or top or bottom:
+        return F.pad(x, (left, right, top, bottom))
+    return x
+
+
+class _SpatialBase(nn.Module):
+    """Shared plumbing: layout, exchanger, rank map."""
+
+    def __init__(
+        self,
+        num_spatial_parts: int,
+        slice_method: str,
+        spatial_local_rank: int,
+        rank_of_tile=None,
+        grad_mode: str = "exact",
+    ):
+        super().__init__()
+        self.layout = TileLayout(num_spatial_parts, slice_method)
+        self.tile = spatial_local_rank
+        self.grad_mode = grad_mode
+        if rank_of_tile is None:
+            rank_of_tile = lambda t: t  # tile index == global rank (tests)
+        self.exchanger = (
+            HaloExchanger(self.layout, self.tile, rank_of_tile)
+            if num_spatial_parts > 1
+            else None
+        )
+
+
+class HaloConv2d(_SpatialBase):
+    """Spatially-distributed Conv2d (reference conv_spatial, spatial.py:25).
+
+    Forward: halo-pad the local tile by ``halo_len`` (= the padding the
+    undistributed conv would use), exchange the ring with up to 8
+    neighbours, then convolve with padding=0. With stride s the local
+    tile size must be divisible by s (power-of-two image sizes — the
+    reference's verify_spatial_config enforces the same).
+
+    ``halo_len=None`` derives (kernel_size-1)//2. ``halo_len=0`` with
+    ``d2=True`` applies asymmetric outer-edge padding instead (D2).
+    """
+
+    def __init__(
+        self,
+        in_channels: int,
+        out_channels: int,
+        kernel_size: int,
+        stride: int = 1,
+        padding: Optional[int] = None,
+        bias: bool = True,
+        num_spatial_parts: int = 1,
+        slice_method: str = "square",
+        spatial_local_rank: int = 0,
+        rank_of_tile=None,
+        grad_mode: str = "exact",
+        d2: bool = False,
+        halo_len: Optional[int] = None,
+    ):
+        super().__init__(
+            num_spatial_parts, slice_method, spatial_local_rank, rank_of_tile, grad_mode
+        )
+        if padding is None:
+            padding = (kernel_size - 1) // 2
+        self.halo_len = padding if halo_len is None else halo_len
+        self.d2 = d2
+        self.outer_pad = padding if d2 else 0
+        self.conv = nn.Conv2d(
+            in_channels, out_channels, kernel_size, stride=stride, padding=0, bias=bias
+        )
+        self.stride = stride
+        self.kernel_size = kernel_size
+
+    def forward(self, x):
+        if self.d2 and self.halo_len == 0:
+            if self.exchanger is not None:
+                xp = outer_pad_only(x, self.layout, self.tile, self.outer_pad)
+            else:
+                xp = F.pad(x, (self.outer_pad,) * 4)
+        else:
+            xp = halo_pad(x, self.halo_len, self.exchanger, self.grad_mode)
+        return self.conv(xp)
+
+
+class HaloExchangeLayer(_SpatialBase):
+    """Standalone pad+exchange module (reference halo_exchange_layer,
+    spatial.py:1032). Output spatial dims grow by 2*halo_len; used by the
+    D2 design to amortise one large exchange over ``fused_layers`` convs."""
+
+    def __init__(
+        self,
+        halo_len: int,
+        num_spatial_parts: int = 1,
+        slice_method: str = "square",
+        spatial_local_rank: int = 0,
+        rank_of_tile=None,
+        grad_mode: str = "exact",
+    ):
+        super().__init__(
+            num_spatial_parts, slice_method, spatial_local_rank, rank_of_tile, grad_mode
+        )
+        self.halo_len = halo_len
+
+    def forward(self, x):
+        return halo_pad(x, self.halo_len, self.exchanger, self.grad_mode)
+
+
+class HaloPool2d(_SpatialBase):
+    """Spatial max/avg pool (reference Pool, spatial.py:1416): halo-pad by
+    (kernel-1)//2 then pool with padding=0.
+
+    NOTE on avg-pool semantics: a single-GPU AvgPool2d with padding>0 and
+    count_include_pad=False is NOT reproducible tile-locally at interior
+    edges (the halo values are real, not pad). We therefore only claim
+    exact parity for max pool and for avg pool with padding=0; the
+    reference has the same caveat (spatial.py:1440-1441 asserts
+    count_include_pad incompatibility).
+    """
+
+    def __init__(
+        self,
+        kind: str,
+        kernel_size: int,
+        stride: Optional[int] = None,
+        padding: int = 0,
+        num_spatial_parts: int = 1,
+        slice_method: str = "square",
+        spatial_local_rank: int = 0,
+        rank_of_tile=None,
+        grad_mode: str = "exact",
+        d2: bool = False,
+    ):
+        super().__init__(
+            num_spatial_parts, slice_method, spatial_local_rank, rank_of_tile, grad_mode
+        )
+        assert kind in ("max", "avg")
+        self.kind = kind
+        self.kernel_size = kernel_size
+        self.stride = stride or kernel_size
+        self.halo_len = padding
+        self.d2 = d2
+
+    def forward(self, x):
+        h = self.halo_len
+        # max pool pads with -inf so image-boundary windows match the
+        # single-GPU op exactly (zero-pad would win over negative inputs)
+        fill = float("-inf") if self.kind == "max" else 0.0
+        if self.d2 and h > 0:
+            xp = (
+                outer_pad_only(x, self.layout, self.tile, h)
+                if self.exchanger is not None
+                else F.pad(x, (h,) * 4, value=fill)
+            )
+        else:
+            xp = (
+                halo_pad(x, h, self.exchanger, self.grad_mode, fill=fill)
+                if h > 0
+                else x
+            )
+        if self.kind == "max":
+            return F.max_pool2d(xp, self.kernel_size, self.stride, padding=0)
+        return F.avg_pool2d(
+            xp, self.kernel_size, self.stride, padding=0, count_include_pad=True
+        )
