@@ -384,7 +384,10 @@ class FlatGrads:
         offset = 0
         for p in self.params:
             n = p.numel()
-            p.grad = self.buffer[offset : offset + n].view_as(p)
+            view = self.buffer[offset : offset + n].view_as(p)
+            if p.grad is not None:
+                view.copy_(p.grad)  # keep grads accumulated before attach
+            p.grad = view
             offset += n
 
     def zero_(self):
@@ -463,10 +466,19 @@ class GradReducer:
 
     def apply_allreduce(self, module: torch.nn.Module):
         """Standard path: spatial group first (tile replicas), then outer DP
-        (reference apply_allreduce, comm.py:506-514)."""
+        (reference apply_allreduce, comm.py:506-514).
+
+        Scaling: tiles PARTITION the pixels of one batch, so the spatial
+        group SUMS (each tile holds a partial weight grad). Local-DP and
+        outer-DP groups each see the full weight grad of a batch shard,
+        so those AVERAGE. This reproduces the serial gradient exactly
+        (the reference folds the same arithmetic into divide_bs,
+        comm.py:440-458)."""
         if self.comm.ENABLE_SPATIAL and self.comm.split_rank < self.comm.spatial_size:
             self.allreduce_grads(
-                module, self.comm.spatial_allreduce_groups.get(self.comm.split_rank)
+                module,
+                self.comm.spatial_allreduce_groups.get(self.comm.split_rank),
+                divide_by=1.0,
             )
         if self.comm.LOCAL_DP_LP > 1 and self.comm.split_rank >= self.comm.spatial_size:
             self.allreduce_grads(

@@ -91,10 +91,10 @@ class BottleneckV2(nn.Module):
 class Head(nn.Module):
     """Global average pool + linear classifier (one cell)."""
 
-    def __init__(self, in_ch: int, num_classes: int, final_bn: bool = False):
+    def __init__(self, in_ch: int, num_classes: int, final_bn: bool = False, mknorm=nn.BatchNorm2d):
         super().__init__()
         self.final = (
-            nn.Sequential(nn.BatchNorm2d(in_ch), nn.ReLU(inplace=True))
+            nn.Sequential(mknorm(in_ch), nn.ReLU(inplace=True))
             if final_bn
             else nn.Identity()
         )

@@ -1,0 +1,90 @@
+"""TileBatchNorm2d — BatchNorm with statistics reduced across the tiles
+of one spatial partition.
+
+The reference uses plain ``nn.BatchNorm2d`` inside spatial partitions,
+so each tile normalises with ITS OWN statistics — distributed training
+silently computes a different function than the undistributed model
+(and than the paper's math). This module all-reduces the per-channel
+sums over the spatial tile group, making spatially-parallel BN exactly
+equal to single-GPU BN (our SP parity tests rely on it).
+
+Works on gloo (CPU tests) and RCCL; autograd-correct via a sum-allreduce
+whose adjoint is another sum-allreduce. Falls back to local statistics
+when no group is set (plain model) and on meta tensors (shape
+inference). torch.nn.SyncBatchNorm is CUDA/NCCL-only and syncs over a
+whole group of its own; this version is backend-agnostic and scoped to
+the tile group.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+class _AllReduceSum(torch.autograd.Function):
+    """Differentiable sum-allreduce: forward sums over the group, backward
+    sums the incoming grads over the group (the adjoint of broadcast-sum)."""
+
+    @staticmethod
+    def forward(ctx, t: torch.Tensor, group):
+        ctx.group = group
+        out = t.clone()
+        dist.all_reduce(out, group=group)
+        return out
+
+    @staticmethod
+    def backward(ctx, g: torch.Tensor):
+        g = g.contiguous().clone()
+        dist.all_reduce(g, group=ctx.group)
+        return g, None
+
+
+def allreduce_sum(t: torch.Tensor, group) -> torch.Tensor:
+    return _AllReduceSum.apply(t, group)
+
+
+class TileBatchNorm2d(nn.BatchNorm2d):
+    """Drop-in BatchNorm2d whose batch statistics span the tile group."""
+
+    def __init__(self, num_features: int, group=None, **kw):
+        super().__init__(num_features, **kw)
+        self.group = group
+
+    def _use_sync(self, x) -> bool:
+        return (
+            self.training
+            and self.group is not None
+            and not x.is_meta
+            and dist.is_initialized()
+        )
+
+    def forward(self, x):
+        if not self._use_sync(x):
+            return super().forward(x)
+        n_local = x.numel() // x.shape[1]
+        # pack [sum, sumsq] into one message: one allreduce per layer
+        s = x.sum(dim=(0, 2, 3))
+        ss = (x * x).sum(dim=(0, 2, 3))
+        packed = torch.cat([s, ss])
+        packed = allreduce_sum(packed, self.group)
+        world = dist.get_world_size(group=self.group)
+        n = n_local * world  # tiles are equal-sized (power-of-two constraint)
+        mean = packed[: self.num_features] / n
+        var = packed[self.num_features :] / n - mean * mean
+        if self.track_running_stats:
+            with torch.no_grad():
+                m = self.momentum if self.momentum is not None else 0.1
+                self.running_mean.mul_(1 - m).add_(mean.detach(), alpha=m)
+                unbiased = var.detach() * (n / max(n - 1, 1))
+                self.running_var.mul_(1 - m).add_(unbiased, alpha=m)
+                self.num_batches_tracked += 1
+        inv = torch.rsqrt(var + self.eps)
+        out = (x - mean.view(1, -1, 1, 1)) * inv.view(1, -1, 1, 1)
+        if self.affine:
+            out = out * self.weight.view(1, -1, 1, 1) + self.bias.view(1, -1, 1, 1)
+        return out

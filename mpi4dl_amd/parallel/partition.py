@@ -144,13 +144,31 @@ def infer_boundary_shapes(
 
 
 def _to_meta(model: nn.Module) -> nn.Module:
-    """A structural copy of ``model`` with meta-device parameters/buffers."""
+    """A structural copy of ``model`` with meta-device parameters/buffers.
+
+    ProcessGroup handles (TileBatchNorm2d.group), HaloExchangers and
+    other comm objects are passed through by identity — they are neither
+    copyable nor needed for a meta forward (spatial modules skip comm on
+    meta tensors).
+    """
     import copy
 
-    meta = copy.deepcopy(model.__class__.__new__(model.__class__))
-    # Cheaper + robust: re-use module objects via to_empty on a deepcopy is
-    # expensive for big models; instead swap tensors to meta on a shallow
-    # structural clone.
-    meta = copy.deepcopy(model)  # weights copied once on CPU; acceptable
+    memo = {}
+    try:
+        from torch.distributed import ProcessGroup
+
+        for m in model.modules():
+            for v in m.__dict__.values():
+                if isinstance(v, ProcessGroup):
+                    memo[id(v)] = v
+    except ImportError:
+        pass
+    from ..ops.halo import HaloExchanger
+
+    for m in model.modules():
+        for v in m.__dict__.values():
+            if isinstance(v, HaloExchanger):
+                memo[id(v)] = v
+    meta = copy.deepcopy(model, memo)
     meta = meta.to(device="meta")
     return meta

@@ -189,26 +189,35 @@ class train_model:
             self.send_output(y, part)
         return self.outputs[part]
 
+    def receive_output_grad(self, part: int):
+        """Grad wrt this stage's outputs, from the next stage."""
+        gbufs = self.grad_buffers[part]
+        p2p.recv_tensors(gbufs, self.next_rank, tag_base=3000 + part * 16)
+        return gbufs
+
+    def send_input_grad(self, part: int):
+        """Grad wrt this stage's received inputs, to the previous stage."""
+        x = self.inputs[part]
+        xs = list(x) if isinstance(x, tuple) else [x]
+        gsend = [
+            (t.grad if t.grad is not None else torch.zeros_like(t)).to(self.act_dtype)
+            for t in xs
+        ]
+        tr = p2p.isend_tensors(gsend, self.prev_rank, tag_base=3000 + part * 16)
+        self._pending.append(tr)
+
     def backward_pass(self, part: int):
         """One micro-batch backward (mp_pipeline.py:475-507)."""
         y = self.outputs[part]
         if self.last_stage:
             y.backward()
         else:
-            gbufs = self.grad_buffers[part]
-            p2p.recv_tensors(gbufs, self.next_rank, tag_base=3000 + part * 16)
+            gbufs = self.receive_output_grad(part)
             ys = list(y) if isinstance(y, tuple) else [y]
             grads = [g.to(t.dtype) for g, t in zip(gbufs, ys)]
             torch.autograd.backward(ys, grads)
         if not self.first_stage:
-            x = self.inputs[part]
-            xs = list(x) if isinstance(x, tuple) else [x]
-            gsend = [
-                (t.grad if t.grad is not None else torch.zeros_like(t)).to(self.act_dtype)
-                for t in xs
-            ]
-            tr = p2p.isend_tensors(gsend, self.prev_rank, tag_base=3000 + part * 16)
-            self._pending.append(tr)
+            self.send_input_grad(part)
         # free graph state for this part
         self.outputs[part] = None
         self.inputs[part] = None
