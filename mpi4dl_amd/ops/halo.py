@@ -94,18 +94,28 @@ class TileLayout:
         return x[..., r * th : (r + 1) * th, c * tw : (c + 1) * tw]
 
 
+def _hpair(h):
+    """halo spec -> (hh, hw): rows halo, cols halo (asymmetric kernels
+    like AmoebaNet's 1x7/7x1 exchange in one axis only)."""
+    if isinstance(h, (tuple, list)):
+        return int(h[0]), int(h[1])
+    return int(h), int(h)
+
+
 def send_region(d, H, W, h):
     """Interior boundary band (in padded coords) sent toward direction d."""
+    hh, hw = _hpair(h)
     dr, dc = d
-    rs = {-1: (h, 2 * h), 0: (h, h + H), 1: (H, h + H)}[dr]
-    cs = {-1: (h, 2 * h), 0: (h, h + W), 1: (W, h + W)}[dc]
+    rs = {-1: (hh, 2 * hh), 0: (hh, hh + H), 1: (H, hh + H)}[dr]
+    cs = {-1: (hw, 2 * hw), 0: (hw, hw + W), 1: (W, hw + W)}[dc]
     return rs, cs
 
 
 def recv_region(d, H, W, h):
+    hh, hw = _hpair(h)
     dr, dc = d
-    rs = {-1: (0, h), 0: (h, h + H), 1: (h + H, 2 * h + H)}[dr]
-    cs = {-1: (0, h), 0: (h, h + W), 1: (h + W, 2 * h + W)}[dc]
+    rs = {-1: (0, hh), 0: (hh, hh + H), 1: (hh + H, 2 * hh + H)}[dr]
+    cs = {-1: (0, hw), 0: (hw, hw + W), 1: (hw + W, 2 * hw + W)}[dc]
     return rs, cs
 
 
@@ -146,11 +156,14 @@ class HaloExchanger:
         ring is filled (async overlap is handled by HaloConv2d's
         interior/boundary split, not here).
         """
-        if h == 0 or not self.neigh:
+        hh, hw = _hpair(h)
+        if (hh == 0 and hw == 0) or not self.neigh:
             return
-        H, W = xp.shape[-2] - 2 * h, xp.shape[-1] - 2 * h
-        sends, recvs, stage = [], [], []
+        H, W = xp.shape[-2] - 2 * hh, xp.shape[-1] - 2 * hw
+        sends, recvs = [], []
         for d, t in self.neigh:
+            if (d[0] != 0 and hh == 0) or (d[1] != 0 and hw == 0):
+                continue  # no halo along that axis
             peer = self.rank_of_tile(t)
             (rs, re), (cs, ce) = send_region(d, H, W, h)
             buf = xp[:, :, rs:re, cs:ce].contiguous()
@@ -180,12 +193,15 @@ class HaloExchanger:
         bands belong to neighbours' interior pixels: send each band to its
         owner; add received bands into my interior edge regions.
         """
-        H, W = gp.shape[-2] - 2 * h, gp.shape[-1] - 2 * h
-        g = gp[:, :, h : h + H, h : h + W].clone()
-        if h == 0 or not self.neigh:
+        hh, hw = _hpair(h)
+        H, W = gp.shape[-2] - 2 * hh, gp.shape[-1] - 2 * hw
+        g = gp[:, :, hh : hh + H, hw : hw + W].clone()
+        if (hh == 0 and hw == 0) or not self.neigh:
             return g
         sends, recvs = [], []
         for d, t in self.neigh:
+            if (d[0] != 0 and hh == 0) or (d[1] != 0 and hw == 0):
+                continue
             peer = self.rank_of_tile(t)
             # the band I received FROM d in forward carries grads for the
             # neighbour's interior: send it back tagged with my direction
@@ -208,7 +224,7 @@ class HaloExchanger:
         tr.wait()
         for rbuf, _, _, (rs, re, cs, ce) in recvs:
             # send_region coords are in padded space; shift to unpadded
-            g[:, :, rs - h : re - h, cs - h : ce - h].add_(rbuf)
+            g[:, :, rs - hh : re - hh, cs - hw : ce - hw].add_(rbuf)
         return g
 
 
@@ -220,20 +236,22 @@ class _HaloPadFn(torch.autograd.Function):
         ctx.h = h
         ctx.exchanger = exchanger
         ctx.grad_mode = grad_mode
-        xp = F.pad(x, (h, h, h, h), value=fill)
+        hh, hw = _hpair(h)
+        xp = F.pad(x, (hw, hw, hh, hh), value=fill)
         if not x.is_meta:
             exchanger.exchange_padded(xp, h)
         return xp
 
     @staticmethod
     def backward(ctx, gp):
-        h = ctx.h
-        if h == 0:
-            return gp, None, None, None
+        hh, hw = _hpair(ctx.h)
+        if hh == 0 and hw == 0:
+            return gp, None, None, None, None
         if ctx.grad_mode == "exact" and not gp.is_meta:
-            g = ctx.exchanger.exchange_grad_padded(gp.contiguous(), h)
+            g = ctx.exchanger.exchange_grad_padded(gp.contiguous(), ctx.h)
         else:
-            g = gp[:, :, h:-h, h:-h]
+            H, W = gp.shape[-2] - 2 * hh, gp.shape[-1] - 2 * hw
+            g = gp[:, :, hh : hh + H, hw : hw + W]
         return g, None, None, None, None
 
 
@@ -241,6 +259,7 @@ def halo_pad(x, h, exchanger: HaloExchanger, grad_mode: str = "exact", fill: flo
     """Pad by h (fill value for outer/image-boundary ring, e.g. -inf for
     max pool to match single-GPU semantics) and fill interior sides from
     neighbours (autograd-aware)."""
+    hh, hw = _hpair(h)
     if exchanger is None or not exchanger.neigh:
-        return F.pad(x, (h, h, h, h), value=fill)
+        return F.pad(x, (hw, hw, hh, hh), value=fill)
     return _HaloPadFn.apply(x, h, exchanger, grad_mode, fill)

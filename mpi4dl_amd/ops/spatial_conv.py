@@ -31,19 +31,20 @@ import torch.nn.functional as F
 from .halo import HaloExchanger, TileLayout, halo_pad
 
 
-def outer_pad_only(x: torch.Tensor, layout: TileLayout, tile: int, pad: int):
+def outer_pad_only(x: torch.Tensor, layout: TileLayout, tile: int, pad, fill: float = 0.0):
     """Asymmetric zero-pad: pad only the sides of the tile that lie on the
     image boundary (D2 design — reference spatial.py:67-111). Interior
     sides are covered by earlier large-halo exchanges."""
-    if pad == 0:
+    ph, pw = (pad, pad) if isinstance(pad, int) else pad
+    if ph == 0 and pw == 0:
         return x
     r, c = layout.pos(tile)
-    left = pad if c == 0 else 0
-    right = pad if c == layout.cols - 1 else 0
-    top = pad if r == 0 else 0
-    bottom = pad if r == layout.rows - 1 else 0
+    left = pw if c == 0 else 0
+    right = pw if c == layout.cols - 1 else 0
+    top = ph if r == 0 else 0
+    bottom = ph if r == layout.rows - 1 else 0
     if left or right or top or bottom:
-        return F.pad(x, (left, right, top, bottom))
+        return F.pad(x, (left, right, top, bottom), value=fill)
     return x
 
 
@@ -88,9 +89,9 @@ class HaloConv2d(_SpatialBase):
         self,
         in_channels: int,
         out_channels: int,
-        kernel_size: int,
-        stride: int = 1,
-        padding: Optional[int] = None,
+        kernel_size,
+        stride=1,
+        padding=None,
         bias: bool = True,
         num_spatial_parts: int = 1,
         slice_method: str = "square",
@@ -98,16 +99,19 @@ class HaloConv2d(_SpatialBase):
         rank_of_tile=None,
         grad_mode: str = "exact",
         d2: bool = False,
-        halo_len: Optional[int] = None,
+        halo_len=None,
     ):
         super().__init__(
             num_spatial_parts, slice_method, spatial_local_rank, rank_of_tile, grad_mode
         )
+        kh, kw = (kernel_size, kernel_size) if isinstance(kernel_size, int) else kernel_size
         if padding is None:
-            padding = (kernel_size - 1) // 2
-        self.halo_len = padding if halo_len is None else halo_len
+            padding = ((kh - 1) // 2, (kw - 1) // 2)
+        elif isinstance(padding, int):
+            padding = (padding, padding)
+        self.halo_len = tuple(padding) if halo_len is None else halo_len
         self.d2 = d2
-        self.outer_pad = padding if d2 else 0
+        self.outer_pad = tuple(padding) if d2 else (0, 0)
         self.conv = nn.Conv2d(
             in_channels, out_channels, kernel_size, stride=stride, padding=0, bias=bias
         )
@@ -115,11 +119,12 @@ class HaloConv2d(_SpatialBase):
         self.kernel_size = kernel_size
 
     def forward(self, x):
-        if self.d2 and self.halo_len == 0:
+        if self.d2 and self.halo_len in (0, (0, 0)):
             if self.exchanger is not None:
                 xp = outer_pad_only(x, self.layout, self.tile, self.outer_pad)
             else:
-                xp = F.pad(x, (self.outer_pad,) * 4)
+                ph, pw = self.outer_pad
+                xp = F.pad(x, (pw, pw, ph, ph))
         else:
             xp = halo_pad(x, self.halo_len, self.exchanger, self.grad_mode)
         return self.conv(xp)
@@ -152,12 +157,12 @@ class HaloPool2d(_SpatialBase):
     """Spatial max/avg pool (reference Pool, spatial.py:1416): halo-pad by
     (kernel-1)//2 then pool with padding=0.
 
-    NOTE on avg-pool semantics: a single-GPU AvgPool2d with padding>0 and
-    count_include_pad=False is NOT reproducible tile-locally at interior
-    edges (the halo values are real, not pad). We therefore only claim
-    exact parity for max pool and for avg pool with padding=0; the
-    reference has the same caveat (spatial.py:1440-1441 asserts
-    count_include_pad incompatibility).
+    avg pool supports BOTH count_include_pad semantics exactly:
+    * True — zero-filled ring, every window divides by k*k (trivially
+      equals the single-GPU op);
+    * False — the divisor is the number of window cells inside the
+      GLOBAL image, computed analytically from the tile's grid position
+      (the reference simply refuses this case, spatial.py:1440-1441).
     """
 
     def __init__(
@@ -172,6 +177,7 @@ class HaloPool2d(_SpatialBase):
         rank_of_tile=None,
         grad_mode: str = "exact",
         d2: bool = False,
+        count_include_pad: bool = True,
     ):
         super().__init__(
             num_spatial_parts, slice_method, spatial_local_rank, rank_of_tile, grad_mode
@@ -182,15 +188,31 @@ class HaloPool2d(_SpatialBase):
         self.stride = stride or kernel_size
         self.halo_len = padding
         self.d2 = d2
+        self.count_include_pad = count_include_pad
+
+    def _avg_divisors(self, out_h, out_w, H_loc, W_loc, device):
+        """Per-output-position count of window cells inside the global
+        image (count_include_pad=False semantics)."""
+        k, s, p = self.kernel_size, self.stride, self.halo_len
+        r, c = self.layout.pos(self.tile)
+        Hg, Wg = H_loc * self.layout.rows, W_loc * self.layout.cols
+        gr0 = r * H_loc - p
+        gc0 = c * W_loc - p
+        rows = gr0 + torch.arange(out_h, device=device) * s
+        cols = gc0 + torch.arange(out_w, device=device) * s
+        rcnt = (torch.clamp(rows + k, max=Hg) - torch.clamp(rows, min=0)).clamp(min=0)
+        ccnt = (torch.clamp(cols + k, max=Wg) - torch.clamp(cols, min=0)).clamp(min=0)
+        return (rcnt.view(-1, 1) * ccnt.view(1, -1)).to(torch.float32)
 
     def forward(self, x):
         h = self.halo_len
         # max pool pads with -inf so image-boundary windows match the
         # single-GPU op exactly (zero-pad would win over negative inputs)
         fill = float("-inf") if self.kind == "max" else 0.0
+        H_loc, W_loc = x.shape[-2], x.shape[-1]
         if self.d2 and h > 0:
             xp = (
-                outer_pad_only(x, self.layout, self.tile, h)
+                outer_pad_only(x, self.layout, self.tile, h, fill=fill)
                 if self.exchanger is not None
                 else F.pad(x, (h,) * 4, value=fill)
             )
@@ -202,6 +224,17 @@ class HaloPool2d(_SpatialBase):
             )
         if self.kind == "max":
             return F.max_pool2d(xp, self.kernel_size, self.stride, padding=0)
-        return F.avg_pool2d(
-            xp, self.kernel_size, self.stride, padding=0, count_include_pad=True
+        if self.count_include_pad or h == 0 or self.exchanger is None:
+            return F.avg_pool2d(
+                xp, self.kernel_size, self.stride, padding=0, count_include_pad=True
+            )
+        if x.is_meta:
+            return F.avg_pool2d(xp, self.kernel_size, self.stride, padding=0)
+        k = self.kernel_size
+        sums = F.avg_pool2d(
+            xp, k, self.stride, padding=0, count_include_pad=True
+        ) * float(k * k)
+        div = self._avg_divisors(
+            sums.shape[-2], sums.shape[-1], H_loc, W_loc, sums.device
         )
+        return sums / div

@@ -213,19 +213,21 @@ class train_model_spatial(train_model):
         if not self.first_stage:
             prev = self.split_rank - 1
             spec = _as_list(self.shape_list[prev])
-            assert len(spec) == 1 or self.in_layout is None, (
-                "tuple activations across spatial seams are not supported "
-                "(the reference has the same restriction)"
-            )
             if self.in_layout is not None:
-                # per-incoming-tile buffers
-                tile_shape = scale_shape_for_tile(spec[0], self.in_layout)
+                # per-edge x per-tensor buffers (tuple activations supported:
+                # each tile sends every tensor of the tuple)
                 mb = self._local_mb()
-                shapes = [(mb,) + tuple(tile_shape[1:])] * len(self.in_edges)
+                tile_shapes = [
+                    (mb,) + tuple(scale_shape_for_tile(sp, self.in_layout)[1:])
+                    for sp in spec
+                ]
                 self.input_buffers = [
                     [
-                        torch.zeros(s, device=self.device, dtype=self.act_dtype)
-                        for s in shapes
+                        [
+                            torch.zeros(ts, device=self.device, dtype=self.act_dtype)
+                            for ts in tile_shapes
+                        ]
+                        for _ in self.in_edges
                     ]
                     for _ in range(self.parts)
                 ]
@@ -243,12 +245,28 @@ class train_model_spatial(train_model):
         if not self.last_stage:
             spec = _as_list(self.shape_list[self.split_rank])
             if self.is_tile_rank:
-                my_shape = scale_shape_for_tile(spec[0], self.layout)
+                my_shapes = [
+                    tuple(scale_shape_for_tile(sp, self.layout)) for sp in spec
+                ]
                 if self.out_mode == "joint_dp":
                     mb = self.mb // L
-                    shapes = [(mb,) + tuple(my_shape[1:])] * L
-                else:
-                    shapes = [(self.mb,) + tuple(my_shape[1:])]
+                    # per-DP-peer x per-tensor
+                    self.grad_buffers = [
+                        [
+                            [
+                                torch.zeros(
+                                    (mb,) + tuple(ms[1:]),
+                                    device=self.device,
+                                    dtype=self.act_dtype,
+                                )
+                                for ms in my_shapes
+                            ]
+                            for _ in range(L)
+                        ]
+                        for _ in range(self.parts)
+                    ]
+                    return
+                shapes = [(self.mb,) + tuple(ms[1:]) for ms in my_shapes]
             else:
                 shapes = [(self._local_mb(),) + tuple(s[1:]) for s in spec]
             self.grad_buffers = [
@@ -269,38 +287,54 @@ class train_model_spatial(train_model):
             p2p.recv_tensors(bufs, self.prev_rank, tag_base=1000 + part * 16)
             leaves = [b.clone().requires_grad_(True) for b in bufs]
             return leaves[0] if len(leaves) == 1 else tuple(leaves)
-        # gather tiles (joint / skewed / joint_dp receiving side)
+        # gather tiles (joint / skewed / joint_dp receiving side);
+        # bufs[i][j] = tensor j of the tuple from edge i
         recvs = [
-            (bufs[i], peer, 100 + i) for i, peer in enumerate(self.in_edges)
+            (b, peer, 100 + i * 4 + j)
+            for i, peer in enumerate(self.in_edges)
+            for j, b in enumerate(bufs[i])
         ]
         p2p.exchange([], recvs).wait()
-        leaves = [b.clone().requires_grad_(True) for b in bufs]
+        leaves = [[b.clone().requires_grad_(True) for b in eb] for eb in bufs]
         self._seam_leaves = getattr(self, "_seam_leaves", [None] * self.parts)
         self._seam_leaves[part] = leaves
-        # merge grid: cat cols within row, then rows (reference
-        # merge_inputs_joint_cat, train_spatial.py:1083)
-        rows = [
-            torch.cat([leaves[i] for i in row], dim=3) for row in self.in_grid
-        ]
-        return torch.cat(rows, dim=2) if len(rows) > 1 else rows[0]
+        # merge grid per tuple slot: cat cols within row, then rows
+        # (reference merge_inputs_joint_cat, train_spatial.py:1083)
+        ntens = len(bufs[0])
+        merged = []
+        for j in range(ntens):
+            rows = [
+                torch.cat([leaves[i][j] for i in row], dim=3)
+                for row in self.in_grid
+            ]
+            merged.append(torch.cat(rows, dim=2) if len(rows) > 1 else rows[0])
+        return merged[0] if ntens == 1 else tuple(merged)
 
     def send_output(self, y, part: int):
-        assert not isinstance(y, tuple), "tuple activations only on plain edges"
+        ys = list(y) if isinstance(y, tuple) else [y]
         if self.out_mode == "joint_dp":
             # batch-split across the local-DP ranks (reference :809-853)
-            chunks = y.to(self.act_dtype).chunk(len(self.out_edges), dim=0)
+            sends = []
+            for j, t in enumerate(ys):
+                chunks = t.to(self.act_dtype).chunk(len(self.out_edges), dim=0)
+                for c, peer in zip(chunks, self.out_edges):
+                    sends.append(
+                        (c.contiguous(), peer, 100 + self._joint_dp_slot() * 4 + j)
+                    )
+            self._pending.append(p2p.exchange(sends, []))
+        elif self.out_mode in ("joint", "skewed"):
+            slot = self._seam_slot()
             sends = [
-                (c.contiguous(), peer, 100 + self._joint_dp_slot())
-                for c, peer in zip(chunks, self.out_edges)
+                (t.to(self.act_dtype).contiguous(), self.out_edges[0], 100 + slot * 4 + j)
+                for j, t in enumerate(ys)
             ]
             self._pending.append(p2p.exchange(sends, []))
         else:
-            tag = (
-                100 + self._seam_slot()
-                if self.out_mode in ("joint", "skewed")
-                else 1000 + part * 16
+            tr = p2p.isend_tensors(
+                [t.to(self.act_dtype) for t in ys],
+                self.out_edges[0],
+                tag_base=1000 + part * 16,
             )
-            tr = p2p.isend_tensors([y.to(self.act_dtype)], self.out_edges[0], tag_base=tag)
             self._pending.append(tr)
 
     def _seam_slot(self) -> int:
@@ -324,19 +358,29 @@ class train_model_spatial(train_model):
     def receive_output_grad(self, part: int):
         bufs = self.grad_buffers[part]
         if self.out_mode in ("joint", "skewed"):
-            recvs = [(bufs[0], self.out_edges[0], 200 + self._seam_slot())]
+            slot = self._seam_slot()
+            recvs = [
+                (b, self.out_edges[0], 200 + slot * 4 + j)
+                for j, b in enumerate(bufs)
+            ]
             p2p.exchange([], recvs).wait()
             return bufs
         if self.out_mode == "joint_dp":
             recvs = [
-                (bufs[d], peer, 200 + self._joint_dp_slot())
+                (b, peer, 200 + self._joint_dp_slot() * 4 + j)
                 for d, peer in enumerate(self.out_edges)
+                for j, b in enumerate(bufs[d])
             ]
             p2p.exchange([], recvs).wait()
             # each local-DP rank's loss is the mean over its mb/L shard;
             # the full-batch mean is 1/L of the shard-mean sum — rescale
             # so tile-side gradients match the serial trajectory.
-            return [torch.cat(list(bufs), dim=0).div_(len(self.out_edges))]
+            L = len(self.out_edges)
+            ntens = len(bufs[0])
+            return [
+                torch.cat([bufs[d][j] for d in range(L)], dim=0).div_(L)
+                for j in range(ntens)
+            ]
         p2p.recv_tensors(bufs, self.next_rank, tag_base=3000 + part * 16)
         return bufs
 
@@ -345,11 +389,12 @@ class train_model_spatial(train_model):
             return super().send_input_grad(part)
         leaves = self._seam_leaves[part]
         sends = []
-        for i, (leaf, peer) in enumerate(zip(leaves, self.in_edges)):
-            g = (
-                leaf.grad if leaf.grad is not None else torch.zeros_like(leaf)
-            ).to(self.act_dtype)
-            sends.append((g.contiguous(), peer, 200 + i))
+        for i, (edge_leaves, peer) in enumerate(zip(leaves, self.in_edges)):
+            for j, leaf in enumerate(edge_leaves):
+                g = (
+                    leaf.grad if leaf.grad is not None else torch.zeros_like(leaf)
+                ).to(self.act_dtype)
+                sends.append((g.contiguous(), peer, 200 + i * 4 + j))
         self._pending.append(p2p.exchange(sends, []))
         self._seam_leaves[part] = None
 
