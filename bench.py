@@ -1,0 +1,222 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: AmoebaNet-D training at 2048x2048 with SP+PP
+(BASELINE.json metric: "img/sec AmoebaNet-D 2048x2048 SP+PP at 1/2/4/8
+MI355X").
+
+Launch (driver contract):
+  python bench.py --gpus N --steps K --warmup W
+  # N>1 via: python -m torch.distributed.run --nnodes=1 --nproc-per-node N
+  #          --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+Topology per N (strong scaling, fixed global batch):
+  N=1: whole model on one GPU (engine degenerates, same code path)
+  N=2: 2 pipeline stages (PP)
+  N=4: 2 spatial tiles on the first partition + 2 more LP stages (SP+PP)
+  N=8: 4 spatial tiles + 4 more LP stages (SP+PP)
+
+Synthetic data (no network for datasets), random-init weights, bf16
+autocast compute (>= the reference's fp32), loss/optimizer step included
+in the timed region. One JSON line on rank 0.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+import torch.distributed as dist
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+# reference best published number on this metric/config:
+# AmoebaNet-D 2048^2 SP-vert-D2 B=2 ~= 5.0 img/s
+BASELINE_IMGS = 5.0
+
+
+def topology(n: int):
+    if n == 1:
+        return dict(split_size=1, nsp=1, spatial_size=0)
+    if n == 2:
+        return dict(split_size=2, nsp=1, spatial_size=0)
+    if n == 4:
+        return dict(split_size=3, nsp=2, spatial_size=1)
+    if n == 8:
+        return dict(split_size=5, nsp=4, spatial_size=1)
+    raise SystemExit(f"unsupported GPU count {n} (use 1/2/4/8)")
+
+
+def make_balance(ncells: int, split: int, spatial_size: int):
+    """Spatial partitions hold the early high-resolution cells (stem + 2
+    reduction cells); remaining cells spread evenly over LP stages."""
+    if split == 1:
+        return [ncells]
+    if spatial_size == 0:
+        base, rem = divmod(ncells, split)
+        return [base + (1 if i < rem else 0) for i in range(split)]
+    sp_cells = 3  # stem + stem2 + stem3: the 1024^2..256^2 activations
+    rest = ncells - sp_cells
+    lp = split - spatial_size
+    base, rem = divmod(rest, lp)
+    return [sp_cells] + [base + (1 if i < rem else 0) for i in range(lp)]
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=3)
+    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--image-size", type=int, default=2048)
+    ap.add_argument("--batch", type=int, default=8, help="global batch per step")
+    ap.add_argument("--parts", type=int, default=8, help="pipeline micro-batches")
+    ap.add_argument("--num-layers", type=int, default=18)
+    ap.add_argument("--num-filters", type=int, default=416)
+    ap.add_argument("--num-classes", type=int, default=1000)
+    ap.add_argument("--slice-method", default="vertical")
+    ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    args = ap.parse_args()
+
+    from mpi4dl_amd.comm import Communicator, GradReducer, init_distributed
+    from mpi4dl_amd.models.amoebanet import amoebanetd
+    from mpi4dl_amd.ops.plan import SpatialPlan
+    from mpi4dl_amd.parallel.partition import model_generator
+    from mpi4dl_amd.parallel.pipeline import train_model
+    from mpi4dl_amd.parallel.spatial import train_model_spatial
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    n = args.gpus if args.gpus else world
+    assert world in (1, n), f"WORLD_SIZE {world} != --gpus {n}"
+    on_gpu = torch.cuda.is_available()
+    device = None
+    init_distributed()
+    rank = dist.get_rank()
+    if on_gpu:
+        device = torch.device("cuda", torch.cuda.current_device())
+    else:
+        device = torch.device("cpu")
+
+    topo = topology(n)
+    S = args.image_size
+    B, parts = args.batch, args.parts
+    mb = B // parts
+    autocast_dtype = torch.bfloat16 if (args.dtype == "bf16" and on_gpu) else None
+    act_dtype = torch.bfloat16 if autocast_dtype else torch.float32
+
+    spatial = topo["spatial_size"] > 0 and world > 1
+    comm = Communicator(
+        split_size=topo["split_size"],
+        ENABLE_SPATIAL=spatial,
+        num_spatial_parts=topo["nsp"],
+        spatial_size=topo["spatial_size"] if spatial else 0,
+    )
+
+    # model (full, cells) — probe cell count cheaply on meta
+    def build(plan=None):
+        torch.manual_seed(0)
+        return amoebanetd(args.num_classes, args.num_layers, args.num_filters, plan)
+
+    with torch.device("meta"):
+        ncells = len(build())
+    balance = make_balance(ncells, topo["split_size"], topo["spatial_size"] if spatial else 0)
+
+    plan = (
+        SpatialPlan(comm, balance, args.slice_method)
+        if spatial
+        else None
+    )
+    model = build(plan)
+    gen = model_generator(model, topo["split_size"], (mb, 3, S, S), balance=balance)
+    gen.get_output_shapes()
+    gen.ready_model(comm.split_rank, device=device)
+    del model  # free remote stages
+
+    opt = torch.optim.SGD(gen.models.parameters(), lr=0.01, momentum=0.9)
+    eng_kw = dict(
+        optimizer=opt,
+        device=device,
+        autocast_dtype=autocast_dtype,
+        act_dtype=act_dtype,
+    )
+    if spatial:
+        eng = train_model_spatial(
+            gen, comm.local_rank, B, parts, comm,
+            slice_method=args.slice_method, **eng_kw,
+        )
+    else:
+        eng = train_model(gen, comm.local_rank, B, parts, comm, **eng_kw)
+    reducer = GradReducer(comm)
+
+    # synthetic data of the benchmark shape
+    torch.manual_seed(1234 + rank)
+    x = torch.randn(B, 3, S, S, device=device, dtype=torch.float32)
+    y = torch.randint(0, args.num_classes, (B,), device=device)
+
+    def step():
+        loss, _, _ = eng.run_step(x, y)
+        if spatial or comm.dp_size > 1:
+            reducer.apply_allreduce(eng.models)
+        eng.update()
+        return loss
+
+    def fence():
+        if world > 1:
+            dist.barrier()
+        if on_gpu:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        step()
+    fence()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    fence()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    t = torch.tensor([elapsed], dtype=torch.float64)
+    if world > 1:
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = float(t[0])
+
+    imgs_per_s = B * args.steps / elapsed
+    if rank == 0:
+        par = "single" if n == 1 else (
+            f"pp{topo['split_size']}" if not spatial
+            else f"sp{topo['nsp']}+pp{topo['split_size']}"
+        )
+        print(
+            json.dumps(
+                {
+                    "metric": "img/sec AmoebaNet-D 2048x2048 SP+PP",
+                    "value": imgs_per_s,
+                    "unit": "img/s",
+                    "n_gpus": n,
+                    "steps": args.steps,
+                    "warmup": args.warmup,
+                    "ms_per_step": elapsed / args.steps * 1e3,
+                    "higher_is_better": True,
+                    "scaling": "strong",
+                    "vs_baseline": imgs_per_s / BASELINE_IMGS,
+                    "dtype": args.dtype if on_gpu else "fp32",
+                    "data": "synthetic",
+                    "config": {
+                        "model": f"amoebanet-d(L{args.num_layers},F{args.num_filters})",
+                        "global_batch": B,
+                        "parts": parts,
+                        "seq_len": None,
+                        "image_size": S,
+                        "parallelism": par,
+                        "slice_method": args.slice_method if spatial else None,
+                    },
+                }
+            ),
+            flush=True,
+        )
+
+
+if __name__ == "__main__":
+    main()

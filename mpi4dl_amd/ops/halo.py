@@ -159,6 +159,8 @@ class HaloExchanger:
         hh, hw = _hpair(h)
         if (hh == 0 and hw == 0) or not self.neigh:
             return
+        if xp.is_cuda:
+            return self._exchange_padded_gpu(xp, h)
         H, W = xp.shape[-2] - 2 * hh, xp.shape[-1] - 2 * hw
         sends, recvs = [], []
         for d, t in self.neigh:
@@ -184,6 +186,83 @@ class HaloExchanger:
         for rbuf, _, _, (rs, re, cs, ce) in recvs:
             xp[:, :, rs:re, cs:ce].copy_(rbuf)
 
+    # -- GPU fast path: gemscore pack/unpack + flat staging buffers ---------
+
+    def _plan_gpu(self, xp, h, grad: bool):
+        """Cached (descs, flat buffers, per-strip views, peers/tags)."""
+        key = (tuple(xp.shape), _hpair(h), xp.dtype, grad)
+        cache = getattr(self, "_gpu_plans", None)
+        if cache is None:
+            cache = self._gpu_plans = {}
+        if key in cache:
+            return cache[key]
+        hh, hw = _hpair(h)
+        H, W = xp.shape[-2] - 2 * hh, xp.shape[-1] - 2 * hw
+        n, c = xp.shape[0], xp.shape[1]
+        tagb = 8 if grad else 0
+        out_rows, in_rows, sends, recvs = [], [], [], []
+        s_off = r_off = 0
+        for d, t in self.neigh:
+            if (d[0] != 0 and hh == 0) or (d[1] != 0 and hw == 0):
+                continue
+            peer = self.rank_of_tile(t)
+            # forward: pack send_region / unpack recv_region.
+            # grad (transposed): pack recv_region / unpack-add send_region.
+            (rs, re), (cs, ce) = (recv_region if grad else send_region)(d, H, W, h)
+            sz = n * c * (re - rs) * (ce - cs)
+            out_rows.append([rs, cs, re - rs, ce - cs])
+            sends.append((peer, tagb + _DIR_IDX[_opposite(d)], s_off, sz))
+            s_off += sz
+            (rs, re), (cs, ce) = (send_region if grad else recv_region)(d, H, W, h)
+            sz = n * c * (re - rs) * (ce - cs)
+            if grad:
+                # unpack-add targets the UNpadded grad tile
+                in_rows.append([rs - hh, cs - hw, re - rs, ce - cs])
+            else:
+                in_rows.append([rs, cs, re - rs, ce - cs])
+            recvs.append((peer, tagb + _DIR_IDX[d], r_off, sz))
+            r_off += sz
+        plan = {
+            "sdesc": torch.tensor(out_rows, dtype=torch.int64),
+            "rdesc": torch.tensor(in_rows, dtype=torch.int64),
+            "sbuf": torch.empty(s_off, device=xp.device, dtype=xp.dtype),
+            "rbuf": torch.empty(r_off, device=xp.device, dtype=xp.dtype),
+            "sends": sends,
+            "recvs": recvs,
+        }
+        cache[key] = plan
+        return plan
+
+    def _exchange_padded_gpu(self, xp, h):
+        from . import backend
+
+        ge = backend.ext()
+        pl = self._plan_gpu(xp, h, grad=False)
+        ge.halo_pack(xp, pl["sbuf"], pl["sdesc"])
+        tr = p2p.exchange(
+            [(pl["sbuf"].narrow(0, o, s), peer, tag) for peer, tag, o, s in pl["sends"]],
+            [(pl["rbuf"].narrow(0, o, s), peer, tag) for peer, tag, o, s in pl["recvs"]],
+        )
+        tr.wait()
+        ge.halo_unpack(xp, pl["rbuf"], pl["rdesc"])
+
+    def _exchange_grad_padded_gpu(self, gp, h):
+        from . import backend
+
+        ge = backend.ext()
+        hh, hw = _hpair(h)
+        H, W = gp.shape[-2] - 2 * hh, gp.shape[-1] - 2 * hw
+        g = gp[:, :, hh : hh + H, hw : hw + W].clone().contiguous()
+        pl = self._plan_gpu(gp, h, grad=True)
+        ge.halo_pack(gp, pl["sbuf"], pl["sdesc"])
+        tr = p2p.exchange(
+            [(pl["sbuf"].narrow(0, o, s), peer, tag) for peer, tag, o, s in pl["sends"]],
+            [(pl["rbuf"].narrow(0, o, s), peer, tag) for peer, tag, o, s in pl["recvs"]],
+        )
+        tr.wait()
+        ge.halo_unpack_add(g, pl["rbuf"], pl["rdesc"])
+        return g
+
     # -- backward (transposed) ------------------------------------------------
 
     def exchange_grad_padded(self, gp: torch.Tensor, h: int) -> torch.Tensor:
@@ -195,6 +274,8 @@ class HaloExchanger:
         """
         hh, hw = _hpair(h)
         H, W = gp.shape[-2] - 2 * hh, gp.shape[-1] - 2 * hw
+        if ((hh or hw) and self.neigh) and gp.is_cuda:
+            return self._exchange_grad_padded_gpu(gp, h)
         g = gp[:, :, hh : hh + H, hw : hw + W].clone()
         if (hh == 0 and hw == 0) or not self.neigh:
             return g
