@@ -149,7 +149,13 @@ class Communicator:
         LOCAL_DP_LP: int = 1,
         DISABLE_INIT: bool = False,
         backend: Optional[str] = None,
+        ENABLE_GEMS: bool = False,
     ):
+        # ENABLE_GEMS builds the GEMS pair/mirror groups WITHOUT the
+        # local-rank inversion (our single-communicator GEMS design:
+        # parallel/gems.py supplies mirrored positions itself; the
+        # reference instead builds a second inverted MPIComm,
+        # comm.py:77-80 + sync_comms_for_master).
         if not ENABLE_SPATIAL:
             spatial_size = 0
             num_spatial_parts = 1
@@ -158,6 +164,7 @@ class Communicator:
         assert dist.is_initialized(), "torch.distributed must be initialised"
 
         self.ENABLE_MASTER = ENABLE_MASTER
+        self.ENABLE_GEMS = ENABLE_GEMS or ENABLE_MASTER
         self.ENABLE_SPATIAL = ENABLE_SPATIAL
         self.split_size = split_size
         self.spatial_size = spatial_size
@@ -268,17 +275,31 @@ class Communicator:
         # GEMS master pairs: {j, mp-1-j} per replica (comm.py:170-195) — the
         # two global ranks that host the same partition in the two mirrored
         # engines, used to average the paired grads.
-        if self.ENABLE_MASTER or True:  # cheap; built unconditionally when useful
-            if self.mp_size > 1 and self.ENABLE_MASTER:
-                for rep in range(self.dp_size):
-                    for j in range(self.mp_size):
-                        a, b = j, self.mp_size - 1 - j
-                        if a > b:
-                            continue
-                        g = self._new_group([rep * self.mp_size + a, rep * self.mp_size + b])
-                        if rep == self.replica:
-                            self.master_pair_groups[a] = g
-                            self.master_pair_groups[b] = g
+        if self.mp_size > 1 and self.ENABLE_GEMS:
+            for rep in range(self.dp_size):
+                for j in range(self.mp_size):
+                    a, b = j, self.mp_size - 1 - j
+                    if a > b:
+                        continue
+                    g = self._new_group([rep * self.mp_size + a, rep * self.mp_size + b])
+                    if rep == self.replica:
+                        self.master_pair_groups[a] = g
+                        self.master_pair_groups[b] = g
+
+        # Mirror spatial groups for the GEMS-inverse engine: partition p's
+        # tiles of replica 2 live on global ranks mp-1-lr (used for its
+        # spatial grad reduction and TileBatchNorm sync).
+        self.mirror_spatial_groups = {}
+        if self.ENABLE_GEMS and self.spatial_size > 0:
+            for rep in range(self.dp_size):
+                for part in range(self.spatial_size):
+                    ranks = [
+                        rep * self.mp_size + (self.mp_size - 1 - lr)
+                        for lr in self.ranks_of_partition(part)
+                    ]
+                    g = self._new_group(ranks)
+                    if rep == self.replica:
+                        self.mirror_spatial_groups[part] = g
 
         # SP<->LP scatter/gather seam groups for LOCAL_DP_LP (comm.py:250-276):
         # one group per tile rank of the *last* spatial partition: that tile

@@ -109,12 +109,37 @@ void halo_copy(torch::Tensor tile, torch::Tensor buf, torch::Tensor desc) {
   if (total == 0) return;
   TORCH_CHECK(buf.numel() >= total, "staging buffer too small");
   const int block = 256;
-  int64_t g = (total + block - 1) / block;
-  const int grid = (int)std::min<int64_t>(g, 2048);  // grid-stride (G11)
   auto stream = at::cuda::getCurrentCUDAStream();
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::Half, at::ScalarType::BFloat16, tile.scalar_type(),
       "halo_copy", [&] {
+        if (MODE == 2) {
+          // grad strips may OVERLAP at corners (each neighbour received a
+          // replicated copy of the corner in forward, so each contributes
+          // a gradient): launch per strip, stream-ordered, so the adds
+          // into the overlap are sequential instead of racing.
+          for (int i = 0; i < a.nstrips; ++i) {
+            StripArgs one;
+            one.nstrips = 1;
+            one.nc = a.nc;
+            one.plane_stride = a.plane_stride;
+            one.wp = a.wp;
+            one.s[0] = a.s[i];
+            const int64_t t1 = (int64_t)a.s[i].rows * a.s[i].cols * a.nc;
+            one.s[0].buf_off = a.s[i].buf_off;
+            const int grid1 =
+                (int)std::min<int64_t>((t1 + block - 1) / block, 2048);
+            hipLaunchKernelGGL((halo_copy_kernel<scalar_t, MODE>), dim3(grid1),
+                               dim3(block), 0, stream.stream(),
+                               tile.data_ptr<scalar_t>(),
+                               buf.data_ptr<scalar_t>() + a.s[i].buf_off -
+                                   one.s[0].buf_off,
+                               one, t1);
+          }
+          return;
+        }
+        const int grid =
+            (int)std::min<int64_t>((total + block - 1) / block, 2048);
         hipLaunchKernelGGL((halo_copy_kernel<scalar_t, MODE>), dim3(grid),
                            dim3(block), 0, stream.stream(),
                            tile.data_ptr<scalar_t>(), buf.data_ptr<scalar_t>(),

@@ -51,7 +51,14 @@ class SpatialPlan:
         nparts = comm.spatial_parts[part]
         # tile index of THIS rank in that partition; ranks outside the
         # partition never execute these cells for real (meta only).
-        tile = comm.local_rank - start
+        # A GEMS-inverse engine occupies the MIRRORED in-clique position
+        # (reference spatial.py:913-918).
+        pos = (
+            comm.mp_size - 1 - comm.local_rank
+            if self.gems_inverse
+            else comm.local_rank
+        )
+        tile = pos - start
         if not (0 <= tile < nparts):
             tile = 0
         inv = self.gems_inverse
