@@ -102,20 +102,27 @@ def _hpair(h):
     return int(h), int(h)
 
 
-def send_region(d, H, W, h):
-    """Interior boundary band (in padded coords) sent toward direction d."""
+def send_region(d, H, W, h, off=None):
+    """Interior boundary band (in padded coords) sent toward direction d.
+
+    ``off`` = (top, left) pad offsets of the nominal region inside the
+    padded tensor; defaults to (hh, hw) (symmetric pad). D2 tiles pad
+    interior sides only, so boundary tiles pass off 0 for those sides.
+    """
     hh, hw = _hpair(h)
+    t, l = (hh, hw) if off is None else off
     dr, dc = d
-    rs = {-1: (hh, 2 * hh), 0: (hh, hh + H), 1: (H, hh + H)}[dr]
-    cs = {-1: (hw, 2 * hw), 0: (hw, hw + W), 1: (W, hw + W)}[dc]
+    rs = {-1: (t, t + hh), 0: (t, t + H), 1: (t + H - hh, t + H)}[dr]
+    cs = {-1: (l, l + hw), 0: (l, l + W), 1: (l + W - hw, l + W)}[dc]
     return rs, cs
 
 
-def recv_region(d, H, W, h):
+def recv_region(d, H, W, h, off=None):
     hh, hw = _hpair(h)
+    t, l = (hh, hw) if off is None else off
     dr, dc = d
-    rs = {-1: (0, hh), 0: (hh, hh + H), 1: (hh + H, 2 * hh + H)}[dr]
-    cs = {-1: (0, hw), 0: (hw, hw + W), 1: (hw + W, 2 * hw + W)}[dc]
+    rs = {-1: (t - hh, t), 0: (t, t + H), 1: (t + H, t + H + hh)}[dr]
+    cs = {-1: (l - hw, l), 0: (l, l + W), 1: (l + W, l + W + hw)}[dc]
     return rs, cs
 
 
@@ -149,29 +156,46 @@ class HaloExchanger:
 
     # -- forward -------------------------------------------------------------
 
-    def exchange_padded(self, xp: torch.Tensor, h: int) -> None:
+    def pads_d2(self, h):
+        """Per-side pad amounts (top, bottom, left, right) for D2 mode:
+        interior sides get the halo pad, image-boundary sides get none
+        (each conv re-applies its own zero pad there)."""
+        hh, hw = _hpair(h)
+        r, c = self.layout.pos(self.tile)
+        t = hh if r > 0 else 0
+        b = hh if r < self.layout.rows - 1 else 0
+        l = hw if c > 0 else 0
+        rr = hw if c < self.layout.cols - 1 else 0
+        return t, b, l, rr
+
+    def exchange_padded(self, xp: torch.Tensor, h: int, off=None, nominal=None) -> None:
         """In-place: fill xp's pad ring (width h) from neighbours.
 
-        xp: (N, C, H+2h, W+2h), already zero-padded. Blocks until the
-        ring is filled (async overlap is handled by HaloConv2d's
-        interior/boundary split, not here).
+        xp: (N, C, H+2h, W+2h), already zero-padded (symmetric) — or, in
+        D2 mode, padded on interior sides only, with ``off`` = (top,
+        left) pad offsets and ``nominal`` = (H, W) of the nominal
+        region. Blocks until the ring is filled (async overlap is
+        handled by HaloConv2d's interior/boundary split, not here).
         """
         hh, hw = _hpair(h)
         if (hh == 0 and hw == 0) or not self.neigh:
             return
         if xp.is_cuda:
-            return self._exchange_padded_gpu(xp, h)
-        H, W = xp.shape[-2] - 2 * hh, xp.shape[-1] - 2 * hw
+            return self._exchange_padded_gpu(xp, h, off, nominal)
+        if nominal is None:
+            H, W = xp.shape[-2] - 2 * hh, xp.shape[-1] - 2 * hw
+        else:
+            H, W = nominal
         sends, recvs = [], []
         for d, t in self.neigh:
             if (d[0] != 0 and hh == 0) or (d[1] != 0 and hw == 0):
                 continue  # no halo along that axis
             peer = self.rank_of_tile(t)
-            (rs, re), (cs, ce) = send_region(d, H, W, h)
+            (rs, re), (cs, ce) = send_region(d, H, W, h, off)
             buf = xp[:, :, rs:re, cs:ce].contiguous()
             # receiver tags by arrival direction = opposite of my send dir
             sends.append((buf, peer, _DIR_IDX[_opposite(d)]))
-            (rs, re), (cs, ce) = recv_region(d, H, W, h)
+            (rs, re), (cs, ce) = recv_region(d, H, W, h, off)
             rbuf = torch.empty(
                 (xp.shape[0], xp.shape[1], re - rs, ce - cs),
                 device=xp.device,
@@ -188,16 +212,20 @@ class HaloExchanger:
 
     # -- GPU fast path: gemscore pack/unpack + flat staging buffers ---------
 
-    def _plan_gpu(self, xp, h, grad: bool):
+    def _plan_gpu(self, xp, h, grad: bool, off=None, nominal=None):
         """Cached (descs, flat buffers, per-strip views, peers/tags)."""
-        key = (tuple(xp.shape), _hpair(h), xp.dtype, grad)
+        key = (tuple(xp.shape), _hpair(h), xp.dtype, grad, off, nominal)
         cache = getattr(self, "_gpu_plans", None)
         if cache is None:
             cache = self._gpu_plans = {}
         if key in cache:
             return cache[key]
         hh, hw = _hpair(h)
-        H, W = xp.shape[-2] - 2 * hh, xp.shape[-1] - 2 * hw
+        if nominal is None:
+            H, W = xp.shape[-2] - 2 * hh, xp.shape[-1] - 2 * hw
+        else:
+            H, W = nominal
+        t0, l0 = (hh, hw) if off is None else off
         n, c = xp.shape[0], xp.shape[1]
         tagb = 8 if grad else 0
         out_rows, in_rows, sends, recvs = [], [], [], []
@@ -208,16 +236,20 @@ class HaloExchanger:
             peer = self.rank_of_tile(t)
             # forward: pack send_region / unpack recv_region.
             # grad (transposed): pack recv_region / unpack-add send_region.
-            (rs, re), (cs, ce) = (recv_region if grad else send_region)(d, H, W, h)
+            (rs, re), (cs, ce) = (recv_region if grad else send_region)(
+                d, H, W, h, off
+            )
             sz = n * c * (re - rs) * (ce - cs)
             out_rows.append([rs, cs, re - rs, ce - cs])
             sends.append((peer, tagb + _DIR_IDX[_opposite(d)], s_off, sz))
             s_off += sz
-            (rs, re), (cs, ce) = (send_region if grad else recv_region)(d, H, W, h)
+            (rs, re), (cs, ce) = (send_region if grad else recv_region)(
+                d, H, W, h, off
+            )
             sz = n * c * (re - rs) * (ce - cs)
             if grad:
                 # unpack-add targets the UNpadded grad tile
-                in_rows.append([rs - hh, cs - hw, re - rs, ce - cs])
+                in_rows.append([rs - t0, cs - l0, re - rs, ce - cs])
             else:
                 in_rows.append([rs, cs, re - rs, ce - cs])
             recvs.append((peer, tagb + _DIR_IDX[d], r_off, sz))
@@ -233,11 +265,11 @@ class HaloExchanger:
         cache[key] = plan
         return plan
 
-    def _exchange_padded_gpu(self, xp, h):
+    def _exchange_padded_gpu(self, xp, h, off=None, nominal=None):
         from . import backend
 
         ge = backend.ext()
-        pl = self._plan_gpu(xp, h, grad=False)
+        pl = self._plan_gpu(xp, h, grad=False, off=off, nominal=nominal)
         ge.halo_pack(xp, pl["sbuf"], pl["sdesc"])
         tr = p2p.exchange(
             [(pl["sbuf"].narrow(0, o, s), peer, tag) for peer, tag, o, s in pl["sends"]],
@@ -246,14 +278,19 @@ class HaloExchanger:
         tr.wait()
         ge.halo_unpack(xp, pl["rbuf"], pl["rdesc"])
 
-    def _exchange_grad_padded_gpu(self, gp, h):
+    def _exchange_grad_padded_gpu(self, gp, h, off=None, nominal=None):
         from . import backend
 
         ge = backend.ext()
         hh, hw = _hpair(h)
-        H, W = gp.shape[-2] - 2 * hh, gp.shape[-1] - 2 * hw
-        g = gp[:, :, hh : hh + H, hw : hw + W].clone().contiguous()
-        pl = self._plan_gpu(gp, h, grad=True)
+        if nominal is None:
+            H, W = gp.shape[-2] - 2 * hh, gp.shape[-1] - 2 * hw
+            t0, l0 = hh, hw
+        else:
+            H, W = nominal
+            t0, l0 = off
+        g = gp[:, :, t0 : t0 + H, l0 : l0 + W].clone().contiguous()
+        pl = self._plan_gpu(gp, h, grad=True, off=off, nominal=nominal)
         ge.halo_pack(gp, pl["sbuf"], pl["sdesc"])
         tr = p2p.exchange(
             [(pl["sbuf"].narrow(0, o, s), peer, tag) for peer, tag, o, s in pl["sends"]],
@@ -265,7 +302,9 @@ class HaloExchanger:
 
     # -- backward (transposed) ------------------------------------------------
 
-    def exchange_grad_padded(self, gp: torch.Tensor, h: int) -> torch.Tensor:
+    def exchange_grad_padded(
+        self, gp: torch.Tensor, h: int, off=None, nominal=None
+    ) -> torch.Tensor:
         """Transposed halo exchange: return grad wrt the UNpadded tile.
 
         gp: gradient wrt the padded tile (N, C, H+2h, W+2h). The pad-ring
@@ -273,10 +312,15 @@ class HaloExchanger:
         owner; add received bands into my interior edge regions.
         """
         hh, hw = _hpair(h)
-        H, W = gp.shape[-2] - 2 * hh, gp.shape[-1] - 2 * hw
+        if nominal is None:
+            H, W = gp.shape[-2] - 2 * hh, gp.shape[-1] - 2 * hw
+            t0, l0 = hh, hw
+        else:
+            H, W = nominal
+            t0, l0 = off
         if ((hh or hw) and self.neigh) and gp.is_cuda:
-            return self._exchange_grad_padded_gpu(gp, h)
-        g = gp[:, :, hh : hh + H, hw : hw + W].clone()
+            return self._exchange_grad_padded_gpu(gp, h, off, nominal)
+        g = gp[:, :, t0 : t0 + H, l0 : l0 + W].clone()
         if (hh == 0 and hw == 0) or not self.neigh:
             return g
         sends, recvs = [], []
@@ -287,11 +331,11 @@ class HaloExchanger:
             # the band I received FROM d in forward carries grads for the
             # neighbour's interior: send it back tagged with my direction
             # as seen by the receiver (= opposite(d)).
-            (rs, re), (cs, ce) = recv_region(d, H, W, h)
+            (rs, re), (cs, ce) = recv_region(d, H, W, h, off)
             buf = gp[:, :, rs:re, cs:ce].contiguous()
             sends.append((buf, peer, 8 + _DIR_IDX[_opposite(d)]))
             # I get back grads for the strips I SENT in forward
-            (rs, re), (cs, ce) = send_region(d, H, W, h)
+            (rs, re), (cs, ce) = send_region(d, H, W, h, off)
             rbuf = torch.empty(
                 (gp.shape[0], gp.shape[1], re - rs, ce - cs),
                 device=gp.device,
@@ -305,7 +349,7 @@ class HaloExchanger:
         tr.wait()
         for rbuf, _, _, (rs, re, cs, ce) in recvs:
             # send_region coords are in padded space; shift to unpadded
-            g[:, :, rs - hh : re - hh, cs - hw : ce - hw].add_(rbuf)
+            g[:, :, rs - t0 : re - t0, cs - l0 : ce - l0].add_(rbuf)
         return g
 
 
@@ -344,3 +388,42 @@ def halo_pad(x, h, exchanger: HaloExchanger, grad_mode: str = "exact", fill: flo
     if exchanger is None or not exchanger.neigh:
         return F.pad(x, (hw, hw, hh, hh), value=fill)
     return _HaloPadFn.apply(x, h, exchanger, grad_mode, fill)
+
+
+class _HaloPadD2Fn(torch.autograd.Function):
+    """D2 pad: interior sides only (boundary sides stay unpadded — each
+    conv re-applies its own zero pad there, reference spatial.py:67-111)."""
+
+    @staticmethod
+    def forward(ctx, x, h, exchanger, grad_mode):
+        t, b, l, r = exchanger.pads_d2(h)
+        ctx.h = h
+        ctx.exchanger = exchanger
+        ctx.grad_mode = grad_mode
+        ctx.nominal = (x.shape[-2], x.shape[-1])
+        ctx.off = (t, l)
+        xp = F.pad(x, (l, r, t, b))
+        if not x.is_meta:
+            exchanger.exchange_padded(xp, h, off=(t, l), nominal=ctx.nominal)
+        return xp
+
+    @staticmethod
+    def backward(ctx, gp):
+        t, l = ctx.off
+        H, W = ctx.nominal
+        if ctx.grad_mode == "exact" and not gp.is_meta:
+            g = ctx.exchanger.exchange_grad_padded(
+                gp.contiguous(), ctx.h, off=ctx.off, nominal=ctx.nominal
+            )
+        else:
+            g = gp[:, :, t : t + H, l : l + W]
+        return g, None, None, None
+
+
+def halo_pad_d2(x, h, exchanger: HaloExchanger, grad_mode: str = "exact"):
+    """D2 fused-halo pad: grow the tile by h on interior sides only and
+    fill from neighbours; the surplus is then consumed by `fused_layers`
+    unpadded convs (reference resnet_spatial_d2.py design)."""
+    if exchanger is None or not exchanger.neigh:
+        return x
+    return _HaloPadD2Fn.apply(x, h, exchanger, grad_mode)
