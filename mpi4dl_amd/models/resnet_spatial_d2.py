@@ -65,11 +65,15 @@ class BottleneckV2D2(nn.Module):
     def __init__(self, in_ch, mid_ch, ctx, mknorm=nn.BatchNorm2d, pre_h: int = 0):
         super().__init__()
         out_ch = mid_ch * self.expansion
-        # build a throwaway HaloConv2d only to own an exchanger/layout
-        probe = HaloConv2d(1, 1, 3, **ctx)
-        self.exchanger = probe.exchanger
-        self.layout = probe.layout
-        self.tile = probe.tile
+        from ..ops.halo import HaloExchanger, TileLayout
+
+        self.layout = TileLayout(ctx["num_spatial_parts"], ctx["slice_method"])
+        self.tile = ctx["spatial_local_rank"]
+        self.exchanger = (
+            HaloExchanger(self.layout, self.tile, ctx["rank_of_tile"])
+            if ctx["num_spatial_parts"] > 1
+            else None
+        )
         self.grad_mode = ctx.get("grad_mode", "exact")
         self.pre_h = pre_h
         self.pre = nn.Sequential(mknorm(in_ch), nn.ReLU(inplace=True))
