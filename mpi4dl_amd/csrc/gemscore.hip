@@ -201,6 +201,480 @@ torch::Tensor bn_stats(torch::Tensor x) {
   return out;
 }
 
+
+// ---------------------------------------------------------------------------
+// Pooling (NCHW), padding-aware, global-geometry divisors.
+//
+// Replaces torch's eager max_pool NCHW kernels (15.5% of step time in
+// profiles/r01_bench_single_gpu_kernel_stats.txt). Backward is
+// gather-formulated (each input position scans the <=ceil(k/s)^2 windows
+// that contain it) — no atomics.
+//
+// Geometry: output (oh,ow) covers input rows [oh*s-p, oh*s-p+k). For
+// tile-parallel exact avg (count_include_pad=False) the divisor counts
+// window cells inside the GLOBAL image: global row = gr0 + oh*s + kh
+// with gr0 = tile_row_offset - p; plain case: gr0 = -p, Hg = H.
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void maxpool_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
+                                   int32_t* __restrict__ idx, int64_t NC,
+                                   int H, int W, int OH, int OW, int k, int s,
+                                   int p) {
+  const int64_t total = NC * OH * OW;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    const int ow = (int)(i % OW);
+    const int oh = (int)((i / OW) % OH);
+    const int64_t plane = i / ((int64_t)OW * OH);
+    const T* xp = x + plane * H * W;
+    const int h0 = oh * s - p, w0 = ow * s - p;
+    float best = -INFINITY;
+    int32_t bidx = 0;
+    for (int kh = 0; kh < k; ++kh) {
+      const int h = h0 + kh;
+      if (h < 0 || h >= H) continue;
+      for (int kw = 0; kw < k; ++kw) {
+        const int w = w0 + kw;
+        if (w < 0 || w >= W) continue;
+        const float v = (float)xp[h * W + w];
+        if (v > best) { best = v; bidx = h * W + w; }
+      }
+    }
+    y[i] = (T)best;
+    idx[i] = bidx;
+  }
+}
+
+template <typename T>
+__global__ void maxpool_bwd_kernel(const T* __restrict__ go,
+                                   const int32_t* __restrict__ idx,
+                                   T* __restrict__ gi, int64_t NC, int H,
+                                   int W, int OH, int OW, int k, int s,
+                                   int p) {
+  const int64_t total = NC * H * W;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    const int w = (int)(i % W);
+    const int h = (int)((i / W) % H);
+    const int64_t plane = i / ((int64_t)W * H);
+    const int32_t me = h * W + w;
+    // windows with oh*s - p <= h <= oh*s - p + k - 1
+    int oh_lo = (h + p - k + 1 + s - 1) / s; if (oh_lo < 0) oh_lo = 0;
+    int oh_hi = (h + p) / s; if (oh_hi > OH - 1) oh_hi = OH - 1;
+    int ow_lo = (w + p - k + 1 + s - 1) / s; if (ow_lo < 0) ow_lo = 0;
+    int ow_hi = (w + p) / s; if (ow_hi > OW - 1) ow_hi = OW - 1;
+    float acc = 0.f;
+    const T* gop = go + plane * OH * OW;
+    const int32_t* ip = idx + plane * OH * OW;
+    for (int oh = oh_lo; oh <= oh_hi; ++oh)
+      for (int ow = ow_lo; ow <= ow_hi; ++ow) {
+        const int o = oh * OW + ow;
+        if (ip[o] == me) acc += (float)gop[o];
+      }
+    gi[i] = (T)acc;
+  }
+}
+
+__device__ __forceinline__ float avg_div(int oh, int ow, int s, int k,
+                                         long gr0, long gc0, long Hg, long Wg,
+                                         int include_pad) {
+  if (include_pad) return (float)(k * k);
+  long r0 = gr0 + (long)oh * s, c0 = gc0 + (long)ow * s;
+  long r1 = r0 + k, c1 = c0 + k;
+  if (r0 < 0) r0 = 0; if (c0 < 0) c0 = 0;
+  if (r1 > Hg) r1 = Hg; if (c1 > Wg) c1 = Wg;
+  long cnt = (r1 - r0) * (c1 - c0);
+  return cnt > 0 ? (float)cnt : 1.f;
+}
+
+template <typename T>
+__global__ void avgpool_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
+                                   int64_t NC, int H, int W, int OH, int OW,
+                                   int k, int s, int p, long gr0, long gc0,
+                                   long Hg, long Wg, int include_pad) {
+  const int64_t total = NC * OH * OW;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    const int ow = (int)(i % OW);
+    const int oh = (int)((i / OW) % OH);
+    const int64_t plane = i / ((int64_t)OW * OH);
+    const T* xp = x + plane * H * W;
+    const int h0 = oh * s - p, w0 = ow * s - p;
+    float acc = 0.f;
+    for (int kh = 0; kh < k; ++kh) {
+      const int h = h0 + kh;
+      if (h < 0 || h >= H) continue;
+      for (int kw = 0; kw < k; ++kw) {
+        const int w = w0 + kw;
+        if (w < 0 || w >= W) continue;
+        acc += (float)xp[h * W + w];
+      }
+    }
+    y[i] = (T)(acc / avg_div(oh, ow, s, k, gr0, gc0, Hg, Wg, include_pad));
+  }
+}
+
+template <typename T>
+__global__ void avgpool_bwd_kernel(const T* __restrict__ go, T* __restrict__ gi,
+                                   int64_t NC, int H, int W, int OH, int OW,
+                                   int k, int s, int p, long gr0, long gc0,
+                                   long Hg, long Wg, int include_pad) {
+  const int64_t total = NC * H * W;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    const int w = (int)(i % W);
+    const int h = (int)((i / W) % H);
+    const int64_t plane = i / ((int64_t)W * H);
+    int oh_lo = (h + p - k + 1 + s - 1) / s; if (oh_lo < 0) oh_lo = 0;
+    int oh_hi = (h + p) / s; if (oh_hi > OH - 1) oh_hi = OH - 1;
+    int ow_lo = (w + p - k + 1 + s - 1) / s; if (ow_lo < 0) ow_lo = 0;
+    int ow_hi = (w + p) / s; if (ow_hi > OW - 1) ow_hi = OW - 1;
+    float acc = 0.f;
+    const T* gop = go + plane * OH * OW;
+    for (int oh = oh_lo; oh <= oh_hi; ++oh)
+      for (int ow = ow_lo; ow <= ow_hi; ++ow)
+        acc += (float)gop[oh * OW + ow] /
+               avg_div(oh, ow, s, k, gr0, gc0, Hg, Wg, include_pad);
+    gi[i] = (T)acc;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Fused BatchNorm (NCHW): fp64-accumulated channel stats, one
+// normalize+affine(+ReLU) pass, gather-formulated backward.
+// Replaces MIOpenBatchNorm{Fwd,Bwd}Spatial (14.3% of step time) and the
+// surrounding unfused elementwise casts; doubles as the local-stats leg
+// of TileBatchNorm2d's cross-tile sync (python does one allreduce of the
+// [sum,sumsq] / [gsum,gxsum] vectors between the two kernels).
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void bn_stats64_kernel(const T* __restrict__ x,
+                                  float* __restrict__ out, int64_t N,
+                                  int64_t C, int64_t HW) {
+  for (int64_t ch = blockIdx.x; ch < C; ch += gridDim.x) {
+    double s = 0.0, ss = 0.0;
+    for (int64_t n = 0; n < N; ++n) {
+      const T* p = x + (n * C + ch) * HW;
+      for (int64_t i = threadIdx.x; i < HW; i += blockDim.x) {
+        const double v = (double)(float)p[i];
+        s += v;
+        ss += v * v;
+      }
+    }
+    for (int off = 32; off > 0; off >>= 1) {
+      s += __shfl_down(s, off, 64);
+      ss += __shfl_down(ss, off, 64);
+    }
+    __shared__ double ls[8], lss[8];
+    const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
+    if (lane == 0) { ls[wid] = s; lss[wid] = ss; }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      double ts = 0.0, tss = 0.0;
+      for (int wv = 0; wv < (int)(blockDim.x >> 6); ++wv) { ts += ls[wv]; tss += lss[wv]; }
+      out[ch] = (float)ts;
+      out[C + ch] = (float)tss;
+    }
+    __syncthreads();
+  }
+}
+
+template <typename T, bool RELU>
+__global__ void bn_apply_kernel(const T* __restrict__ x, T* __restrict__ y,
+                                const float* __restrict__ mean,
+                                const float* __restrict__ invstd,
+                                const float* __restrict__ wgt,
+                                const float* __restrict__ bias, int64_t N,
+                                int64_t C, int64_t HW) {
+  const int64_t total = N * C * HW;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    const int64_t ch = (i / HW) % C;
+    float v = ((float)x[i] - mean[ch]) * invstd[ch];
+    v = v * wgt[ch] + bias[ch];
+    if (RELU) v = v > 0.f ? v : 0.f;
+    y[i] = (T)v;
+  }
+}
+
+// per-channel [gsum, gxsum] where gxsum = sum(go * xhat); RELU masks go
+// by (y > 0) — pass y=nullptr when no fusion.
+template <typename T, bool RELU>
+__global__ void bn_bwd_stats_kernel(const T* __restrict__ go,
+                                    const T* __restrict__ x,
+                                    const T* __restrict__ y,
+                                    const float* __restrict__ mean,
+                                    const float* __restrict__ invstd,
+                                    float* __restrict__ out, int64_t N,
+                                    int64_t C, int64_t HW) {
+  for (int64_t ch = blockIdx.x; ch < C; ch += gridDim.x) {
+    const float m = mean[ch], inv = invstd[ch];
+    double gs = 0.0, gx = 0.0;
+    for (int64_t n = 0; n < N; ++n) {
+      const int64_t off = (n * C + ch) * HW;
+      for (int64_t i = threadIdx.x; i < HW; i += blockDim.x) {
+        float g = (float)go[off + i];
+        if (RELU && (float)y[off + i] <= 0.f) g = 0.f;
+        const float xh = ((float)x[off + i] - m) * inv;
+        gs += (double)g;
+        gx += (double)(g * xh);
+      }
+    }
+    for (int off2 = 32; off2 > 0; off2 >>= 1) {
+      gs += __shfl_down(gs, off2, 64);
+      gx += __shfl_down(gx, off2, 64);
+    }
+    __shared__ double l1[8], l2[8];
+    const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
+    if (lane == 0) { l1[wid] = gs; l2[wid] = gx; }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      double t1 = 0.0, t2 = 0.0;
+      for (int wv = 0; wv < (int)(blockDim.x >> 6); ++wv) { t1 += l1[wv]; t2 += l2[wv]; }
+      out[ch] = (float)t1;
+      out[C + ch] = (float)t2;
+    }
+    __syncthreads();
+  }
+}
+
+template <typename T, bool RELU>
+__global__ void bn_bwd_apply_kernel(const T* __restrict__ go,
+                                    const T* __restrict__ x,
+                                    const T* __restrict__ y,
+                                    T* __restrict__ gi,
+                                    const float* __restrict__ mean,
+                                    const float* __restrict__ invstd,
+                                    const float* __restrict__ wgt,
+                                    const float* __restrict__ gsum,
+                                    const float* __restrict__ gxsum,
+                                    double inv_n, int64_t N, int64_t C,
+                                    int64_t HW) {
+  const int64_t total = N * C * HW;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    const int64_t ch = (i / HW) % C;
+    float g = (float)go[i];
+    if (RELU && (float)y[i] <= 0.f) g = 0.f;
+    const float m = mean[ch], inv = invstd[ch];
+    const float xh = ((float)x[i] - m) * inv;
+    const float t =
+        g - (float)(gsum[ch] * inv_n) - xh * (float)(gxsum[ch] * inv_n);
+    gi[i] = (T)(t * wgt[ch] * inv);
+  }
+}
+
+// ---------------- torch-facing wrappers ----------------
+
+static inline int grid_for(int64_t total, int block) {
+  int64_t g = (total + block - 1) / block;
+  return (int)std::min<int64_t>(g, 4096);
+}
+
+std::vector<torch::Tensor> maxpool_fwd(torch::Tensor x, int64_t k, int64_t s,
+                                       int64_t p) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.is_contiguous());
+  const int64_t N = x.size(0), C = x.size(1);
+  const int H = (int)x.size(2), W = (int)x.size(3);
+  const int OH = (H + 2 * (int)p - (int)k) / (int)s + 1;
+  const int OW = (W + 2 * (int)p - (int)k) / (int)s + 1;
+  auto y = torch::empty({N, C, OH, OW}, x.options());
+  auto idx = torch::empty({N, C, OH, OW}, x.options().dtype(torch::kInt));
+  const int64_t total = N * C * (int64_t)OH * OW;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::Half, at::ScalarType::BFloat16, x.scalar_type(),
+      "maxpool_fwd", [&] {
+        hipLaunchKernelGGL(
+            (maxpool_fwd_kernel<scalar_t>), dim3(grid_for(total, 256)),
+            dim3(256), 0, stream.stream(), x.data_ptr<scalar_t>(),
+            y.data_ptr<scalar_t>(), idx.data_ptr<int32_t>(), N * C, H, W, OH,
+            OW, (int)k, (int)s, (int)p);
+      });
+  return {y, idx};
+}
+
+torch::Tensor maxpool_bwd(torch::Tensor go, torch::Tensor idx, int64_t H,
+                          int64_t W, int64_t k, int64_t s, int64_t p) {
+  TORCH_CHECK(go.is_cuda() && go.is_contiguous());
+  const int64_t N = go.size(0), C = go.size(1);
+  const int OH = (int)go.size(2), OW = (int)go.size(3);
+  auto gi = torch::empty({N, C, H, W}, go.options());
+  const int64_t total = N * C * H * W;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::Half, at::ScalarType::BFloat16, go.scalar_type(),
+      "maxpool_bwd", [&] {
+        hipLaunchKernelGGL(
+            (maxpool_bwd_kernel<scalar_t>), dim3(grid_for(total, 256)),
+            dim3(256), 0, stream.stream(), go.data_ptr<scalar_t>(),
+            idx.data_ptr<int32_t>(), gi.data_ptr<scalar_t>(), N * C, (int)H,
+            (int)W, OH, OW, (int)k, (int)s, (int)p);
+      });
+  return gi;
+}
+
+torch::Tensor avgpool_fwd(torch::Tensor x, int64_t k, int64_t s, int64_t p,
+                          int64_t gr0, int64_t gc0, int64_t Hg, int64_t Wg,
+                          bool include_pad) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.is_contiguous());
+  const int64_t N = x.size(0), C = x.size(1);
+  const int H = (int)x.size(2), W = (int)x.size(3);
+  const int OH = (H + 2 * (int)p - (int)k) / (int)s + 1;
+  const int OW = (W + 2 * (int)p - (int)k) / (int)s + 1;
+  auto y = torch::empty({N, C, OH, OW}, x.options());
+  const int64_t total = N * C * (int64_t)OH * OW;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::Half, at::ScalarType::BFloat16, x.scalar_type(),
+      "avgpool_fwd", [&] {
+        hipLaunchKernelGGL(
+            (avgpool_fwd_kernel<scalar_t>), dim3(grid_for(total, 256)),
+            dim3(256), 0, stream.stream(), x.data_ptr<scalar_t>(),
+            y.data_ptr<scalar_t>(), N * C, H, W, OH, OW, (int)k, (int)s,
+            (int)p, (long)gr0, (long)gc0, (long)Hg, (long)Wg,
+            include_pad ? 1 : 0);
+      });
+  return y;
+}
+
+torch::Tensor avgpool_bwd(torch::Tensor go, int64_t H, int64_t W, int64_t k,
+                          int64_t s, int64_t p, int64_t gr0, int64_t gc0,
+                          int64_t Hg, int64_t Wg, bool include_pad) {
+  TORCH_CHECK(go.is_cuda() && go.is_contiguous());
+  const int64_t N = go.size(0), C = go.size(1);
+  const int OH = (int)go.size(2), OW = (int)go.size(3);
+  auto gi = torch::empty({N, C, H, W}, go.options());
+  const int64_t total = N * C * H * W;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::Half, at::ScalarType::BFloat16, go.scalar_type(),
+      "avgpool_bwd", [&] {
+        hipLaunchKernelGGL(
+            (avgpool_bwd_kernel<scalar_t>), dim3(grid_for(total, 256)),
+            dim3(256), 0, stream.stream(), go.data_ptr<scalar_t>(),
+            gi.data_ptr<scalar_t>(), N * C, (int)H, (int)W, OH, OW, (int)k,
+            (int)s, (int)p, (long)gr0, (long)gc0, (long)Hg, (long)Wg,
+            include_pad ? 1 : 0);
+      });
+  return gi;
+}
+
+torch::Tensor bn_stats64(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.is_contiguous());
+  const int64_t N = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
+  auto out = torch::empty({2 * C}, x.options().dtype(torch::kFloat));
+  const int grid = (int)std::min<int64_t>(C, 2048);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::Half, at::ScalarType::BFloat16, x.scalar_type(),
+      "bn_stats64", [&] {
+        hipLaunchKernelGGL((bn_stats64_kernel<scalar_t>), dim3(grid),
+                           dim3(256), 0, stream.stream(),
+                           x.data_ptr<scalar_t>(), out.data_ptr<float>(), N, C,
+                           HW);
+      });
+  return out;
+}
+
+torch::Tensor bn_apply(torch::Tensor x, torch::Tensor mean,
+                       torch::Tensor invstd, torch::Tensor w, torch::Tensor b,
+                       bool relu) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.is_contiguous());
+  const int64_t N = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
+  auto y = torch::empty_like(x);
+  const int64_t total = N * C * HW;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::Half, at::ScalarType::BFloat16, x.scalar_type(),
+      "bn_apply", [&] {
+        if (relu)
+          hipLaunchKernelGGL((bn_apply_kernel<scalar_t, true>),
+                             dim3(grid_for(total, 256)), dim3(256), 0,
+                             stream.stream(), x.data_ptr<scalar_t>(),
+                             y.data_ptr<scalar_t>(), mean.data_ptr<float>(),
+                             invstd.data_ptr<float>(), w.data_ptr<float>(),
+                             b.data_ptr<float>(), N, C, HW);
+        else
+          hipLaunchKernelGGL((bn_apply_kernel<scalar_t, false>),
+                             dim3(grid_for(total, 256)), dim3(256), 0,
+                             stream.stream(), x.data_ptr<scalar_t>(),
+                             y.data_ptr<scalar_t>(), mean.data_ptr<float>(),
+                             invstd.data_ptr<float>(), w.data_ptr<float>(),
+                             b.data_ptr<float>(), N, C, HW);
+      });
+  return y;
+}
+
+torch::Tensor bn_bwd_stats(torch::Tensor go, torch::Tensor x, torch::Tensor y,
+                           torch::Tensor mean, torch::Tensor invstd,
+                           bool relu) {
+  const int64_t N = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
+  auto out = torch::empty({2 * C}, x.options().dtype(torch::kFloat));
+  const int grid = (int)std::min<int64_t>(C, 2048);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::Half, at::ScalarType::BFloat16, x.scalar_type(),
+      "bn_bwd_stats", [&] {
+        if (relu)
+          hipLaunchKernelGGL((bn_bwd_stats_kernel<scalar_t, true>), dim3(grid),
+                             dim3(256), 0, stream.stream(),
+                             go.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
+                             y.data_ptr<scalar_t>(), mean.data_ptr<float>(),
+                             invstd.data_ptr<float>(), out.data_ptr<float>(),
+                             N, C, HW);
+        else
+          hipLaunchKernelGGL((bn_bwd_stats_kernel<scalar_t, false>),
+                             dim3(grid), dim3(256), 0, stream.stream(),
+                             go.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
+                             (scalar_t*)nullptr, mean.data_ptr<float>(),
+                             invstd.data_ptr<float>(), out.data_ptr<float>(),
+                             N, C, HW);
+      });
+  return out;
+}
+
+torch::Tensor bn_bwd_apply(torch::Tensor go, torch::Tensor x, torch::Tensor y,
+                           torch::Tensor mean, torch::Tensor invstd,
+                           torch::Tensor w, torch::Tensor gsum,
+                           torch::Tensor gxsum, double n, bool relu) {
+  const int64_t N = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
+  auto gi = torch::empty_like(x);
+  const int64_t total = N * C * HW;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::Half, at::ScalarType::BFloat16, x.scalar_type(),
+      "bn_bwd_apply", [&] {
+        if (relu)
+          hipLaunchKernelGGL(
+              (bn_bwd_apply_kernel<scalar_t, true>),
+              dim3(grid_for(total, 256)), dim3(256), 0, stream.stream(),
+              go.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
+              y.data_ptr<scalar_t>(), gi.data_ptr<scalar_t>(),
+              mean.data_ptr<float>(), invstd.data_ptr<float>(),
+              w.data_ptr<float>(), gsum.data_ptr<float>(),
+              gxsum.data_ptr<float>(), 1.0 / n, N, C, HW);
+        else
+          hipLaunchKernelGGL(
+              (bn_bwd_apply_kernel<scalar_t, false>),
+              dim3(grid_for(total, 256)), dim3(256), 0, stream.stream(),
+              go.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
+              (scalar_t*)nullptr, gi.data_ptr<scalar_t>(),
+              mean.data_ptr<float>(), invstd.data_ptr<float>(),
+              w.data_ptr<float>(), gsum.data_ptr<float>(),
+              gxsum.data_ptr<float>(), 1.0 / n, N, C, HW);
+      });
+  return gi;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -208,5 +682,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("halo_unpack", &halo_copy<1>, "unpack halo strips buffer->tile");
   m.def("halo_unpack_add", &halo_copy<2>, "accumulate grad strips into tile");
   m.def("bn_stats", &bn_stats, "per-channel [sum, sumsq] in one pass");
+  m.def("maxpool_fwd", &maxpool_fwd);
+  m.def("maxpool_bwd", &maxpool_bwd);
+  m.def("avgpool_fwd", &avgpool_fwd);
+  m.def("avgpool_bwd", &avgpool_bwd);
+  m.def("bn_stats64", &bn_stats64);
+  m.def("bn_apply", &bn_apply);
+  m.def("bn_bwd_stats", &bn_bwd_stats);
+  m.def("bn_bwd_apply", &bn_bwd_apply);
   m.attr("gfx_arch") = "gfx950";
 }

@@ -114,18 +114,17 @@ def make_op(name: str, c: int, stride: int, ctx, mknorm) -> nn.Module:
             nn.Identity() if stride == 1 else FactorizedReduce(c, c, mknorm)
         )
     if name == "avg_pool_3x3":
-        if ctx is None:
-            return nn.AvgPool2d(3, stride=stride, padding=1, count_include_pad=False)
+        # plain path also routes through HaloPool2d (num_spatial_parts=1)
+        # so the gemscore pool kernels run on GPU
         return HaloPool2d(
-            "avg", 3, stride=stride, padding=1, count_include_pad=False, **ctx
+            "avg", 3, stride=stride, padding=1, count_include_pad=False,
+            **(ctx or {})
         )
     if name == "max_pool_3x3":
-        if ctx is None:
-            return nn.MaxPool2d(3, stride=stride, padding=1)
-        return HaloPool2d("max", 3, stride=stride, padding=1, **ctx)
+        return HaloPool2d("max", 3, stride=stride, padding=1, **(ctx or {}))
     if name == "max_pool_2x2":
         # padding 0, stride 2 in reduction cells: tile-local, no halo
-        return nn.MaxPool2d(2, stride=stride, padding=0)
+        return HaloPool2d("max", 2, stride=stride, padding=0)
     if name == "conv_1x1":
         return nn.Sequential(
             nn.ReLU(inplace=False),
@@ -269,7 +268,11 @@ def amoebanetd(
 
     def mknorm():
         if plan is None:
-            return nn.BatchNorm2d
+            # TileBatchNorm2d(group=None) == BatchNorm2d numerically but
+            # takes the gemscore fused path on GPU
+            from ..ops.norm import TileBatchNorm2d
+
+            return TileBatchNorm2d
         i = len(cells)
         return lambda ch: plan.norm(ch, i)
 

@@ -87,9 +87,7 @@ class StemS(nn.Module):
                 sconv(in_ch, filters, 7, 2, ctx),
                 mknorm(filters),
                 nn.ReLU(inplace=True),
-                HaloPool2d("max", 3, stride=2, padding=1, **ctx)
-                if ctx is not None
-                else nn.MaxPool2d(3, stride=2, padding=1),
+                HaloPool2d("max", 3, stride=2, padding=1, **(ctx or {})),
             )
         else:
             self.ops = nn.Sequential(

@@ -51,9 +51,10 @@ def allreduce_sum(t: torch.Tensor, group) -> torch.Tensor:
 class TileBatchNorm2d(nn.BatchNorm2d):
     """Drop-in BatchNorm2d whose batch statistics span the tile group."""
 
-    def __init__(self, num_features: int, group=None, **kw):
+    def __init__(self, num_features: int, group=None, relu: bool = False, **kw):
         super().__init__(num_features, **kw)
         self.group = group
+        self.relu = relu  # fused BN+ReLU (native GPU path only)
 
     def _use_sync(self, x) -> bool:
         return (
@@ -64,8 +65,11 @@ class TileBatchNorm2d(nn.BatchNorm2d):
         )
 
     def forward(self, x):
+        if x.is_cuda and not x.is_meta:
+            return self._forward_native(x)
         if not self._use_sync(x):
-            return super().forward(x)
+            y = super().forward(x)
+            return torch.relu_(y) if self.relu else y
         n_local = x.numel() // x.shape[1]
         world = dist.get_world_size(group=self.group)
         n = n_local * world  # tiles are equal-sized (power-of-two constraint)
@@ -86,4 +90,38 @@ class TileBatchNorm2d(nn.BatchNorm2d):
         out = centred * inv.view(1, -1, 1, 1)
         if self.affine:
             out = out * self.weight.view(1, -1, 1, 1) + self.bias.view(1, -1, 1, 1)
-        return out
+        return torch.relu(out) if self.relu else out
+
+    # -- gemscore fused path (MI355X) ----------------------------------
+
+    def _forward_native(self, x):
+        from . import backend
+        from .native import native_batchnorm
+
+        ge = backend.ext()
+        x = x.contiguous()
+        C = self.num_features
+        n = x.numel() // C
+        group = self.group if (self.group is not None and dist.is_initialized()) else None
+        if self.training:
+            stats = ge.bn_stats64(x)
+            if group is not None:
+                dist.all_reduce(stats, group=group)
+                n *= dist.get_world_size(group=group)
+            mean = stats[:C] / n
+            var = stats[C:] / n - mean * mean
+            if self.track_running_stats:
+                with torch.no_grad():
+                    m = self.momentum if self.momentum is not None else 0.1
+                    self.running_mean.mul_(1 - m).add_(mean, alpha=m)
+                    unbiased = var * (n / max(n - 1, 1))
+                    self.running_var.mul_(1 - m).add_(unbiased, alpha=m)
+                    self.num_batches_tracked += 1
+            invstd = torch.rsqrt(var + self.eps)
+        else:
+            mean = self.running_mean.float()
+            invstd = torch.rsqrt(self.running_var.float() + self.eps)
+        return native_batchnorm(
+            x, self.weight, self.bias, mean.contiguous(), invstd.contiguous(),
+            group, float(n), self.training, self.relu,
+        )

@@ -206,10 +206,30 @@ class HaloPool2d(_SpatialBase):
 
     def forward(self, x):
         h = self.halo_len
+        k, s = self.kernel_size, self.stride
+        on_gpu = x.is_cuda and not x.is_meta
         # max pool pads with -inf so image-boundary windows match the
         # single-GPU op exactly (zero-pad would win over negative inputs)
         fill = float("-inf") if self.kind == "max" else 0.0
         H_loc, W_loc = x.shape[-2], x.shape[-1]
+
+        # ---- plain (no tiles): padding handled inside the op ------------
+        if self.exchanger is None and not self.d2:
+            if self.kind == "max":
+                if on_gpu:
+                    from .native import native_maxpool
+
+                    return native_maxpool(x, k, s, h)
+                return F.max_pool2d(x, k, s, padding=h)
+            if on_gpu:
+                from .native import native_avgpool
+
+                return native_avgpool(x, k, s, h, include_pad=self.count_include_pad)
+            return F.avg_pool2d(
+                x, k, s, padding=h, count_include_pad=self.count_include_pad
+            )
+
+        # ---- tiled: halo-pad then pool with padding=0 -------------------
         if self.d2 and h > 0:
             xp = (
                 outer_pad_only(x, self.layout, self.tile, h, fill=fill)
@@ -223,17 +243,27 @@ class HaloPool2d(_SpatialBase):
                 else x
             )
         if self.kind == "max":
-            return F.max_pool2d(xp, self.kernel_size, self.stride, padding=0)
-        if self.count_include_pad or h == 0 or self.exchanger is None:
-            return F.avg_pool2d(
-                xp, self.kernel_size, self.stride, padding=0, count_include_pad=True
+            if on_gpu and not self.d2:
+                from .native import native_maxpool
+
+                return native_maxpool(xp, k, s, 0)
+            return F.max_pool2d(xp, k, s, padding=0)
+        # avg: divisors from GLOBAL geometry when count_include_pad=False
+        r, c = self.layout.pos(self.tile)
+        Hg, Wg = H_loc * self.layout.rows, W_loc * self.layout.cols
+        gr0, gc0 = r * H_loc - h, c * W_loc - h
+        if on_gpu and not self.d2:
+            from .native import native_avgpool
+
+            return native_avgpool(
+                xp, k, s, 0, gr0=gr0, gc0=gc0, Hg=Hg, Wg=Wg,
+                include_pad=self.count_include_pad,
             )
-        if x.is_meta:
-            return F.avg_pool2d(xp, self.kernel_size, self.stride, padding=0)
-        k = self.kernel_size
-        sums = F.avg_pool2d(
-            xp, k, self.stride, padding=0, count_include_pad=True
-        ) * float(k * k)
+        if self.count_include_pad or h == 0 or x.is_meta:
+            return F.avg_pool2d(xp, k, s, padding=0, count_include_pad=True)
+        sums = F.avg_pool2d(xp, k, s, padding=0, count_include_pad=True) * float(
+            k * k
+        )
         div = self._avg_divisors(
             sums.shape[-2], sums.shape[-1], H_loc, W_loc, sums.device
         )

@@ -126,3 +126,90 @@ def test_amoebanet_gpu_train_step():
     loss.backward()
     torch.cuda.synchronize()
     assert torch.isfinite(loss)
+
+
+@gpu
+@requires_gpu
+@pytest.mark.parametrize("k,s,p", [(3, 2, 1), (2, 2, 0), (3, 1, 1)])
+def test_native_maxpool_matches_torch(k, s, p):
+    from mpi4dl_amd.ops.native import native_maxpool
+
+    torch.manual_seed(0)
+    x = torch.randn(2, 7, 33, 47, device="cuda", requires_grad=True)
+    x2 = x.detach().clone().requires_grad_(True)
+    y = native_maxpool(x, k, s, p)
+    y_ref = torch.nn.functional.max_pool2d(x2, k, s, padding=p)
+    assert torch.allclose(y, y_ref, atol=1e-6)
+    g = torch.randn_like(y)
+    y.backward(g)
+    y_ref.backward(g)
+    assert torch.allclose(x.grad, x2.grad, atol=1e-5)
+
+
+@gpu
+@requires_gpu
+@pytest.mark.parametrize("include_pad", [True, False])
+def test_native_avgpool_matches_torch(include_pad):
+    from mpi4dl_amd.ops.native import native_avgpool
+
+    torch.manual_seed(0)
+    x = torch.randn(2, 5, 32, 48, device="cuda", requires_grad=True)
+    x2 = x.detach().clone().requires_grad_(True)
+    y = native_avgpool(x, 3, 2, 1, include_pad=include_pad)
+    y_ref = torch.nn.functional.avg_pool2d(
+        x2, 3, 2, padding=1, count_include_pad=include_pad
+    )
+    assert torch.allclose(y, y_ref, atol=1e-5), (y - y_ref).abs().max()
+    g = torch.randn_like(y)
+    y.backward(g)
+    y_ref.backward(g)
+    assert torch.allclose(x.grad, x2.grad, atol=1e-5)
+
+
+@gpu
+@requires_gpu
+def test_native_batchnorm_matches_torch():
+    from mpi4dl_amd.ops.norm import TileBatchNorm2d
+
+    torch.manual_seed(0)
+    bn = TileBatchNorm2d(9).cuda()
+    ref = torch.nn.BatchNorm2d(9).cuda()
+    ref.load_state_dict({k: v for k, v in bn.state_dict().items()})
+    x = torch.randn(3, 9, 17, 21, device="cuda", requires_grad=True)
+    x2 = x.detach().clone().requires_grad_(True)
+    y = bn(x)
+    y_ref = ref(x2)
+    assert torch.allclose(y, y_ref, atol=1e-5), (y - y_ref).abs().max()
+    g = torch.randn_like(y)
+    y.backward(g)
+    y_ref.backward(g)
+    assert torch.allclose(x.grad, x2.grad, atol=1e-4), (x.grad - x2.grad).abs().max()
+    assert torch.allclose(bn.weight.grad, ref.weight.grad, atol=1e-3)
+    assert torch.allclose(bn.bias.grad, ref.bias.grad, atol=1e-3)
+    assert torch.allclose(bn.running_mean, ref.running_mean, atol=1e-5)
+    assert torch.allclose(bn.running_var, ref.running_var, atol=1e-4)
+    # eval path
+    bn.eval(); ref.eval()
+    ye = bn(x.detach())
+    ye_ref = ref(x.detach())
+    assert torch.allclose(ye, ye_ref, atol=1e-5)
+
+
+@gpu
+@requires_gpu
+def test_native_batchnorm_relu_fused():
+    from mpi4dl_amd.ops.norm import TileBatchNorm2d
+
+    torch.manual_seed(0)
+    bn = TileBatchNorm2d(4, relu=True).cuda()
+    ref = torch.nn.BatchNorm2d(4).cuda()
+    ref.load_state_dict({k: v for k, v in bn.state_dict().items()})
+    x = torch.randn(2, 4, 8, 8, device="cuda", requires_grad=True)
+    x2 = x.detach().clone().requires_grad_(True)
+    y = bn(x)
+    y_ref = torch.relu(ref(x2))
+    assert torch.allclose(y, y_ref, atol=1e-5)
+    g = torch.randn_like(y)
+    y.backward(g)
+    y_ref.backward(g)
+    assert torch.allclose(x.grad, x2.grad, atol=1e-4)
