@@ -133,7 +133,12 @@ def main():
     gen.ready_model(comm.split_rank, device=device)
     del model  # free remote stages
 
-    opt = torch.optim.SGD(gen.models.parameters(), lr=0.01, momentum=0.9)
+    if on_gpu:
+        from mpi4dl_amd.optim import FusedSGD
+
+        opt = FusedSGD(gen.models, lr=0.01, momentum=0.9)
+    else:
+        opt = torch.optim.SGD(gen.models.parameters(), lr=0.01, momentum=0.9)
     eng_kw = dict(
         optimizer=opt,
         device=device,

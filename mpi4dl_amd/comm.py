@@ -393,7 +393,12 @@ class FlatGrads:
     re-concatenates all grads every step — comm.py:414-438).
     """
 
-    def __init__(self, module: torch.nn.Module, dtype: Optional[torch.dtype] = None):
+    def __init__(
+        self,
+        module: torch.nn.Module,
+        dtype: Optional[torch.dtype] = None,
+        pad_to: int = 0,
+    ):
         self.params = [p for p in module.parameters() if p.requires_grad]
         if not self.params:
             self.buffer = torch.zeros(0)
@@ -401,6 +406,7 @@ class FlatGrads:
         dev = self.params[0].device
         dtype = dtype or self.params[0].dtype
         total = sum(p.numel() for p in self.params)
+        total = max((total + 3) // 4 * 4, pad_to)  # float4 kernels
         self.buffer = torch.zeros(total, device=dev, dtype=dtype)
         offset = 0
         for p in self.params:
@@ -418,6 +424,16 @@ class FlatGrads:
     def rescale_(self, divide_by: float):
         if divide_by != 1.0 and self.buffer.numel():
             self.buffer.div_(divide_by)
+
+    @classmethod
+    def get(cls, module: torch.nn.Module, pad_to: int = 0) -> "FlatGrads":
+        """One FlatGrads per module (shared by GradReducer and FusedSGD
+        so collectives and the optimizer use the SAME buffer)."""
+        fg = getattr(module, "_mpi4dl_flatgrads", None)
+        if fg is None or (pad_to and fg.buffer.numel() < pad_to):
+            fg = cls(module, pad_to=pad_to)
+            module._mpi4dl_flatgrads = fg
+        return fg
 
 
 class GradReducer:
@@ -471,10 +487,7 @@ class GradReducer:
     # -- grads ---------------------------------------------------------------
 
     def flat(self, module: torch.nn.Module) -> FlatGrads:
-        key = id(module)
-        if key not in self._flat:
-            self._flat[key] = FlatGrads(module)
-        return self._flat[key]
+        return FlatGrads.get(module)
 
     def allreduce_grads(self, module: torch.nn.Module, group, divide_by: float = None):
         """Average grads over ``group`` (no-op for group=None / size 1)."""

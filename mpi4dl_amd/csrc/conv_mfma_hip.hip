@@ -1,0 +1,452 @@
+// Implicit-GEMM MFMA convolution for gfx950 (CDNA4) — forward,
+// backward-weight; backward-data reuses the forward kernel on
+// transformed weights (stride 1).
+//
+// This is the hand-written compute path behind HaloConv2d: conv on
+// halo-pre-padded NCHW bf16 tiles (padding usually 0 after the
+// exchange; plain padding also supported). Design notes
+// (/opt/skills/guides/cdna_hip_programming.md):
+//
+// * GEMM view: OUT[Kout][N*OH*OW] = W[Kout][C*R*S] x X[C*R*S][N*OH*OW].
+//   With NCHW, pixels are contiguous for fixed (c,r,s), so the X
+//   ("B") operand loads coalesce; weights are (c,r,s)-contiguous, so
+//   the W ("A") operand is directly fragment-shaped.
+// * N-tiles are OUTPUT-ROW SEGMENTS (one (n,oh) row per block), so the
+//   per-k input address is base + c*HW + r*W + s and never wraps —
+//   no per-pixel div/mod in the hot loop.
+// * mfma_f32_16x16x32_bf16; block = 256 threads = 4 waves (2x2), tile
+//   BM=128 (out-ch) x BN=128 (pixels) x BK=32; wave tile 64x64,
+//   acc 4x4 fragments (64 f32/lane).
+// * A (weights) LDS image [BM][BK] row-major -> ds_read_b128 fragments.
+//   B (pixels) LDS image stores element (k, px) of each 16-px block at
+//   off = px + (k&3)*16 + (k>>3)*64 + ((k>>2)&1)*256 (bf16 units), so
+//   ONE ds_read_b64_tr_b16 pair per fragment delivers the full k=32
+//   per-lane B fragment (the attention-V recipe, guide S2 tr-read).
+// * fp32 accumulate; epilogue adds bias and stores bf16.
+//
+// Reference behaviour being replaced: cuDNN/MIOpen conv under
+// torchgems' conv_spatial (/root/reference/src/torchgems/spatial.py:1027).
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include <cstdint>
+
+namespace conv_mfma {
+
+using bf16 = __hip_bfloat16;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(4))) short s16x4;
+typedef __attribute__((ext_vector_type(8))) short s16x8;
+
+#define BM 128
+#define BN 128
+#define BK 32
+#define WAVES 4
+
+struct ConvGeom {
+  int N, C, H, W;        // input
+  int K, R, S;           // weights
+  int OH, OW;            // output
+  int sh, sw, ph, pw;    // stride / pad
+  int CRS;               // C*R*S
+  int row_tiles;         // ceil(OW / BN)
+};
+
+// ---------------------------------------------------------------------------
+// Forward kernel
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void conv_fwd_kernel(
+    const bf16* __restrict__ x, const bf16* __restrict__ w,
+    const float* __restrict__ bias, bf16* __restrict__ out, ConvGeom g) {
+  // block -> (m_tile, n, oh, ow_tile)
+  const int m_tiles = (g.K + BM - 1) / BM;
+  int bid = blockIdx.x;
+  const int mt = bid % m_tiles;
+  bid /= m_tiles;
+  const int owt = bid % g.row_tiles;
+  bid /= g.row_tiles;
+  const int oh = bid % g.OH;
+  const int n = bid / g.OH;
+
+  const int k0out = mt * BM;           // first out-channel of tile
+  const int ow0 = owt * BN;            // first output col of tile
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;            // 4 waves: (wm, wn) = (wid>>1, wid&1)
+  const int wm = wid >> 1, wn = wid & 1;
+
+  // LDS: A image [BM][BK] bf16 (+8 pad per row vs bank conflicts),
+  //      B image (BN/16) blocks x 512 elems, double buffered.
+  __shared__ __attribute__((aligned(16))) short lds[2 * (BM * (BK + 8)) +
+                                                    2 * (BN * BK)];
+  auto ldsA = [&](int buf) { return lds + buf * (BM * (BK + 8)); };
+  auto ldsB = [&](int buf) {
+    return lds + 2 * (BM * (BK + 8)) + buf * (BN * BK);
+  };
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  const int64_t HW = (int64_t)g.H * g.W;
+  const int64_t in_n = (int64_t)n * g.C * HW;
+  const int RS = g.R * g.S;
+
+  // ---- staging helpers ----------------------------------------------------
+  // A: 256 threads load BM*BK = 4096 bf16 in 2 passes of dwordx4 (8 bf16).
+  // thread -> (row = idx/4, kchunk = idx%4)
+  // B: 512 (k, px8) chunks -> 2 per thread.
+
+  auto stage = [&](int buf, int kk0) {
+    // ---- A (weights) ----
+#pragma unroll
+    for (int pass = 0; pass < 2; ++pass) {
+      const int idx = pass * 256 + tid;       // 0..511
+      const int row = idx >> 2;               // 0..127
+      const int kc = (idx & 3) * 8;           // 0,8,16,24
+      const int kout = k0out + row;
+      short v[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (kout < g.K) {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const int k = kk0 + kc + e;
+          if (k < g.CRS) v[e] = ((const short*)w)[(int64_t)kout * g.CRS + k];
+        }
+      }
+      short* dst = ldsA(buf) + row * (BK + 8) + kc;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) dst[e] = v[e];
+    }
+    // ---- B (input pixels) ----
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      const int idx = it * 256 + tid;         // 0..511
+      const int kk = idx & 31;                // k within tile
+      const int pxc = idx >> 5;               // 16 chunks of 8 px
+      const int px0 = pxc * 8;
+      const int k = kk0 + kk;
+      short v[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (k < g.CRS) {
+        const int c = k / RS;
+        const int rs = k - c * RS;
+        const int r = rs / g.S;
+        const int s = rs - r * g.S;
+        const int ih = oh * g.sh - g.ph + r;
+        if (ih >= 0 && ih < g.H) {
+          const bf16* src = x + in_n + c * HW + (int64_t)ih * g.W;
+#pragma unroll
+          for (int e = 0; e < 8; ++e) {
+            const int ow = ow0 + px0 + e;
+            const int iw = ow * g.sw - g.pw + s;
+            if (ow < g.OW && iw >= 0 && iw < g.W)
+              v[e] = ((const short*)src)[iw];
+          }
+        }
+      }
+      // B image offset: block pb = px/16; within block:
+      // off = (px&15) + (k&3)*16 + (k>>3)*64 + ((k>>2)&1)*256
+      const int pb = (px0 >> 4);
+      const int base = pb * 512 + (kk & 3) * 16 + ((kk >> 3) << 6) +
+                       (((kk >> 2) & 1) << 8);
+      short* dst = ldsB(buf) + base + (px0 & 15);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) dst[e] = v[e];
+    }
+  };
+
+  stage(0, 0);
+  __syncthreads();
+
+  const int a_row0 = wm * 64;     // my wave's 64 out-channels
+  const int b_px0 = wn * 64;      // my wave's 64 pixels
+
+  for (int kk0 = 0; kk0 < g.CRS; kk0 += BK) {
+    const int buf = (kk0 / BK) & 1;
+    if (kk0 + BK < g.CRS) {
+      stage(buf ^ 1, kk0 + BK);
+    }
+    // fragments + MFMA
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf) {
+      const short* arow =
+          ldsA(buf) + (a_row0 + mf * 16 + (lane & 15)) * (BK + 8) +
+          ((lane >> 4) << 3);
+      s16x8 afrag = *(const s16x8*)arow;
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        const int pb = (b_px0 >> 4) + nf;
+        const short* bbase = ldsB(buf) + pb * 512;
+        // tr reads: lane-dependent addressing handled by the instruction
+        s16x4 b0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (__attribute__((address_space(3))) s16x4*)(
+                (__attribute__((address_space(3))) short*)(bbase) +
+                (lane & 15) + ((lane >> 4) << 6)));
+        s16x4 b1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (__attribute__((address_space(3))) s16x4*)(
+                (__attribute__((address_space(3))) short*)(bbase + 256) +
+                (lane & 15) + ((lane >> 4) << 6)));
+        s16x8 bfrag;
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          bfrag[e] = b0[e];
+          bfrag[e + 4] = b1[e];
+        }
+        acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag, bfrag, acc[mf][nf], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: bias + bf16 store ---------------------------------------
+  const int64_t out_n = ((int64_t)n * g.K) * g.OH * g.OW;
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int kout = k0out + a_row0 + mf * 16 + ((lane >> 4) << 2) + reg;
+      if (kout >= g.K) continue;
+      const float b = bias ? bias[kout] : 0.f;
+      const int64_t orow = out_n + (int64_t)kout * g.OH * g.OW +
+                           (int64_t)oh * g.OW;
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        const int ow = ow0 + b_px0 + nf * 16 + (lane & 15);
+        if (ow < g.OW)
+          out[orow + ow] = (bf16)(acc[mf][nf][reg] + b);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Backward-weight kernel: GW[Kout][CRS] += GO[Kout][px] * X[CRS][px]^T
+// Both operands are pixel-contiguous rows -> plain b128 fragments.
+// Tile 64x64, BKpx = 64 pixels per step, one (n, oh-row-chunk) per
+// blockIdx.y slice, fp32 atomicAdd into the fp32 workspace.
+// ---------------------------------------------------------------------------
+
+#define WBM 64
+#define WBN 64
+#define WBK 64
+
+__global__ __launch_bounds__(256) void conv_bwdw_kernel(
+    const bf16* __restrict__ go, const bf16* __restrict__ x,
+    float* __restrict__ gw, ConvGeom g) {
+  const int m_tiles = (g.K + WBM - 1) / WBM;
+  const int n_tiles = (g.CRS + WBN - 1) / WBN;
+  int bid = blockIdx.x;
+  const int mt = bid % m_tiles;
+  bid /= m_tiles;
+  const int nt = bid % n_tiles;
+  const int slice = bid / n_tiles;  // (n, oh) slice index
+  const int n = slice / g.OH;
+  const int oh = slice % g.OH;
+
+  const int kout0 = mt * WBM;
+  const int crs0 = nt * WBN;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;  // 4 waves (2x2): wave tile 32x32? -> use 2x2 of 16
+  const int wm = wid >> 1, wn = wid & 1;
+
+  __shared__ __attribute__((aligned(16))) short lds[WBM * (WBK + 8) +
+                                                    WBN * (WBK + 8)];
+  short* ldsA = lds;                       // GO [kout][px]
+  short* ldsB = lds + WBM * (WBK + 8);     // X  [crs][px]
+
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  const int64_t HW = (int64_t)g.H * g.W;
+  const int64_t in_n = (int64_t)n * g.C * HW;
+  const int64_t go_row = ((int64_t)n * g.K) * g.OH * g.OW +
+                         (int64_t)oh * g.OW;
+  const int RS = g.R * g.S;
+
+  for (int px0 = 0; px0 < g.OW; px0 += WBK) {
+    // stage GO tile [WBM][WBK]: idx -> (row, px8)
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      const int idx = it * 256 + tid;  // 512 chunks of 8
+      const int row = idx >> 3;
+      const int pc = (idx & 7) * 8;
+      short v[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+      const int kout = kout0 + row;
+      if (kout < g.K) {
+        const bf16* src = go + go_row + (int64_t)kout * g.OH * g.OW;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const int ow = px0 + pc + e;
+          if (ow < g.OW) v[e] = ((const short*)src)[ow];
+        }
+      }
+      short* dst = ldsA + row * (WBK + 8) + pc;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) dst[e] = v[e];
+    }
+    // stage X tile [WBN][WBK]
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      const int idx = it * 256 + tid;
+      const int row = idx >> 3;
+      const int pc = (idx & 7) * 8;
+      short v[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+      const int k = crs0 + row;
+      if (k < g.CRS) {
+        const int c = k / RS;
+        const int rs = k - c * RS;
+        const int r = rs / g.S;
+        const int s = rs - r * g.S;
+        const int ih = oh * g.sh - g.ph + r;
+        if (ih >= 0 && ih < g.H) {
+          const bf16* src = x + in_n + c * HW + (int64_t)ih * g.W;
+#pragma unroll
+          for (int e = 0; e < 8; ++e) {
+            const int ow = px0 + pc + e;
+            const int iw = ow * g.sw - g.pw + s;
+            if (ow < g.OW && iw >= 0 && iw < g.W)
+              v[e] = ((const short*)src)[iw];
+          }
+        }
+      }
+      short* dst = ldsB + row * (WBK + 8) + pc;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) dst[e] = v[e];
+    }
+    __syncthreads();
+
+    // wave (wm, wn) computes 32x32: 2x2 fragments of 16x16, k = WBK
+#pragma unroll
+    for (int kk = 0; kk < WBK; kk += 32) {
+#pragma unroll
+      for (int mf = 0; mf < 2; ++mf) {
+        const short* arow = ldsA + (wm * 32 + mf * 16 + (lane & 15)) *
+                                       (WBK + 8) +
+                            kk + ((lane >> 4) << 3);
+        s16x8 afrag = *(const s16x8*)arow;
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf) {
+          const short* brow = ldsB + (wn * 32 + nf * 16 + (lane & 15)) *
+                                         (WBK + 8) +
+                              kk + ((lane >> 4) << 3);
+          s16x8 bfrag = *(const s16x8*)brow;
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag, bfrag, acc[mf][nf], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  // accumulate into gw: D[m][n] position: col = lane&15, row = (lane>>4)*4+reg
+#pragma unroll
+  for (int mf = 0; mf < 2; ++mf) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int kout = kout0 + wm * 32 + mf * 16 + ((lane >> 4) << 2) + reg;
+      if (kout >= g.K) continue;
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf) {
+        const int crs = crs0 + wn * 32 + nf * 16 + (lane & 15);
+        if (crs < g.CRS)
+          atomicAdd(&gw[(int64_t)kout * g.CRS + crs], acc[mf][nf][reg]);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Host wrappers
+// ---------------------------------------------------------------------------
+
+static ConvGeom make_geom(const torch::Tensor& x, const torch::Tensor& w,
+                          int64_t sh, int64_t sw, int64_t ph, int64_t pw) {
+  ConvGeom g;
+  g.N = (int)x.size(0);
+  g.C = (int)x.size(1);
+  g.H = (int)x.size(2);
+  g.W = (int)x.size(3);
+  g.K = (int)w.size(0);
+  g.R = (int)w.size(2);
+  g.S = (int)w.size(3);
+  g.sh = (int)sh; g.sw = (int)sw; g.ph = (int)ph; g.pw = (int)pw;
+  g.OH = (g.H + 2 * g.ph - g.R) / g.sh + 1;
+  g.OW = (g.W + 2 * g.pw - g.S) / g.sw + 1;
+  g.CRS = g.C * g.R * g.S;
+  g.row_tiles = (g.OW + BN - 1) / BN;
+  return g;
+}
+
+torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor w,
+                       c10::optional<torch::Tensor> bias, int64_t sh,
+                       int64_t sw, int64_t ph, int64_t pw) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && w.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 &&
+              w.scalar_type() == torch::kBFloat16,
+              "conv_fwd: bf16 only");
+  TORCH_CHECK(x.size(1) == w.size(1), "channel mismatch");
+  ConvGeom g = make_geom(x, w, sh, sw, ph, pw);
+  auto out = torch::empty({g.N, g.K, g.OH, g.OW},
+                          x.options().dtype(torch::kBFloat16));
+  const float* bptr = nullptr;
+  torch::Tensor b32;
+  if (bias.has_value()) {
+    b32 = bias->to(torch::kFloat).contiguous();
+    bptr = b32.data_ptr<float>();
+  }
+  const int m_tiles = (g.K + BM - 1) / BM;
+  const int64_t blocks = (int64_t)m_tiles * g.row_tiles * g.OH * g.N;
+  TORCH_CHECK(blocks < (1LL << 31), "grid too large");
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  hipLaunchKernelGGL(conv_fwd_kernel, dim3((uint32_t)blocks), dim3(256), 0,
+                     stream.stream(), (const bf16*)x.data_ptr(),
+                     (const bf16*)w.data_ptr(), bptr, (bf16*)out.data_ptr(),
+                     g);
+  return out;
+}
+
+torch::Tensor conv_bwd_weight(torch::Tensor go, torch::Tensor x,
+                              int64_t R, int64_t S, int64_t sh, int64_t sw,
+                              int64_t ph, int64_t pw) {
+  TORCH_CHECK(go.is_cuda() && go.is_contiguous() && x.is_contiguous());
+  TORCH_CHECK(go.scalar_type() == torch::kBFloat16 &&
+              x.scalar_type() == torch::kBFloat16);
+  const int K = (int)go.size(1), C = (int)x.size(1);
+  ConvGeom g;
+  g.N = (int)x.size(0); g.C = C; g.H = (int)x.size(2); g.W = (int)x.size(3);
+  g.K = K; g.R = (int)R; g.S = (int)S;
+  g.sh = (int)sh; g.sw = (int)sw; g.ph = (int)ph; g.pw = (int)pw;
+  g.OH = (int)go.size(2); g.OW = (int)go.size(3);
+  g.CRS = C * (int)R * (int)S;
+  g.row_tiles = 0;
+  auto gw = torch::zeros({(int64_t)K, (int64_t)g.CRS},
+                         x.options().dtype(torch::kFloat));
+  const int m_tiles = (K + WBM - 1) / WBM;
+  const int n_tiles = (g.CRS + WBN - 1) / WBN;
+  const int64_t blocks = (int64_t)m_tiles * n_tiles * g.N * g.OH;
+  TORCH_CHECK(blocks < (1LL << 31), "grid too large");
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  hipLaunchKernelGGL(conv_bwdw_kernel, dim3((uint32_t)blocks), dim3(256), 0,
+                     stream.stream(), (const bf16*)go.data_ptr(),
+                     (const bf16*)x.data_ptr(), gw.data_ptr<float>(), g);
+  return gw.view({(int64_t)K, (int64_t)C, R, S});
+}
+
+}  // namespace conv_mfma
+
+void register_conv_mfma(pybind11::module_& m) {
+  m.def("conv_fwd", &conv_mfma::conv_fwd,
+        "implicit-GEMM MFMA conv forward (bf16 NCHW)");
+  m.def("conv_bwd_weight", &conv_mfma::conv_bwd_weight,
+        "implicit-GEMM MFMA conv weight gradient (fp32 out)");
+}

@@ -216,65 +216,92 @@ torch::Tensor bn_stats(torch::Tensor x) {
 // with gr0 = tile_row_offset - p; plain case: gr0 = -p, Hg = H.
 // ---------------------------------------------------------------------------
 
+// 8 consecutive outputs per thread: the input row segment is read once
+// into a register sliding window (guide G13 — vectorize ALWAYS).
 template <typename T>
 __global__ void maxpool_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
                                    int32_t* __restrict__ idx, int64_t NC,
                                    int H, int W, int OH, int OW, int k, int s,
                                    int p) {
-  const int64_t total = NC * OH * OW;
+  const int OW8 = (OW + 7) / 8;
+  const int64_t total = NC * OH * OW8;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += stride) {
-    const int ow = (int)(i % OW);
-    const int oh = (int)((i / OW) % OH);
-    const int64_t plane = i / ((int64_t)OW * OH);
+  for (int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; t < total;
+       t += stride) {
+    const int owc = (int)(t % OW8);
+    const int oh = (int)((t / OW8) % OH);
+    const int64_t plane = t / ((int64_t)OW8 * OH);
     const T* xp = x + plane * H * W;
-    const int h0 = oh * s - p, w0 = ow * s - p;
-    float best = -INFINITY;
-    int32_t bidx = 0;
+    const int ow0 = owc * 8;
+    const int h0 = oh * s - p;
+    float best[8];
+    int32_t bidx[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) { best[e] = -INFINITY; bidx[e] = 0; }
     for (int kh = 0; kh < k; ++kh) {
       const int h = h0 + kh;
       if (h < 0 || h >= H) continue;
-      for (int kw = 0; kw < k; ++kw) {
-        const int w = w0 + kw;
-        if (w < 0 || w >= W) continue;
-        const float v = (float)xp[h * W + w];
-        if (v > best) { best = v; bidx = h * W + w; }
+      const T* row = xp + h * W;
+      // window of this row needed: [ow0*s-p, (ow0+7)*s-p+k)
+      const int w_lo = ow0 * s - p;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const int wbase = w_lo + e * s;
+        for (int kw = 0; kw < k; ++kw) {
+          const int w = wbase + kw;
+          if (w < 0 || w >= W) continue;
+          const float v = (float)row[w];
+          if (v > best[e]) { best[e] = v; bidx[e] = h * W + w; }
+        }
       }
     }
-    y[i] = (T)best;
-    idx[i] = bidx;
+    const int64_t obase = plane * (int64_t)OH * OW + (int64_t)oh * OW;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const int ow = ow0 + e;
+      if (ow < OW) { y[obase + ow] = (T)best[e]; idx[obase + ow] = bidx[e]; }
+    }
   }
 }
 
+// 8 consecutive input columns per thread: the candidate output windows
+// of the 8 inputs overlap, so go/idx rows are read once per (oh, thread).
 template <typename T>
 __global__ void maxpool_bwd_kernel(const T* __restrict__ go,
                                    const int32_t* __restrict__ idx,
                                    T* __restrict__ gi, int64_t NC, int H,
                                    int W, int OH, int OW, int k, int s,
                                    int p) {
-  const int64_t total = NC * H * W;
+  const int W8 = (W + 7) / 8;
+  const int64_t total = NC * H * W8;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += stride) {
-    const int w = (int)(i % W);
-    const int h = (int)((i / W) % H);
-    const int64_t plane = i / ((int64_t)W * H);
-    const int32_t me = h * W + w;
-    // windows with oh*s - p <= h <= oh*s - p + k - 1
+  for (int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; t < total;
+       t += stride) {
+    const int wc = (int)(t % W8);
+    const int h = (int)((t / W8) % H);
+    const int64_t plane = t / ((int64_t)W8 * H);
+    const int w0 = wc * 8;
+    float acc[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) acc[e] = 0.f;
     int oh_lo = (h + p - k + 1 + s - 1) / s; if (oh_lo < 0) oh_lo = 0;
     int oh_hi = (h + p) / s; if (oh_hi > OH - 1) oh_hi = OH - 1;
-    int ow_lo = (w + p - k + 1 + s - 1) / s; if (ow_lo < 0) ow_lo = 0;
-    int ow_hi = (w + p) / s; if (ow_hi > OW - 1) ow_hi = OW - 1;
-    float acc = 0.f;
+    int ow_lo = (w0 + p - k + 1 + s - 1) / s; if (ow_lo < 0) ow_lo = 0;
+    int ow_hi = (w0 + 7 + p) / s; if (ow_hi > OW - 1) ow_hi = OW - 1;
     const T* gop = go + plane * OH * OW;
     const int32_t* ip = idx + plane * OH * OW;
-    for (int oh = oh_lo; oh <= oh_hi; ++oh)
+    for (int oh = oh_lo; oh <= oh_hi; ++oh) {
       for (int ow = ow_lo; ow <= ow_hi; ++ow) {
         const int o = oh * OW + ow;
-        if (ip[o] == me) acc += (float)gop[o];
+        const int32_t winner = ip[o];
+        const int dw = winner - h * W - w0;  // which of my 8 inputs (if any)
+        if (dw >= 0 && dw < 8) acc[dw] += (float)gop[o];
       }
-    gi[i] = (T)acc;
+    }
+    const int64_t ibase = plane * (int64_t)H * W + (int64_t)h * W;
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+      if (w0 + e < W) gi[ibase + w0 + e] = (T)acc[e];
   }
 }
 
@@ -353,35 +380,70 @@ __global__ void avgpool_bwd_kernel(const T* __restrict__ go, T* __restrict__ gi,
 // [sum,sumsq] / [gsum,gxsum] vectors between the two kernels).
 // ---------------------------------------------------------------------------
 
+typedef __attribute__((ext_vector_type(8))) short short8v;
+
+template <typename T>
+__device__ __forceinline__ float to_f32(T v) { return (float)v; }
+
+// 2D grid: x = channel, y = HW chunk (the chip needs >>256 workgroups —
+// guide G1/G11); 16-byte vector loads (G13); fp64 accumulation with one
+// fp64 atomicAdd per block per output.
 template <typename T>
 __global__ void bn_stats64_kernel(const T* __restrict__ x,
-                                  float* __restrict__ out, int64_t N,
-                                  int64_t C, int64_t HW) {
-  for (int64_t ch = blockIdx.x; ch < C; ch += gridDim.x) {
-    double s = 0.0, ss = 0.0;
-    for (int64_t n = 0; n < N; ++n) {
-      const T* p = x + (n * C + ch) * HW;
-      for (int64_t i = threadIdx.x; i < HW; i += blockDim.x) {
-        const double v = (double)(float)p[i];
-        s += v;
-        ss += v * v;
+                                  double* __restrict__ out, int64_t N,
+                                  int64_t C, int64_t HW, int64_t chunk) {
+  const int64_t ch = blockIdx.x;
+  const int64_t i0 = (int64_t)blockIdx.y * chunk;
+  const int64_t i1 = min(i0 + chunk, HW);
+  double s = 0.0, ss = 0.0;
+  const int VEC = sizeof(T) == 2 ? 8 : 4;
+  for (int64_t n = 0; n < N; ++n) {
+    const T* p = x + (n * C + ch) * HW;
+    int64_t i = i0 + (int64_t)threadIdx.x * VEC;
+    if (sizeof(T) == 2) {
+      for (; i + 8 <= i1; i += (int64_t)blockDim.x * 8) {
+        short8v v = *(const short8v*)(p + i);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const float f = to_f32(((const T*)&v)[e]);
+          s += (double)f;
+          ss += (double)(f * f);
+        }
+      }
+    } else {
+      for (; i + 4 <= i1; i += (int64_t)blockDim.x * 4) {
+        const float4 v = *(const float4*)((const float*)p + i);
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          const float f = ((const float*)&v)[e];
+          s += (double)f;
+          ss += (double)(f * f);
+        }
       }
     }
-    for (int off = 32; off > 0; off >>= 1) {
-      s += __shfl_down(s, off, 64);
-      ss += __shfl_down(ss, off, 64);
+    // tail (only the last chunk can be ragged)
+    if (blockIdx.y == gridDim.y - 1) {
+      int64_t tail0 = i1 - (i1 % VEC);
+      for (int64_t t = tail0 + threadIdx.x; t < i1; t += blockDim.x) {
+        const float f = to_f32(p[t]);
+        s += (double)f;
+        ss += (double)(f * f);
+      }
     }
-    __shared__ double ls[8], lss[8];
-    const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
-    if (lane == 0) { ls[wid] = s; lss[wid] = ss; }
-    __syncthreads();
-    if (threadIdx.x == 0) {
-      double ts = 0.0, tss = 0.0;
-      for (int wv = 0; wv < (int)(blockDim.x >> 6); ++wv) { ts += ls[wv]; tss += lss[wv]; }
-      out[ch] = (float)ts;
-      out[C + ch] = (float)tss;
-    }
-    __syncthreads();
+  }
+  for (int off = 32; off > 0; off >>= 1) {
+    s += __shfl_down(s, off, 64);
+    ss += __shfl_down(ss, off, 64);
+  }
+  __shared__ double ls[8], lss[8];
+  const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
+  if (lane == 0) { ls[wid] = s; lss[wid] = ss; }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    double ts = 0.0, tss = 0.0;
+    for (int wv = 0; wv < (int)(blockDim.x >> 6); ++wv) { ts += ls[wv]; tss += lss[wv]; }
+    atomicAdd(&out[ch], ts);
+    atomicAdd(&out[C + ch], tss);
   }
 }
 
@@ -392,34 +454,80 @@ __global__ void bn_apply_kernel(const T* __restrict__ x, T* __restrict__ y,
                                 const float* __restrict__ wgt,
                                 const float* __restrict__ bias, int64_t N,
                                 int64_t C, int64_t HW) {
-  const int64_t total = N * C * HW;
+  // 8 elements per thread (HW % 8 == 0 fast path enforced host-side)
+  const int64_t total8 = N * C * HW / 8;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += stride) {
+  for (int64_t i8 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i8 < total8; i8 += stride) {
+    const int64_t i = i8 * 8;
     const int64_t ch = (i / HW) % C;
-    float v = ((float)x[i] - mean[ch]) * invstd[ch];
-    v = v * wgt[ch] + bias[ch];
-    if (RELU) v = v > 0.f ? v : 0.f;
-    y[i] = (T)v;
+    const float m = mean[ch], inv = invstd[ch], w = wgt[ch], b = bias[ch];
+    if (sizeof(T) == 2) {
+      short8v vx = *(const short8v*)((const short*)x + i);
+      short8v vy;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        float v = (to_f32(((const T*)&vx)[e]) - m) * inv * w + b;
+        if (RELU) v = v > 0.f ? v : 0.f;
+        ((T*)&vy)[e] = (T)v;
+      }
+      *(short8v*)((short*)y + i) = vy;
+    } else {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        float v = ((float)x[i + e] - m) * inv * w + b;
+        if (RELU) v = v > 0.f ? v : 0.f;
+        y[i + e] = (T)v;
+      }
+    }
   }
 }
 
 // per-channel [gsum, gxsum] where gxsum = sum(go * xhat); RELU masks go
-// by (y > 0) — pass y=nullptr when no fusion.
+// by (y > 0) — pass y=nullptr when no fusion. 2D grid + vector loads.
 template <typename T, bool RELU>
 __global__ void bn_bwd_stats_kernel(const T* __restrict__ go,
                                     const T* __restrict__ x,
                                     const T* __restrict__ y,
                                     const float* __restrict__ mean,
                                     const float* __restrict__ invstd,
-                                    float* __restrict__ out, int64_t N,
-                                    int64_t C, int64_t HW) {
-  for (int64_t ch = blockIdx.x; ch < C; ch += gridDim.x) {
-    const float m = mean[ch], inv = invstd[ch];
-    double gs = 0.0, gx = 0.0;
-    for (int64_t n = 0; n < N; ++n) {
-      const int64_t off = (n * C + ch) * HW;
-      for (int64_t i = threadIdx.x; i < HW; i += blockDim.x) {
+                                    double* __restrict__ out, int64_t N,
+                                    int64_t C, int64_t HW, int64_t chunk) {
+  const int64_t ch = blockIdx.x;
+  const int64_t i0 = (int64_t)blockIdx.y * chunk;
+  const int64_t i1 = min(i0 + chunk, HW);
+  const float m = mean[ch], inv = invstd[ch];
+  double gs = 0.0, gx = 0.0;
+  for (int64_t n = 0; n < N; ++n) {
+    const int64_t off = (n * C + ch) * HW;
+    if (sizeof(T) == 2) {
+      int64_t i = i0 + (int64_t)threadIdx.x * 8;
+      for (; i + 8 <= i1; i += (int64_t)blockDim.x * 8) {
+        short8v vg = *(const short8v*)((const short*)go + off + i);
+        short8v vx = *(const short8v*)((const short*)x + off + i);
+        short8v vy;
+        if (RELU) vy = *(const short8v*)((const short*)y + off + i);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          float g = to_f32(((const T*)&vg)[e]);
+          if (RELU && to_f32(((const T*)&vy)[e]) <= 0.f) g = 0.f;
+          const float xh = (to_f32(((const T*)&vx)[e]) - m) * inv;
+          gs += (double)g;
+          gx += (double)(g * xh);
+        }
+      }
+      if (blockIdx.y == gridDim.y - 1) {
+        int64_t tail0 = i1 - (i1 % 8);
+        for (int64_t t = tail0 + threadIdx.x; t < i1; t += blockDim.x) {
+          float g = (float)go[off + t];
+          if (RELU && (float)y[off + t] <= 0.f) g = 0.f;
+          const float xh = ((float)x[off + t] - m) * inv;
+          gs += (double)g;
+          gx += (double)(g * xh);
+        }
+      }
+    } else {
+      for (int64_t i = i0 + threadIdx.x; i < i1; i += blockDim.x) {
         float g = (float)go[off + i];
         if (RELU && (float)y[off + i] <= 0.f) g = 0.f;
         const float xh = ((float)x[off + i] - m) * inv;
@@ -427,21 +535,20 @@ __global__ void bn_bwd_stats_kernel(const T* __restrict__ go,
         gx += (double)(g * xh);
       }
     }
-    for (int off2 = 32; off2 > 0; off2 >>= 1) {
-      gs += __shfl_down(gs, off2, 64);
-      gx += __shfl_down(gx, off2, 64);
-    }
-    __shared__ double l1[8], l2[8];
-    const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
-    if (lane == 0) { l1[wid] = gs; l2[wid] = gx; }
-    __syncthreads();
-    if (threadIdx.x == 0) {
-      double t1 = 0.0, t2 = 0.0;
-      for (int wv = 0; wv < (int)(blockDim.x >> 6); ++wv) { t1 += l1[wv]; t2 += l2[wv]; }
-      out[ch] = (float)t1;
-      out[C + ch] = (float)t2;
-    }
-    __syncthreads();
+  }
+  for (int off2 = 32; off2 > 0; off2 >>= 1) {
+    gs += __shfl_down(gs, off2, 64);
+    gx += __shfl_down(gx, off2, 64);
+  }
+  __shared__ double l1[8], l2[8];
+  const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
+  if (lane == 0) { l1[wid] = gs; l2[wid] = gx; }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    double t1 = 0.0, t2 = 0.0;
+    for (int wv = 0; wv < (int)(blockDim.x >> 6); ++wv) { t1 += l1[wv]; t2 += l2[wv]; }
+    atomicAdd(&out[ch], t1);
+    atomicAdd(&out[C + ch], t2);
   }
 }
 
@@ -457,19 +564,85 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ go,
                                     const float* __restrict__ gxsum,
                                     double inv_n, int64_t N, int64_t C,
                                     int64_t HW) {
-  const int64_t total = N * C * HW;
+  const int64_t total8 = N * C * HW / 8;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += stride) {
+  for (int64_t i8 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i8 < total8; i8 += stride) {
+    const int64_t i = i8 * 8;
     const int64_t ch = (i / HW) % C;
-    float g = (float)go[i];
-    if (RELU && (float)y[i] <= 0.f) g = 0.f;
-    const float m = mean[ch], inv = invstd[ch];
-    const float xh = ((float)x[i] - m) * inv;
-    const float t =
-        g - (float)(gsum[ch] * inv_n) - xh * (float)(gxsum[ch] * inv_n);
-    gi[i] = (T)(t * wgt[ch] * inv);
+    const float m = mean[ch], inv = invstd[ch], w = wgt[ch];
+    const float gsn = (float)(gsum[ch] * inv_n);
+    const float gxn = (float)(gxsum[ch] * inv_n);
+    if (sizeof(T) == 2) {
+      short8v vg = *(const short8v*)((const short*)go + i);
+      short8v vx = *(const short8v*)((const short*)x + i);
+      short8v vy;
+      if (RELU) vy = *(const short8v*)((const short*)y + i);
+      short8v vo;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        float g = to_f32(((const T*)&vg)[e]);
+        if (RELU && to_f32(((const T*)&vy)[e]) <= 0.f) g = 0.f;
+        const float xh = (to_f32(((const T*)&vx)[e]) - m) * inv;
+        ((T*)&vo)[e] = (T)((g - gsn - xh * gxn) * w * inv);
+      }
+      *(short8v*)((short*)gi + i) = vo;
+    } else {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        float g = (float)go[i + e];
+        if (RELU && (float)y[i + e] <= 0.f) g = 0.f;
+        const float xh = ((float)x[i + e] - m) * inv;
+        gi[i + e] = (T)((g - gsn - xh * gxn) * w * inv);
+      }
+    }
   }
+}
+
+// ---------------------------------------------------------------------------
+// Fused momentum-SGD on flat fp32 buffers: one kernel for the whole
+// model step (replaces ~3 torch launches per parameter — the profile
+// showed 5.8k tiny CUDAFunctor_add kernels per step).
+//   v = mu*v + g (+ wd*p);  p -= lr*v;  then g = 0.
+// ---------------------------------------------------------------------------
+
+__global__ void sgd_momentum_kernel(float* __restrict__ p,
+                                    float* __restrict__ g,
+                                    float* __restrict__ v, float lr, float mu,
+                                    float wd, int64_t n4) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n4;
+       i += stride) {
+    float4 pv = *(float4*)(p + i * 4);
+    float4 gv = *(float4*)(g + i * 4);
+    float4 vv = *(float4*)(v + i * 4);
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      float grad = ((float*)&gv)[e] + wd * ((float*)&pv)[e];
+      float vel = mu * ((float*)&vv)[e] + grad;
+      ((float*)&vv)[e] = vel;
+      ((float*)&pv)[e] = ((float*)&pv)[e] - lr * vel;
+      ((float*)&gv)[e] = 0.f;
+    }
+    *(float4*)(p + i * 4) = pv;
+    *(float4*)(v + i * 4) = vv;
+    *(float4*)(g + i * 4) = gv;
+  }
+}
+
+void sgd_momentum(torch::Tensor p, torch::Tensor g, torch::Tensor v,
+                  double lr, double mu, double wd) {
+  TORCH_CHECK(p.is_cuda() && p.is_contiguous() && g.is_contiguous() &&
+              v.is_contiguous());
+  TORCH_CHECK(p.scalar_type() == torch::kFloat);
+  TORCH_CHECK(p.numel() % 4 == 0, "flat buffer must be padded to 4");
+  const int64_t n4 = p.numel() / 4;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  const int grid = (int)std::min<int64_t>((n4 + 255) / 256, 2048);
+  hipLaunchKernelGGL(sgd_momentum_kernel, dim3(grid), dim3(256), 0,
+                     stream.stream(), p.data_ptr<float>(),
+                     g.data_ptr<float>(), v.data_ptr<float>(), (float)lr,
+                     (float)mu, (float)wd, n4);
 }
 
 // ---------------- torch-facing wrappers ----------------
@@ -568,19 +741,29 @@ torch::Tensor avgpool_bwd(torch::Tensor go, int64_t H, int64_t W, int64_t k,
   return gi;
 }
 
+static int64_t stats_splits(int64_t C, int64_t HW) {
+  // target ~2048 workgroups (256 CUs x 8); chunk in multiples of 2048
+  int64_t splits = std::max<int64_t>(1, 2048 / std::max<int64_t>(C, 1));
+  int64_t max_splits = std::max<int64_t>(1, (HW + 2047) / 2048);
+  return std::min(splits, max_splits);
+}
+
 torch::Tensor bn_stats64(torch::Tensor x) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.is_contiguous());
   const int64_t N = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
-  auto out = torch::empty({2 * C}, x.options().dtype(torch::kFloat));
-  const int grid = (int)std::min<int64_t>(C, 2048);
+  auto out = torch::zeros({2 * C}, x.options().dtype(torch::kDouble));
+  const int64_t splits = stats_splits(C, HW);
+  int64_t chunk = (HW + splits - 1) / splits;
+  chunk = ((chunk + 2047) / 2048) * 2048;
+  const int64_t gy = (HW + chunk - 1) / chunk;
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::Half, at::ScalarType::BFloat16, x.scalar_type(),
       "bn_stats64", [&] {
-        hipLaunchKernelGGL((bn_stats64_kernel<scalar_t>), dim3(grid),
-                           dim3(256), 0, stream.stream(),
-                           x.data_ptr<scalar_t>(), out.data_ptr<float>(), N, C,
-                           HW);
+        hipLaunchKernelGGL((bn_stats64_kernel<scalar_t>),
+                           dim3((uint32_t)C, (uint32_t)gy), dim3(256), 0,
+                           stream.stream(), x.data_ptr<scalar_t>(),
+                           out.data_ptr<double>(), N, C, HW, chunk);
       });
   return out;
 }
@@ -618,26 +801,29 @@ torch::Tensor bn_bwd_stats(torch::Tensor go, torch::Tensor x, torch::Tensor y,
                            torch::Tensor mean, torch::Tensor invstd,
                            bool relu) {
   const int64_t N = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
-  auto out = torch::empty({2 * C}, x.options().dtype(torch::kFloat));
-  const int grid = (int)std::min<int64_t>(C, 2048);
+  auto out = torch::zeros({2 * C}, x.options().dtype(torch::kDouble));
+  const int64_t splits = stats_splits(C, HW);
+  int64_t chunk = (HW + splits - 1) / splits;
+  chunk = ((chunk + 2047) / 2048) * 2048;
+  const int64_t gy = (HW + chunk - 1) / chunk;
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::Half, at::ScalarType::BFloat16, x.scalar_type(),
       "bn_bwd_stats", [&] {
         if (relu)
-          hipLaunchKernelGGL((bn_bwd_stats_kernel<scalar_t, true>), dim3(grid),
-                             dim3(256), 0, stream.stream(),
-                             go.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
-                             y.data_ptr<scalar_t>(), mean.data_ptr<float>(),
-                             invstd.data_ptr<float>(), out.data_ptr<float>(),
-                             N, C, HW);
+          hipLaunchKernelGGL((bn_bwd_stats_kernel<scalar_t, true>),
+                             dim3((uint32_t)C, (uint32_t)gy), dim3(256), 0,
+                             stream.stream(), go.data_ptr<scalar_t>(),
+                             x.data_ptr<scalar_t>(), y.data_ptr<scalar_t>(),
+                             mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                             out.data_ptr<double>(), N, C, HW, chunk);
         else
           hipLaunchKernelGGL((bn_bwd_stats_kernel<scalar_t, false>),
-                             dim3(grid), dim3(256), 0, stream.stream(),
-                             go.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
-                             (scalar_t*)nullptr, mean.data_ptr<float>(),
-                             invstd.data_ptr<float>(), out.data_ptr<float>(),
-                             N, C, HW);
+                             dim3((uint32_t)C, (uint32_t)gy), dim3(256), 0,
+                             stream.stream(), go.data_ptr<scalar_t>(),
+                             x.data_ptr<scalar_t>(), (scalar_t*)nullptr,
+                             mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                             out.data_ptr<double>(), N, C, HW, chunk);
       });
   return out;
 }
@@ -677,7 +863,10 @@ torch::Tensor bn_bwd_apply(torch::Tensor go, torch::Tensor x, torch::Tensor y,
 
 }  // namespace
 
+void register_conv_mfma(pybind11::module_& m);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  register_conv_mfma(m);
   m.def("halo_pack", &halo_copy<0>, "pack halo strips tile->flat buffer");
   m.def("halo_unpack", &halo_copy<1>, "unpack halo strips buffer->tile");
   m.def("halo_unpack_add", &halo_copy<2>, "accumulate grad strips into tile");
@@ -690,5 +879,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_apply", &bn_apply);
   m.def("bn_bwd_stats", &bn_bwd_stats);
   m.def("bn_bwd_apply", &bn_bwd_apply);
+  m.def("sgd_momentum", &sgd_momentum);
   m.attr("gfx_arch") = "gfx950";
 }

@@ -1,0 +1,99 @@
+"""NativeConv2d: implicit-GEMM MFMA conv (csrc/conv_mfma.hip) behind an
+nn.Conv2d-compatible module.
+
+Forward and weight-gradient run on the hand-written gfx950 kernels;
+data-gradient for stride-1 convs is the SAME forward kernel applied to
+the (C<->K transposed, 180-rotated) weights — the classic transposed
+convolution identity — so one kernel covers both. Stride-2 data-grad
+falls back to torch (MIOpen) for now.
+
+Dispatch: CUDA + bf16 (or under autocast) + dilation/groups == 1 + OW
+wide enough for the row-segment tiling to be efficient. Everything else
+uses F.conv2d. MPI4DL_NATIVE_CONV=0 disables, =1 forces (for A/B
+benchmarking).
+"""
+
+from __future__ import annotations
+
+import os
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from . import backend
+
+_MIN_OW = 64
+
+
+def _pair(v):
+    return (v, v) if isinstance(v, int) else tuple(v)
+
+
+class ConvFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, bias, sh, sw, ph, pw):
+        ge = backend.ext()
+        xb = x.contiguous().to(torch.bfloat16)
+        wb = w.contiguous().to(torch.bfloat16)
+        y = ge.conv_fwd(xb, wb, bias, sh, sw, ph, pw)
+        ctx.save_for_backward(xb, wb)
+        ctx.geom = (sh, sw, ph, pw)
+        ctx.wdtype = w.dtype
+        ctx.has_bias = bias is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, go):
+        ge = backend.ext()
+        xb, wb = ctx.saved_tensors
+        sh, sw, ph, pw = ctx.geom
+        go = go.contiguous().to(torch.bfloat16)
+        K, C, R, S = wb.shape
+        gw = ge.conv_bwd_weight(go, xb, R, S, sh, sw, ph, pw).to(ctx.wdtype)
+        gb = go.sum(dim=(0, 2, 3)).to(ctx.wdtype) if ctx.has_bias else None
+        if sh == 1 and sw == 1:
+            wt = wb.transpose(0, 1).flip(2, 3).contiguous()
+            gx = ge.conv_fwd(go, wt, None, 1, 1, R - 1 - ph, S - 1 - pw)
+        else:
+            gx = torch.nn.grad.conv2d_input(
+                list(xb.shape), wb, go, stride=(sh, sw), padding=(ph, pw)
+            )
+        return gx, gw, gb, None, None, None, None
+
+
+def native_conv2d(x, w, bias, stride, padding):
+    sh, sw = _pair(stride)
+    ph, pw = _pair(padding)
+    b32 = bias.float().contiguous() if bias is not None else None
+    return ConvFn.apply(x, w, b32, sh, sw, ph, pw)
+
+
+def _dispatchable(x, conv: nn.Conv2d) -> bool:
+    mode = os.environ.get("MPI4DL_NATIVE_CONV", "auto")
+    if mode == "0":
+        return False
+    if not (x.is_cuda and not x.is_meta):
+        return False
+    if conv.groups != 1 or _pair(conv.dilation) != (1, 1):
+        return False
+    bf16_ok = x.dtype == torch.bfloat16 or torch.is_autocast_enabled()
+    if not bf16_ok:
+        return False
+    if mode == "1":
+        return True
+    sh, sw = _pair(conv.stride)
+    ph, pw = _pair(conv.padding)
+    kh, kw = _pair(conv.kernel_size)
+    ow = (x.shape[-1] + 2 * pw - kw) // sw + 1
+    return ow >= _MIN_OW
+
+
+class NativeConv2d(nn.Conv2d):
+    """nn.Conv2d that runs on the gemscore implicit-GEMM MFMA kernels
+    when dispatchable (state-dict compatible with nn.Conv2d)."""
+
+    def forward(self, x):
+        if _dispatchable(x, self):
+            return native_conv2d(x, self.weight, self.bias, self.stride, self.padding)
+        return super().forward(x)

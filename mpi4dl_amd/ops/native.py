@@ -92,22 +92,23 @@ class BatchNormFn(torch.autograd.Function):
         x, y, mean, invstd, w32 = ctx.saved_tensors
         go = go.contiguous()
         C = x.shape[1]
-        stats = ge.bn_bwd_stats(go, x, y, mean, invstd, ctx.relu)
-        gw = stats[C:].clone()  # local sum(go*xhat)
-        gb = stats[:C].clone()  # local sum(go)
+        stats = ge.bn_bwd_stats(go, x, y, mean, invstd, ctx.relu)  # fp64
+        gw = stats[C:].float()  # local sum(go*xhat)
+        gb = stats[:C].float()  # local sum(go)
         if ctx.training:
             g_global = stats
             if ctx.group is not None:
                 g_global = stats.clone()
                 dist.all_reduce(g_global, group=ctx.group)
+            gf = g_global.float()
             gi = ge.bn_bwd_apply(
                 go, x, y, mean, invstd, w32,
-                g_global[:C].contiguous(), g_global[C:].contiguous(),
+                gf[:C].contiguous(), gf[C:].contiguous(),
                 float(ctx.n_global), ctx.relu,
             )
         else:
             # eval: mean/var are constants -> gi = go_eff * w * invstd
-            zeros = torch.zeros_like(stats)
+            zeros = torch.zeros(2 * C, device=x.device, dtype=torch.float32)
             gi = ge.bn_bwd_apply(
                 go, x, y, mean, invstd, w32,
                 zeros[:C].contiguous(), zeros[C:].contiguous(),

@@ -65,7 +65,7 @@ class TileBatchNorm2d(nn.BatchNorm2d):
         )
 
     def forward(self, x):
-        if x.is_cuda and not x.is_meta:
+        if x.is_cuda and not x.is_meta and (x.shape[-2] * x.shape[-1]) % 8 == 0:
             return self._forward_native(x)
         if not self._use_sync(x):
             y = super().forward(x)
@@ -104,12 +104,14 @@ class TileBatchNorm2d(nn.BatchNorm2d):
         n = x.numel() // C
         group = self.group if (self.group is not None and dist.is_initialized()) else None
         if self.training:
-            stats = ge.bn_stats64(x)
+            stats = ge.bn_stats64(x)  # fp64 [sum, sumsq]
             if group is not None:
                 dist.all_reduce(stats, group=group)
                 n *= dist.get_world_size(group=group)
-            mean = stats[:C] / n
-            var = stats[C:] / n - mean * mean
+            mean64 = stats[:C] / n
+            var64 = stats[C:] / n - mean64 * mean64
+            mean = mean64.float()
+            var = var64.float()
             if self.track_running_stats:
                 with torch.no_grad():
                     m = self.momentum if self.momentum is not None else 0.1

@@ -28,6 +28,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from .conv_native import NativeConv2d
 from .halo import HaloExchanger, TileLayout, halo_pad
 
 
@@ -112,7 +113,7 @@ class HaloConv2d(_SpatialBase):
         self.halo_len = tuple(padding) if halo_len is None else halo_len
         self.d2 = d2
         self.outer_pad = tuple(padding) if d2 else (0, 0)
-        self.conv = nn.Conv2d(
+        self.conv = NativeConv2d(
             in_channels, out_channels, kernel_size, stride=stride, padding=0, bias=bias
         )
         self.stride = stride

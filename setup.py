@@ -21,6 +21,7 @@ ext = CUDAExtension(
     name="mpi4dl_amd._gemscore",
     sources=[
         "mpi4dl_amd/csrc/gemscore.hip",
+        "mpi4dl_amd/csrc/conv_mfma.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

@@ -213,3 +213,60 @@ def test_native_batchnorm_relu_fused():
     y.backward(g)
     y_ref.backward(g)
     assert torch.allclose(x.grad, x2.grad, atol=1e-4)
+
+
+@gpu
+@requires_gpu
+@pytest.mark.parametrize(
+    "C,K,HW,k,s,p",
+    [
+        (32, 64, 128, 3, 1, 1),
+        (27, 104, 256, 3, 2, 1),   # stem-like: CRS=243, stride 2
+        (3, 104, 128, 3, 2, 1),    # CRS=27 < BK (masking)
+        (64, 64, 128, 1, 1, 0),    # 1x1
+        (64, 64, 128, 7, 1, 3),    # 7x7
+    ],
+)
+def test_native_conv_fwd_matches_torch(C, K, HW, k, s, p):
+    from mpi4dl_amd.ops import backend
+
+    ge = backend.ext()
+    torch.manual_seed(0)
+    x = torch.randn(2, C, HW, HW, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(K, C, k, k, device="cuda", dtype=torch.bfloat16) * 0.1
+    b = torch.randn(K, device="cuda", dtype=torch.float32)
+    y = ge.conv_fwd(x, w, b, s, s, p, p)
+    ref = torch.nn.functional.conv2d(
+        x.float(), w.float(), b, stride=s, padding=p
+    )
+    err = (y.float() - ref).abs().max() / max(ref.abs().max().item(), 1e-3)
+    assert err < 0.05, f"rel err {err}"
+
+
+@gpu
+@requires_gpu
+def test_native_conv_autograd_matches_torch():
+    from mpi4dl_amd.ops.conv_native import native_conv2d
+
+    torch.manual_seed(0)
+    C, K, HW, k = 16, 32, 96, 3
+    x = torch.randn(2, C, HW, HW, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    w = torch.randn(K, C, k, k, device="cuda", dtype=torch.float32, requires_grad=True) * 0.1
+    b = torch.randn(K, device="cuda", requires_grad=True)
+    y = native_conv2d(x, w, b, 1, 1)
+    x2 = x.detach().float().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    b2 = b.detach().clone().requires_grad_(True)
+    ref = torch.nn.functional.conv2d(x2, w2, b2, stride=1, padding=1)
+    rel = (y.float() - ref).abs().max() / ref.abs().max()
+    assert rel < 0.05, rel
+    g = torch.randn_like(ref)
+    y.backward(g.to(torch.bfloat16))
+    ref.backward(g)
+    for a, bb, name in [
+        (x.grad.float(), x2.grad, "gx"),
+        (w.grad, w2.grad, "gw"),
+        (b.grad, b2.grad, "gb"),
+    ]:
+        rel = (a - bb).abs().max() / max(bb.abs().max().item(), 1e-3)
+        assert rel < 0.08, (name, rel)
