@@ -175,7 +175,7 @@ def test_native_batchnorm_matches_torch():
     bn = TileBatchNorm2d(9).cuda()
     ref = torch.nn.BatchNorm2d(9).cuda()
     ref.load_state_dict({k: v for k, v in bn.state_dict().items()})
-    x = torch.randn(3, 9, 17, 21, device="cuda", requires_grad=True)
+    x = torch.randn(3, 9, 16, 24, device="cuda", requires_grad=True)  # HW%8==0 -> native path
     x2 = x.detach().clone().requires_grad_(True)
     y = bn(x)
     y_ref = ref(x2)
