@@ -82,10 +82,10 @@ __global__ __launch_bounds__(256) void conv_fwd_kernel(
   // LDS: A image [BM][BK] bf16 (+8 pad per row vs bank conflicts),
   //      B image (BN/16) blocks x 512 elems, double buffered.
   __shared__ __attribute__((aligned(16))) short lds[2 * (BM * (BK + 8)) +
-                                                    2 * (BN * BK)];
+                                                    2 * (BN * (BK + 8))];
   auto ldsA = [&](int buf) { return lds + buf * (BM * (BK + 8)); };
   auto ldsB = [&](int buf) {
-    return lds + 2 * (BM * (BK + 8)) + buf * (BN * BK);
+    return lds + 2 * (BM * (BK + 8)) + buf * (BN * (BK + 8));
   };
 
   f32x4 acc[4][4];
@@ -149,14 +149,12 @@ __global__ __launch_bounds__(256) void conv_fwd_kernel(
           }
         }
       }
-      // B image offset: block pb = px/16; within block:
-      // off = (px&15) + (k&3)*16 + (k>>3)*64 + ((k>>2)&1)*256
-      const int pb = (px0 >> 4);
-      const int base = pb * 512 + (kk & 3) * 16 + ((kk >> 3) << 6) +
-                       (((kk >> 2) & 1) << 8);
-      short* dst = ldsB(buf) + base + (px0 & 15);
+      // B image [BN][BK+8]: row = pixel, k contiguous -> fragment reads
+      // are plain 16-byte ds_read_b128; the (global pixel-contiguous ->
+      // LDS k-contiguous) transpose happens here as 8 b16 scatter writes.
+      short* dstB = ldsB(buf) + kk;
 #pragma unroll
-      for (int e = 0; e < 8; ++e) dst[e] = v[e];
+      for (int e = 0; e < 8; ++e) dstB[(px0 + e) * (BK + 8)] = v[e];
     }
   };
 
@@ -180,23 +178,10 @@ __global__ __launch_bounds__(256) void conv_fwd_kernel(
       s16x8 afrag = *(const s16x8*)arow;
 #pragma unroll
       for (int nf = 0; nf < 4; ++nf) {
-        const int pb = (b_px0 >> 4) + nf;
-        const short* bbase = ldsB(buf) + pb * 512;
-        // tr reads: lane-dependent addressing handled by the instruction
-        s16x4 b0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-            (__attribute__((address_space(3))) s16x4*)(
-                (__attribute__((address_space(3))) short*)(bbase) +
-                (lane & 15) + ((lane >> 4) << 6)));
-        s16x4 b1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-            (__attribute__((address_space(3))) s16x4*)(
-                (__attribute__((address_space(3))) short*)(bbase + 256) +
-                (lane & 15) + ((lane >> 4) << 6)));
-        s16x8 bfrag;
-#pragma unroll
-        for (int e = 0; e < 4; ++e) {
-          bfrag[e] = b0[e];
-          bfrag[e + 4] = b1[e];
-        }
+        const short* brow = ldsB(buf) +
+                            (b_px0 + nf * 16 + (lane & 15)) * (BK + 8) +
+                            ((lane >> 4) << 3);
+        s16x8 bfrag = *(const s16x8*)brow;
         acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             afrag, bfrag, acc[mf][nf], 0, 0, 0);
       }
@@ -442,6 +427,41 @@ torch::Tensor conv_bwd_weight(torch::Tensor go, torch::Tensor x,
   return gw.view({(int64_t)K, (int64_t)C, R, S});
 }
 
+// ---------------------------------------------------------------------------
+// ds_read_b64_tr_b16 semantics probe: LDS holds arange shorts; each lane
+// does one tr read with a configurable per-lane address pattern; output
+// [64][4] shows exactly which LDS elements land in which lane/elem.
+// ---------------------------------------------------------------------------
+
+__global__ void tr16_probe_kernel(short* __restrict__ out, int mode) {
+  __shared__ __attribute__((aligned(16))) short lds[1024];
+  for (int i = threadIdx.x; i < 1024; i += blockDim.x) lds[i] = (short)i;
+  __syncthreads();
+  if (threadIdx.x < 64) {
+    const int l = threadIdx.x;
+    int off;
+    if (mode == 0) off = 0;                               // uniform
+    else if (mode == 1) off = (l & 15) + ((l >> 4) << 6); // my assumed map
+    else if (mode == 2) off = l * 4;                      // linear x4
+    else off = (l & 15) * 4 + ((l >> 4) << 6);            // 4-elem rows
+    s16x4 v = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+        (__attribute__((address_space(3))) s16x4*)(
+            (__attribute__((address_space(3))) short*)lds + off));
+#pragma unroll
+    for (int e = 0; e < 4; ++e) out[l * 4 + e] = v[e];
+  }
+}
+
+torch::Tensor tr16_probe(int64_t mode) {
+  auto out = torch::zeros({64, 4}, torch::TensorOptions()
+                                       .dtype(torch::kShort)
+                                       .device(torch::kCUDA));
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  hipLaunchKernelGGL(tr16_probe_kernel, dim3(1), dim3(64), 0, stream.stream(),
+                     out.data_ptr<short>(), (int)mode);
+  return out;
+}
+
 }  // namespace conv_mfma
 
 void register_conv_mfma(pybind11::module_& m) {
@@ -449,4 +469,6 @@ void register_conv_mfma(pybind11::module_& m) {
         "implicit-GEMM MFMA conv forward (bf16 NCHW)");
   m.def("conv_bwd_weight", &conv_mfma::conv_bwd_weight,
         "implicit-GEMM MFMA conv weight gradient (fp32 out)");
+  m.def("tr16_probe", &conv_mfma::tr16_probe,
+        "ds_read_b64_tr_b16 lane-mapping probe");
 }
