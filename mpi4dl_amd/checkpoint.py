@@ -153,9 +153,8 @@ def consolidate(path: str) -> dict:
         if split_rank in seen_stages:
             continue
         seen_stages.add(split_rank)
-        start = sum(balance[:split_rank]) if balance else 0
+        # nn.Sequential slices preserve the ORIGINAL child names, so stage
+        # state_dict keys are already cell-global.
         for k, v in shard["model"].items():
-            head, _, rest = k.partition(".")
-            gk = f"{int(head) + start}.{rest}" if head.isdigit() else k
-            full[gk] = v
+            full[k] = v
     return full
