@@ -238,20 +238,39 @@ __global__ void maxpool_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
     int32_t bidx[8];
 #pragma unroll
     for (int e = 0; e < 8; ++e) { best[e] = -INFINITY; bidx[e] = 0; }
+    const int w_lo = ow0 * s - p;
+    const int span = 8 * s + k - s;  // input row span (<= 17 for k<=3,s<=2)
     for (int kh = 0; kh < k; ++kh) {
       const int h = h0 + kh;
       if (h < 0 || h >= H) continue;
       const T* row = xp + h * W;
-      // window of this row needed: [ow0*s-p, (ow0+7)*s-p+k)
-      const int w_lo = ow0 * s - p;
+      if (span <= 24) {
+        // load the span once into registers (bounds -> -inf), then pure
+        // register max per output
+        float seg[24];
 #pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        const int wbase = w_lo + e * s;
-        for (int kw = 0; kw < k; ++kw) {
-          const int w = wbase + kw;
-          if (w < 0 || w >= W) continue;
-          const float v = (float)row[w];
-          if (v > best[e]) { best[e] = v; bidx[e] = h * W + w; }
+        for (int j = 0; j < 24; ++j) {
+          const int w = w_lo + j;
+          seg[j] = (j < span && w >= 0 && w < W) ? (float)row[w] : -INFINITY;
+        }
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const int base = e * s;
+          for (int kw = 0; kw < k; ++kw) {
+            const float v = seg[base + kw];
+            if (v > best[e]) { best[e] = v; bidx[e] = h * W + w_lo + base + kw; }
+          }
+        }
+      } else {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const int wbase = w_lo + e * s;
+          for (int kw = 0; kw < k; ++kw) {
+            const int w = wbase + kw;
+            if (w < 0 || w >= W) continue;
+            const float v = (float)row[w];
+            if (v > best[e]) { best[e] = v; bidx[e] = h * W + w; }
+          }
         }
       }
     }
@@ -645,6 +664,48 @@ void sgd_momentum(torch::Tensor p, torch::Tensor g, torch::Tensor v,
                      (float)mu, (float)wd, n4);
 }
 
+// bn_finalize: fold the fp64 [sum,sumsq] -> [mean, invstd] conversion,
+// the running-stat EMA update and num_batches_tracked into ONE kernel
+// (the python glue cost ~8 small launches per BN call).
+__global__ void bn_finalize_kernel(const double* __restrict__ stats,
+                                   float* __restrict__ out,
+                                   float* __restrict__ rmean,
+                                   float* __restrict__ rvar,
+                                   int64_t* __restrict__ tracked, float mom,
+                                   double n, float eps, int64_t C) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= C) return;
+  const double mean = stats[i] / n;
+  double var = stats[C + i] / n - mean * mean;
+  if (var < 0.0) var = 0.0;
+  out[i] = (float)mean;
+  out[C + i] = rsqrtf((float)var + eps);
+  if (rmean) {
+    rmean[i] = (1.f - mom) * rmean[i] + mom * (float)mean;
+    const float unb = (float)(var * (n / (n > 1.0 ? n - 1.0 : 1.0)));
+    rvar[i] = (1.f - mom) * rvar[i] + mom * unb;
+  }
+  if (i == 0 && tracked) tracked[0] += 1;
+}
+
+torch::Tensor bn_finalize(torch::Tensor stats,
+                          c10::optional<torch::Tensor> rmean,
+                          c10::optional<torch::Tensor> rvar,
+                          c10::optional<torch::Tensor> tracked, double mom,
+                          double n, double eps) {
+  const int64_t C = stats.numel() / 2;
+  auto out = torch::empty({2 * C}, stats.options().dtype(torch::kFloat));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(
+      bn_finalize_kernel, dim3((uint32_t)((C + 255) / 256)), dim3(256), 0,
+      stream.stream(), stats.data_ptr<double>(), out.data_ptr<float>(),
+      rmean.has_value() ? rmean->data_ptr<float>() : nullptr,
+      rvar.has_value() ? rvar->data_ptr<float>() : nullptr,
+      tracked.has_value() ? tracked->data_ptr<int64_t>() : nullptr,
+      (float)mom, n, (float)eps, C);
+  return out;
+}
+
 // ---------------- torch-facing wrappers ----------------
 
 static inline int grid_for(int64_t total, int block) {
@@ -880,5 +941,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_bwd_stats", &bn_bwd_stats);
   m.def("bn_bwd_apply", &bn_bwd_apply);
   m.def("sgd_momentum", &sgd_momentum);
+  m.def("bn_finalize", &bn_finalize);
   m.attr("gfx_arch") = "gfx950";
 }

@@ -125,15 +125,17 @@ def make_op(name: str, c: int, stride: int, ctx, mknorm) -> nn.Module:
     if name == "max_pool_2x2":
         # padding 0, stride 2 in reduction cells: tile-local, no halo
         return HaloPool2d("max", 2, stride=stride, padding=0)
+    # conv ops START with ReLU of their input state in the reference; the
+    # Cell precomputes relu(state) ONCE and passes it to conv ops (pool /
+    # none ops get the raw state), removing ~6 redundant full-tensor ReLU
+    # kernels per cell. The leading ReLU module is therefore dropped here.
     if name == "conv_1x1":
         return nn.Sequential(
-            nn.ReLU(inplace=False),
             nn.Conv2d(c, c, 1, stride=stride, bias=False),
             mknorm(c),
         )
     if name == "conv_3x3":
         return nn.Sequential(
-            nn.ReLU(inplace=False),
             nn.Conv2d(c, c // 4, 1, bias=False),
             mknorm(c // 4),
             nn.ReLU(inplace=False),
@@ -145,7 +147,6 @@ def make_op(name: str, c: int, stride: int, ctx, mknorm) -> nn.Module:
         )
     if name == "conv_1x7_7x1":
         return nn.Sequential(
-            nn.ReLU(inplace=False),
             nn.Conv2d(c, c // 4, 1, stride=1, bias=False),
             mknorm(c // 4),
             nn.ReLU(inplace=False),
@@ -206,6 +207,7 @@ class Cell(nn.Module):
         ops = REDUCTION_OPERATIONS if reduction else NORMAL_OPERATIONS
         self.concat = REDUCTION_CONCAT if reduction else NORMAL_CONCAT
         self.indices = [i for i, _ in ops]
+        self.wants_relu = [name.startswith("conv") for _, name in ops]
         self.operations = nn.ModuleList()
         for i, name in ops:
             stride = 2 if (reduction and i < 2) else 1
@@ -218,9 +220,20 @@ class Cell(nn.Module):
             s1 = s2 = input_or_states
         skip = s1
         states = [self.reduce1(s1), self.reduce2(s2)]
+        relu_cache = {}
+
+        def get(idx, want_relu):
+            if not want_relu:
+                return states[idx]
+            if idx not in relu_cache:
+                relu_cache[idx] = torch.relu(states[idx])
+            return relu_cache[idx]
+
         for i in range(0, len(self.operations), 2):
-            h1 = self.operations[i](states[self.indices[i]])
-            h2 = self.operations[i + 1](states[self.indices[i + 1]])
+            h1 = self.operations[i](get(self.indices[i], self.wants_relu[i]))
+            h2 = self.operations[i + 1](
+                get(self.indices[i + 1], self.wants_relu[i + 1])
+            )
             states.append(h1 + h2)
         return torch.cat([states[i] for i in self.concat], dim=1), skip
 

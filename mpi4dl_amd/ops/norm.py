@@ -108,21 +108,18 @@ class TileBatchNorm2d(nn.BatchNorm2d):
             if group is not None:
                 dist.all_reduce(stats, group=group)
                 n *= dist.get_world_size(group=group)
-            mean64 = stats[:C] / n
-            var64 = stats[C:] / n - mean64 * mean64
-            mean = mean64.float()
-            var = var64.float()
-            if self.track_running_stats:
-                with torch.no_grad():
-                    m = self.momentum if self.momentum is not None else 0.1
-                    self.running_mean.mul_(1 - m).add_(mean, alpha=m)
-                    unbiased = var * (n / max(n - 1, 1))
-                    self.running_var.mul_(1 - m).add_(unbiased, alpha=m)
-                    self.num_batches_tracked += 1
-            invstd = torch.rsqrt(var + self.eps)
+            m = self.momentum if self.momentum is not None else 0.1
+            mv = ge.bn_finalize(
+                stats,
+                self.running_mean if self.track_running_stats else None,
+                self.running_var if self.track_running_stats else None,
+                self.num_batches_tracked if self.track_running_stats else None,
+                m, float(n), self.eps,
+            )
+            mean, invstd = mv[:C], mv[C:]
         else:
-            mean = self.running_mean.float()
-            invstd = torch.rsqrt(self.running_var.float() + self.eps)
+            mean = self.running_mean.float().contiguous()
+            invstd = torch.rsqrt(self.running_var.float() + self.eps).contiguous()
         return native_batchnorm(
             x, self.weight, self.bias, mean.contiguous(), invstd.contiguous(),
             group, float(n), self.training, self.relu,
