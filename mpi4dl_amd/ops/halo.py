@@ -180,12 +180,18 @@ class HaloExchanger:
         hh, hw = _hpair(h)
         if (hh == 0 and hw == 0) or not self.neigh:
             return
-        if xp.is_cuda:
-            return self._exchange_padded_gpu(xp, h, off, nominal)
         if nominal is None:
             H, W = xp.shape[-2] - 2 * hh, xp.shape[-1] - 2 * hw
         else:
             H, W = nominal
+        assert hh <= H and hw <= W, (
+            f"halo ({hh},{hw}) exceeds local tile {H}x{W}: the spatial "
+            "region extends past the resolution where tiling is valid - "
+            "shrink spatial_size / use fewer tiles (the reference has the "
+            "same constraint, it just corrupts silently)"
+        )
+        if xp.is_cuda:
+            return self._exchange_padded_gpu(xp, h, off, nominal)
         sends, recvs = [], []
         for d, t in self.neigh:
             if (d[0] != 0 and hh == 0) or (d[1] != 0 and hw == 0):
