@@ -218,11 +218,13 @@ torch::Tensor bn_stats(torch::Tensor x) {
 
 // 8 consecutive outputs per thread: the input row segment is read once
 // into a register sliding window (guide G13 — vectorize ALWAYS).
-template <typename T>
+template <typename T, int KK, int SS>
 __global__ void maxpool_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
                                    int32_t* __restrict__ idx, int64_t NC,
-                                   int H, int W, int OH, int OW, int k, int s,
+                                   int H, int W, int OH, int OW, int k_, int s_,
                                    int p) {
+  const int k = KK > 0 ? KK : k_;
+  const int s = KK > 0 ? SS : s_;
   const int OW8 = (OW + 7) / 8;
   const int64_t total = NC * OH * OW8;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -239,26 +241,28 @@ __global__ void maxpool_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
 #pragma unroll
     for (int e = 0; e < 8; ++e) { best[e] = -INFINITY; bidx[e] = 0; }
     const int w_lo = ow0 * s - p;
-    const int span = 8 * s + k - s;  // input row span (<= 17 for k<=3,s<=2)
-    for (int kh = 0; kh < k; ++kh) {
+    for (int kh = 0; kh < (KK > 0 ? KK : k); ++kh) {
       const int h = h0 + kh;
       if (h < 0 || h >= H) continue;
       const T* row = xp + h * W;
-      if (span <= 24) {
-        // load the span once into registers (bounds -> -inf), then pure
-        // register max per output
-        float seg[24];
+      if (KK > 0) {
+        // compile-time (k,s): span stays in registers, loops fully unroll
+        constexpr int SPAN = KK > 0 ? 8 * SS + KK - SS : 1;
+        float seg[SPAN];
 #pragma unroll
-        for (int j = 0; j < 24; ++j) {
+        for (int j = 0; j < SPAN; ++j) {
           const int w = w_lo + j;
-          seg[j] = (j < span && w >= 0 && w < W) ? (float)row[w] : -INFINITY;
+          seg[j] = (w >= 0 && w < W) ? (float)row[w] : -INFINITY;
         }
 #pragma unroll
         for (int e = 0; e < 8; ++e) {
-          const int base = e * s;
-          for (int kw = 0; kw < k; ++kw) {
-            const float v = seg[base + kw];
-            if (v > best[e]) { best[e] = v; bidx[e] = h * W + w_lo + base + kw; }
+#pragma unroll
+          for (int kw = 0; kw < KK; ++kw) {
+            const float v = seg[e * SS + kw];
+            if (v > best[e]) {
+              best[e] = v;
+              bidx[e] = h * W + w_lo + e * SS + kw;
+            }
           }
         }
       } else {
@@ -727,11 +731,20 @@ std::vector<torch::Tensor> maxpool_fwd(torch::Tensor x, int64_t k, int64_t s,
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::Half, at::ScalarType::BFloat16, x.scalar_type(),
       "maxpool_fwd", [&] {
-        hipLaunchKernelGGL(
-            (maxpool_fwd_kernel<scalar_t>), dim3(grid_for(total, 256)),
-            dim3(256), 0, stream.stream(), x.data_ptr<scalar_t>(),
-            y.data_ptr<scalar_t>(), idx.data_ptr<int32_t>(), N * C, H, W, OH,
-            OW, (int)k, (int)s, (int)p);
+        auto launch = [&](auto kernel) {
+          hipLaunchKernelGGL(kernel, dim3(grid_for(total, 256)), dim3(256), 0,
+                             stream.stream(), x.data_ptr<scalar_t>(),
+                             y.data_ptr<scalar_t>(), idx.data_ptr<int32_t>(),
+                             N * C, H, W, OH, OW, (int)k, (int)s, (int)p);
+        };
+        if (k == 3 && s == 2)
+          launch(maxpool_fwd_kernel<scalar_t, 3, 2>);
+        else if (k == 3 && s == 1)
+          launch(maxpool_fwd_kernel<scalar_t, 3, 1>);
+        else if (k == 2 && s == 2)
+          launch(maxpool_fwd_kernel<scalar_t, 2, 2>);
+        else
+          launch(maxpool_fwd_kernel<scalar_t, 0, 0>);
       });
   return {y, idx};
 }

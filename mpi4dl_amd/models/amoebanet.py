@@ -74,12 +74,14 @@ REDUCTION_CONCAT = [4, 5, 6]
 
 
 def _conv(cin, cout, k, stride=1, padding=None, ctx=None, bias=False):
+    from ..ops.conv_native import NativeConv2d
+
     if isinstance(k, int):
         k = (k, k)
     if padding is None:
         padding = ((k[0] - 1) // 2, (k[1] - 1) // 2)
     if ctx is None or (padding[0] == 0 and padding[1] == 0):
-        return nn.Conv2d(cin, cout, k, stride=stride, padding=padding, bias=bias)
+        return NativeConv2d(cin, cout, k, stride=stride, padding=padding, bias=bias)
     return HaloConv2d(cin, cout, k, stride=stride, padding=padding, bias=bias, **ctx)
 
 

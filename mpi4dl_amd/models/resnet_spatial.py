@@ -21,8 +21,10 @@ from .resnet import BasicBlockV1, BottleneckV2, Head
 
 
 def sconv(in_ch, out_ch, k, stride, ctx: Optional[dict], bias=False):
+    from ..ops.conv_native import NativeConv2d
+
     if ctx is None:
-        return nn.Conv2d(in_ch, out_ch, k, stride=stride, padding=k // 2, bias=bias)
+        return NativeConv2d(in_ch, out_ch, k, stride=stride, padding=k // 2, bias=bias)
     return HaloConv2d(in_ch, out_ch, k, stride=stride, bias=bias, **ctx)
 
 

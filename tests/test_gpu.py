@@ -251,7 +251,7 @@ def test_native_conv_autograd_matches_torch():
     torch.manual_seed(0)
     C, K, HW, k = 16, 32, 96, 3
     x = torch.randn(2, C, HW, HW, device="cuda", dtype=torch.bfloat16, requires_grad=True)
-    w = torch.randn(K, C, k, k, device="cuda", dtype=torch.float32, requires_grad=True) * 0.1
+    w = (torch.randn(K, C, k, k, device="cuda") * 0.1).requires_grad_(True)
     b = torch.randn(K, device="cuda", requires_grad=True)
     y = native_conv2d(x, w, b, 1, 1)
     x2 = x.detach().float().requires_grad_(True)
