@@ -216,6 +216,64 @@ class HaloExchanger:
         for rbuf, _, _, (rs, re, cs, ce) in recvs:
             xp[:, :, rs:re, cs:ce].copy_(rbuf)
 
+    def exchange_padded_async(self, xp: torch.Tensor, h):
+        """Start the ring exchange and return a finish() callable.
+
+        Used by the halo/compute overlap path (HaloConv2d drop mode):
+        pack+send/recv are issued now; finish() waits the transfer and
+        unpacks the ring, AFTER the caller has queued interior compute.
+        On RCCL everything is stream-ordered; the comm rides RCCL's own
+        streams so the interior conv overlaps the wire time.
+        """
+        hh, hw = _hpair(h)
+        if (hh == 0 and hw == 0) or not self.neigh:
+            return lambda: None
+        if xp.is_cuda:
+            from . import backend
+
+            ge = backend.ext()
+            pl = self._plan_gpu(xp, h, grad=False)
+            ge.halo_pack(xp, pl["sbuf"], pl["sdesc"])
+            tr = p2p.exchange(
+                [(pl["sbuf"].narrow(0, o, sz), peer, tag)
+                 for peer, tag, o, sz in pl["sends"]],
+                [(pl["rbuf"].narrow(0, o, sz), peer, tag)
+                 for peer, tag, o, sz in pl["recvs"]],
+            )
+
+            def finish():
+                tr.wait()
+                ge.halo_unpack(xp, pl["rbuf"], pl["rdesc"])
+
+            return finish
+        # CPU: same structure with torch packing
+        H, W = xp.shape[-2] - 2 * hh, xp.shape[-1] - 2 * hw
+        sends, recvs = [], []
+        for d, t in self.neigh:
+            if (d[0] != 0 and hh == 0) or (d[1] != 0 and hw == 0):
+                continue
+            peer = self.rank_of_tile(t)
+            (rs, re), (cs, ce) = send_region(d, H, W, h)
+            sends.append((xp[:, :, rs:re, cs:ce].contiguous(), peer,
+                          _DIR_IDX[_opposite(d)]))
+            (rs, re), (cs, ce) = recv_region(d, H, W, h)
+            rbuf = torch.empty(
+                (xp.shape[0], xp.shape[1], re - rs, ce - cs),
+                device=xp.device, dtype=xp.dtype,
+            )
+            recvs.append((rbuf, peer, _DIR_IDX[d], (rs, re, cs, ce)))
+        tr = p2p.exchange(
+            [(b, pp, t) for b, pp, t in sends],
+            [(b, pp, t) for b, pp, t, _ in recvs],
+        )
+
+        def finish():
+            tr.wait()
+            for rbuf, _, _, (rs, re, cs, ce) in recvs:
+                xp[:, :, rs:re, cs:ce].copy_(rbuf)
+
+        return finish
+
     # -- GPU fast path: gemscore pack/unpack + flat staging buffers ---------
 
     def _plan_gpu(self, xp, h, grad: bool, off=None, nominal=None):

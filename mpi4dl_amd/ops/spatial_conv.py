@@ -127,8 +127,60 @@ class HaloConv2d(_SpatialBase):
                 ph, pw = self.outer_pad
                 xp = F.pad(x, (pw, pw, ph, ph))
         else:
+            if (
+                self.grad_mode == "drop"
+                and self.exchanger is not None
+                and not x.is_meta
+                and self.stride in (1, (1, 1))
+                and self.halo_len not in (0, (0, 0))
+            ):
+                return self._forward_overlap(x)
             xp = halo_pad(x, self.halo_len, self.exchanger, self.grad_mode)
         return self.conv(xp)
+
+    def _forward_overlap(self, x):
+        """Halo/compute overlap (the reference's dormant Hy-Fi design,
+        spatial.py:415-826, implemented properly): the INTERIOR output
+        depends only on the local tile, so its conv is queued while the
+        ring is still on the wire; border bands are computed after the
+        ring lands and the pieces are concatenated.
+
+        Reference ('drop') gradient semantics: the received ring is a
+        constant — weight grads still include the ring pixels (identical
+        to the blocking drop path); the exact transposed-grad mode uses
+        the blocking path instead.
+        """
+        hh, hw = (
+            self.halo_len
+            if isinstance(self.halo_len, tuple)
+            else (self.halo_len, self.halo_len)
+        )
+        xp = F.pad(x, (hw, hw, hh, hh))
+        with torch.no_grad():
+            finish = self.exchanger.exchange_padded_async(xp, (hh, hw))
+        # interior while the halo is in flight
+        interior = self.conv(x)
+        with torch.no_grad():
+            finish()
+        H, W = x.shape[-2], x.shape[-1]
+        rows = []
+        if hh > 0:
+            rows.append(self.conv(xp[:, :, 0 : 3 * hh, :]))
+        mid = [interior]
+        if hw > 0:
+            mid.insert(0, self.conv(xp[:, :, hh : H + hh, 0 : 3 * hw].contiguous()))
+            mid.append(
+                self.conv(xp[:, :, hh : H + hh, W - hw : W + 2 * hw].contiguous())
+            )
+        rows_mid = torch.cat(mid, dim=3) if len(mid) > 1 else interior
+        if hh > 0:
+            out = torch.cat(
+                [rows[0], rows_mid, self.conv(xp[:, :, H - hh : H + 2 * hh, :])],
+                dim=2,
+            )
+        else:
+            out = rows_mid
+        return out
 
 
 class HaloExchangeLayer(_SpatialBase):

@@ -163,3 +163,54 @@ def test_meta_halo_shapes():
     x = torch.zeros(2, 3, 16, 16, device="meta")
     y = conv.to("meta")(x)
     assert y.shape == (2, 8, 16, 16)
+
+
+def _overlap_body(rank, world, slice_method):
+    """drop-mode overlap path must equal the blocking drop path exactly
+    (forward AND weight grads)."""
+    from mpi4dl_amd.ops.halo import TileLayout
+    from mpi4dl_amd.ops.spatial_conv import HaloConv2d
+
+    torch.manual_seed(7)
+    H = W = 16
+    full = torch.randn(2, 3, H, W)
+    layout = TileLayout(world, slice_method)
+
+    def mk(grad_mode):
+        torch.manual_seed(1)
+        return HaloConv2d(
+            3, 4, 3, num_spatial_parts=world, slice_method=slice_method,
+            spatial_local_rank=rank, grad_mode=grad_mode,
+        )
+
+    conv_overlap = mk("drop")            # takes the overlap path
+    conv_block = mk("drop")
+    conv_block.grad_mode = "drop"
+    # force blocking by pretending exact dispatch conditions fail
+    conv_block._forward_overlap = None
+    tile = layout.slice_input(full, rank).contiguous()
+
+    t1 = tile.clone().requires_grad_(True)
+    y1 = conv_overlap(t1)
+    t2 = tile.clone().requires_grad_(True)
+    from mpi4dl_amd.ops.halo import halo_pad
+
+    xp = halo_pad(t2, 1, conv_block.exchanger, "drop")
+    y2 = conv_block.conv(xp)
+    assert torch.allclose(y1, y2, atol=1e-6), (y1 - y2).abs().max()
+    g = torch.randn_like(y1)
+    y1.backward(g)
+    y2.backward(g)
+    assert torch.allclose(t1.grad, t2.grad, atol=1e-5)
+    assert torch.allclose(
+        conv_overlap.conv.weight.grad, conv_block.conv.weight.grad, atol=1e-5
+    )
+    return True
+
+
+def test_halo_overlap_square():
+    run_distributed(_overlap_body, 4, ("square",))
+
+
+def test_halo_overlap_vertical():
+    run_distributed(_overlap_body, 2, ("vertical",))
