@@ -110,6 +110,16 @@ def relu_conv_bn(cin, cout, mknorm=nn.BatchNorm2d):
     )
 
 
+def _bn_relu(mknorm, ch):
+    """BN with fused ReLU (TileBatchNorm2d relu=True); an Identity keeps
+    the Sequential indices stable where the ReLU module used to sit."""
+    bn = mknorm(ch)
+    if hasattr(bn, "relu"):
+        bn.relu = True
+        return bn
+    return nn.Sequential(bn, nn.ReLU(inplace=False))
+
+
 def make_op(name: str, c: int, stride: int, ctx, mknorm) -> nn.Module:
     if name == "none":
         return (
@@ -139,25 +149,25 @@ def make_op(name: str, c: int, stride: int, ctx, mknorm) -> nn.Module:
     if name == "conv_3x3":
         return nn.Sequential(
             nn.Conv2d(c, c // 4, 1, bias=False),
-            mknorm(c // 4),
-            nn.ReLU(inplace=False),
+            _bn_relu(mknorm, c // 4),
+            nn.Identity(),
             _conv(c // 4, c // 4, 3, stride=stride, ctx=ctx),
-            mknorm(c // 4),
-            nn.ReLU(inplace=False),
+            _bn_relu(mknorm, c // 4),
+            nn.Identity(),
             nn.Conv2d(c // 4, c, 1, bias=False),
             mknorm(c),
         )
     if name == "conv_1x7_7x1":
         return nn.Sequential(
             nn.Conv2d(c, c // 4, 1, stride=1, bias=False),
-            mknorm(c // 4),
-            nn.ReLU(inplace=False),
+            _bn_relu(mknorm, c // 4),
+            nn.Identity(),
             _conv(c // 4, c // 4, (1, 7), stride=(1, stride), padding=(0, 3), ctx=ctx),
-            mknorm(c // 4),
-            nn.ReLU(inplace=False),
+            _bn_relu(mknorm, c // 4),
+            nn.Identity(),
             _conv(c // 4, c // 4, (7, 1), stride=(stride, 1), padding=(3, 0), ctx=ctx),
-            mknorm(c // 4),
-            nn.ReLU(inplace=False),
+            _bn_relu(mknorm, c // 4),
+            nn.Identity(),
             nn.Conv2d(c // 4, c, 1, stride=1, bias=False),
             mknorm(c),
         )

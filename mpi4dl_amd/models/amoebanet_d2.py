@@ -30,6 +30,7 @@ import torch.nn as nn
 from ..ops.halo import HaloExchanger, TileLayout, halo_pad_d2
 from ..ops.plan import SpatialPlan
 from .amoebanet import (
+    _bn_relu,
     NORMAL_CONCAT,
     NORMAL_OPERATIONS,
     REDUCTION_CONCAT,
@@ -59,25 +60,25 @@ def _make_op_d2(name, c, stride, ctx, mknorm):
         if name == "conv_3x3":
             return nn.Sequential(
                 nn.Conv2d(c, c // 4, 1, bias=False),
-                mknorm(c // 4),
-                nn.ReLU(inplace=False),
+                _bn_relu(mknorm, c // 4),
+                nn.Identity(),
                 _d2conv(c // 4, c // 4, (3, 3), (1, 1), d2ctx),
-                mknorm(c // 4),
-                nn.ReLU(inplace=False),
+                _bn_relu(mknorm, c // 4),
+                nn.Identity(),
                 nn.Conv2d(c // 4, c, 1, bias=False),
                 mknorm(c),
             )
         # conv_1x7_7x1
         return nn.Sequential(
             nn.Conv2d(c, c // 4, 1, stride=1, bias=False),
-            mknorm(c // 4),
-            nn.ReLU(inplace=False),
+            _bn_relu(mknorm, c // 4),
+            nn.Identity(),
             _d2conv(c // 4, c // 4, (1, 7), (0, 3), d2ctx),
-            mknorm(c // 4),
-            nn.ReLU(inplace=False),
+            _bn_relu(mknorm, c // 4),
+            nn.Identity(),
             _d2conv(c // 4, c // 4, (7, 1), (3, 0), d2ctx),
-            mknorm(c // 4),
-            nn.ReLU(inplace=False),
+            _bn_relu(mknorm, c // 4),
+            nn.Identity(),
             nn.Conv2d(c // 4, c, 1, stride=1, bias=False),
             mknorm(c),
         )

@@ -20,6 +20,14 @@ from ..ops.spatial_conv import HaloConv2d, HaloPool2d
 from .resnet import BasicBlockV1, BottleneckV2, Head
 
 
+def _bn_relu(mknorm, ch):
+    bn = mknorm(ch)
+    if hasattr(bn, "relu"):
+        bn.relu = True
+        return bn
+    return nn.Sequential(bn, nn.ReLU(inplace=False))
+
+
 def sconv(in_ch, out_ch, k, stride, ctx: Optional[dict], bias=False):
     from ..ops.conv_native import NativeConv2d
 
@@ -35,8 +43,8 @@ class BasicBlockV1S(nn.Module):
         super().__init__()
         self.body = nn.Sequential(
             sconv(in_ch, out_ch, 3, stride, ctx),
-            mknorm(out_ch),
-            nn.ReLU(inplace=True),
+            _bn_relu(mknorm, out_ch),
+            nn.Identity(),
             sconv(out_ch, out_ch, 3, 1, ctx),
             mknorm(out_ch),
         )
@@ -59,14 +67,14 @@ class BottleneckV2S(nn.Module):
     def __init__(self, in_ch, mid_ch, stride, ctx, mknorm=nn.BatchNorm2d):
         super().__init__()
         out_ch = mid_ch * self.expansion
-        self.pre = nn.Sequential(mknorm(in_ch), nn.ReLU(inplace=True))
+        self.pre = nn.Sequential(_bn_relu(mknorm, in_ch), nn.Identity())
         self.body = nn.Sequential(
             nn.Conv2d(in_ch, mid_ch, 1, bias=False),
-            mknorm(mid_ch),
-            nn.ReLU(inplace=True),
+            _bn_relu(mknorm, mid_ch),
+            nn.Identity(),
             sconv(mid_ch, mid_ch, 3, stride, ctx),
-            mknorm(mid_ch),
-            nn.ReLU(inplace=True),
+            _bn_relu(mknorm, mid_ch),
+            nn.Identity(),
             nn.Conv2d(mid_ch, out_ch, 1, bias=False),
         )
         self.proj = None
@@ -87,8 +95,8 @@ class StemS(nn.Module):
         if image_size >= 128:
             self.ops = nn.Sequential(
                 sconv(in_ch, filters, 7, 2, ctx),
-                mknorm(filters),
-                nn.ReLU(inplace=True),
+                _bn_relu(mknorm, filters),
+                nn.Identity(),
                 HaloPool2d("max", 3, stride=2, padding=1, **(ctx or {})),
             )
         else:
