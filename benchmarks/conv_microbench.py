@@ -1,0 +1,67 @@
+#!/usr/bin/env python3
+"""Per-shape A/B of the gemscore implicit-GEMM MFMA conv vs MIOpen
+(torch F.conv2d) on the shapes the flagship model actually runs.
+
+Usage (GPU): python benchmarks/conv_microbench.py [--iters 20]
+Writes a table to stdout; used to set the NativeConv2d dispatch policy.
+"""
+import argparse
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
+
+# (C, K, H, W, kh, kw, s, ph, pw) — amoebanet-D F416 @2048 + resnet shapes
+SHAPES = [
+    (3, 104, 2048, 2048, 3, 3, 2, 1, 1),     # stem
+    (104, 208, 1024, 1024, 1, 1, 1, 0, 0),   # reduce 1x1 (big HW)
+    (208, 52, 512, 512, 1, 7, 1, 0, 3),      # 1x7
+    (208, 52, 512, 512, 7, 1, 1, 3, 0),      # 7x1
+    (416, 104, 256, 256, 1, 7, 1, 0, 3),
+    (416, 104, 256, 256, 3, 3, 1, 1, 1),     # conv_3x3 mid
+    (64, 64, 512, 512, 3, 3, 1, 1, 1),       # resnet body
+    (256, 256, 128, 128, 3, 3, 1, 1, 1),     # deeper
+]
+
+
+def timeit(fn, iters, warmup=5):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters * 1e3
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--iters", type=int, default=20)
+    ap.add_argument("--batch", type=int, default=1)
+    args = ap.parse_args()
+    from mpi4dl_amd.ops import backend
+
+    ge = backend.ext()
+    print(f"{'shape':<42} {'native':>9} {'miopen':>9} {'ratio':>6}")
+    for C, K, H, W, kh, kw, s, ph, pw in SHAPES:
+        x = torch.randn(args.batch, C, H, W, device="cuda", dtype=torch.bfloat16)
+        w = torch.randn(K, C, kh, kw, device="cuda", dtype=torch.bfloat16) * 0.05
+        tn = timeit(lambda: ge.conv_fwd(x, w, None, s, s, ph, pw), args.iters)
+        tm = timeit(
+            lambda: torch.nn.functional.conv2d(x, w, stride=s, padding=(ph, pw)),
+            args.iters,
+        )
+        # correctness spot-check
+        y = ge.conv_fwd(x, w, None, s, s, ph, pw).float()
+        ref = torch.nn.functional.conv2d(x, w, stride=s, padding=(ph, pw)).float()
+        rel = (y - ref).abs().max().item() / max(ref.abs().max().item(), 1e-3)
+        tag = f"C{C}->K{K} {H}x{W} k{kh}x{kw} s{s}"
+        print(f"{tag:<42} {tn:8.3f}ms {tm:8.3f}ms {tm/tn:5.2f}x  relerr {rel:.3g}")
+
+
+if __name__ == "__main__":
+    main()

@@ -62,9 +62,14 @@ struct ConvGeom {
 __global__ __launch_bounds__(256) void conv_fwd_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ w,
     const float* __restrict__ bias, bf16* __restrict__ out, ConvGeom g) {
-  // block -> (m_tile, n, oh, ow_tile)
+  // block -> (m_tile, n, oh, ow_tile); XCD-aware bijective swizzle so
+  // consecutive tiles spread across the 8 XCDs' L2s (guide T1)
   const int m_tiles = (g.K + BM - 1) / BM;
-  int bid = blockIdx.x;
+  const int nwg = gridDim.x;
+  const int q8 = nwg >> 3, r8 = nwg & 7;
+  const int xcd = blockIdx.x & 7, sub = blockIdx.x >> 3;
+  int bid = (xcd < r8 ? xcd * (q8 + 1) : r8 * (q8 + 1) + (xcd - r8) * q8) + sub;
+  if (bid >= nwg) bid = blockIdx.x;  // safety (never taken: mapping bijective)
   const int mt = bid % m_tiles;
   bid /= m_tiles;
   const int owt = bid % g.row_tiles;
@@ -82,10 +87,10 @@ __global__ __launch_bounds__(256) void conv_fwd_kernel(
   // LDS: A image [BM][BK] bf16 (+8 pad per row vs bank conflicts),
   //      B image (BN/16) blocks x 512 elems, double buffered.
   __shared__ __attribute__((aligned(16))) short lds[2 * (BM * (BK + 8)) +
-                                                    2 * (BN * (BK + 8))];
+                                                    2 * (BN * BK)];
   auto ldsA = [&](int buf) { return lds + buf * (BM * (BK + 8)); };
   auto ldsB = [&](int buf) {
-    return lds + 2 * (BM * (BK + 8)) + buf * (BN * (BK + 8));
+    return lds + 2 * (BM * (BK + 8)) + buf * (BN * BK);
   };
 
   f32x4 acc[4][4];
@@ -149,12 +154,16 @@ __global__ __launch_bounds__(256) void conv_fwd_kernel(
           }
         }
       }
-      // B image [BN][BK+8]: row = pixel, k contiguous -> fragment reads
-      // are plain 16-byte ds_read_b128; the (global pixel-contiguous ->
-      // LDS k-contiguous) transpose happens here as 8 b16 scatter writes.
-      short* dstB = ldsB(buf) + kk;
+      // B image (tr-read layout, semantics measured by tr16_probe):
+      // element (k, px) of 16-px block pb lives at block-local offset
+      //   (k>>3)*128 + ((k>>2)&1)*64 + (k&3)*16 + px
+      // so this 8-pixel chunk is 16 CONTIGUOUS bytes (one ds_write_b128),
+      // and the fragment read is one ds_read_b64_tr_b16 pair per k=32.
+      const int base = (px0 >> 4) * 512 + ((kk >> 3) << 7) +
+                       (((kk >> 2) & 1) << 6) + ((kk & 3) << 4) + (px0 & 15);
+      short* dstB = ldsB(buf) + base;
 #pragma unroll
-      for (int e = 0; e < 8; ++e) dstB[(px0 + e) * (BK + 8)] = v[e];
+      for (int e = 0; e < 8; ++e) dstB[e] = v[e];
     }
   };
 
@@ -178,10 +187,22 @@ __global__ __launch_bounds__(256) void conv_fwd_kernel(
       s16x8 afrag = *(const s16x8*)arow;
 #pragma unroll
       for (int nf = 0; nf < 4; ++nf) {
-        const short* brow = ldsB(buf) +
-                            (b_px0 + nf * 16 + (lane & 15)) * (BK + 8) +
-                            ((lane >> 4) << 3);
-        s16x8 bfrag = *(const s16x8*)brow;
+        const int pb = (b_px0 >> 4) + nf;
+        // per-lane tr address: group base (lane>>4)*128 + (lane&15)*4;
+        // read pair covers k = (lane>>4)*8 .. +8 for pixel col lane&15
+        __attribute__((address_space(3))) short* bbase =
+            (__attribute__((address_space(3))) short*)(ldsB(buf)) + pb * 512 +
+            ((lane >> 4) << 7) + ((lane & 15) << 2);
+        s16x4 b0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (__attribute__((address_space(3))) s16x4*)bbase);
+        s16x4 b1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (__attribute__((address_space(3))) s16x4*)(bbase + 64));
+        s16x8 bfrag;
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          bfrag[e] = b0[e];
+          bfrag[e + 4] = b1[e];
+        }
         acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             afrag, bfrag, acc[mf][nf], 0, 0, 0);
       }
