@@ -15,10 +15,18 @@ needed between comm and compute on the same stream.
 
 from __future__ import annotations
 
+import os
+import time
 from typing import List, Sequence
 
 import torch
 import torch.distributed as dist
+
+# Failure detection (the reference hangs forever on any
+# rank failure). MPI4DL_WAIT_TIMEOUT=<seconds> arms a watchdog on every
+# P2P wait; on expiry we raise with the pending peer set instead of
+# deadlocking the whole job. 0 (default) = off (zero overhead).
+_WAIT_TIMEOUT = float(os.environ.get("MPI4DL_WAIT_TIMEOUT", "0") or 0)
 
 
 def _is_nccl() -> bool:
@@ -28,10 +36,24 @@ def _is_nccl() -> bool:
 class Transfer:
     """Handle for an in-flight batched transfer."""
 
-    def __init__(self, reqs):
+    def __init__(self, reqs, peers=None):
         self.reqs = reqs
+        self.peers = peers or []
 
     def wait(self):
+        if _WAIT_TIMEOUT > 0:
+            deadline = time.monotonic() + _WAIT_TIMEOUT
+            for r in self.reqs:
+                while not r.is_completed():
+                    if time.monotonic() > deadline:
+                        raise RuntimeError(
+                            f"P2P wait exceeded {_WAIT_TIMEOUT}s "
+                            f"(peers={sorted(set(self.peers))}): a peer rank "
+                            "is stuck or dead (MPI4DL_WAIT_TIMEOUT watchdog)"
+                        )
+                    time.sleep(0.005)
+            self.reqs = []
+            return
         for r in self.reqs:
             r.wait()
         self.reqs = []
@@ -40,17 +62,17 @@ class Transfer:
 def isend_tensors(tensors: Sequence[torch.Tensor], peer: int, tag_base: int = 0) -> Transfer:
     if _is_nccl():
         ops = [dist.P2POp(dist.isend, t, peer) for t in tensors]
-        return Transfer(dist.batch_isend_irecv(ops)) if ops else Transfer([])
+        return Transfer(dist.batch_isend_irecv(ops), [peer]) if ops else Transfer([])
     reqs = [dist.isend(t, peer, tag=tag_base + i) for i, t in enumerate(tensors)]
-    return Transfer(reqs)
+    return Transfer(reqs, [peer])
 
 
 def irecv_tensors(buffers: Sequence[torch.Tensor], peer: int, tag_base: int = 0) -> Transfer:
     if _is_nccl():
         ops = [dist.P2POp(dist.irecv, t, peer) for t in buffers]
-        return Transfer(dist.batch_isend_irecv(ops)) if ops else Transfer([])
+        return Transfer(dist.batch_isend_irecv(ops), [peer]) if ops else Transfer([])
     reqs = [dist.irecv(t, peer, tag=tag_base + i) for i, t in enumerate(buffers)]
-    return Transfer(reqs)
+    return Transfer(reqs, [peer])
 
 
 def exchange(
@@ -76,13 +98,14 @@ def exchange(
             dist.P2POp(dist.irecv, b, p)
             for b, p, _ in sorted(recv_list, key=lambda x: (x[1], x[2]))
         ]
-        return Transfer(dist.batch_isend_irecv(ops)) if ops else Transfer([])
+        peers = [p for _, p, _ in send_list] + [p for _, p, _ in recv_list]
+        return Transfer(dist.batch_isend_irecv(ops), peers) if ops else Transfer([])
     reqs = []
     for t, p, tag in send_list:
         reqs.append(dist.isend(t, p, tag=tag))
     for b, p, tag in recv_list:
         reqs.append(dist.irecv(b, p, tag=tag))
-    return Transfer(reqs)
+    return Transfer(reqs, [p for _, p, _ in send_list] + [p for _, p, _ in recv_list])
 
 
 def send_tensors(tensors, peer, tag_base: int = 0):
