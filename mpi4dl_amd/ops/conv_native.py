@@ -86,7 +86,13 @@ def _dispatchable(x, conv: nn.Conv2d) -> bool:
     ph, pw = _pair(conv.padding)
     kh, kw = _pair(conv.kernel_size)
     ow = (x.shape[-1] + 2 * pw - kw) // sw + 1
-    return ow >= _MIN_OW
+    if ow < _MIN_OW:
+        return False
+    # measured policy (profiles/r01_conv_microbench.txt): the implicit-GEMM
+    # kernel beats hipBLASLt on degenerate small-C shapes (the image stem,
+    # where the im2col path dominates); mid-size shapes stay on the vendor
+    # GEMM until the staging pipeline lands (round 2).
+    return x.shape[1] <= 8
 
 
 class NativeConv2d(nn.Conv2d):
