@@ -255,3 +255,28 @@ class train_model:
         views into a FlatGrads buffer — never set_to_none."""
         self.optimizer.step()
         self.optimizer.zero_grad(set_to_none=False)
+
+    @torch.no_grad()
+    def run_eval(self, inputs, labels):
+        """Forward-only pipeline pass (reference benchmarks' optional
+        evaluation loop, e.g. benchmark_resnet_sp.py --enable-evaluation).
+        Returns (mean loss, correct, seen) on the last stage."""
+        was_training = self.models.training
+        self.models.eval()
+        self.loss_sum = 0.0
+        self.correct_sum = 0
+        self.seen = 0
+        parts_x = [None] * self.parts
+        parts_y = [None] * self.parts
+        if inputs is not None and self.first_stage:
+            parts_x = list(inputs.chunk(self.parts, dim=0))
+        if labels is not None and self.last_stage:
+            parts_y = list(labels.chunk(self.parts, dim=0))
+        for part in range(self.parts):
+            self.forward_pass(parts_x[part], parts_y[part], part)
+            self.outputs[part] = None
+            self.inputs[part] = None
+        self._drain()
+        if was_training:
+            self.models.train()
+        return self.loss_sum / max(self.parts, 1), self.correct_sum, self.seen

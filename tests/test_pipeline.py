@@ -95,3 +95,84 @@ def test_single_rank_pipeline():
     results = run_distributed(_pipeline_body, 1, (steps, batch, parts, lr))
     for e, g in zip(expected, results[0]):
         assert abs(e - g) < 1e-4
+
+
+def _dp_body(rank, world, steps, batch, parts, lr):
+    """Outer DP x LP: world 4 = 2 replicas x 2 stages; each replica
+    trains half the global batch; grads averaged across replicas."""
+    import torch.distributed as dist
+
+    from mpi4dl_amd.comm import Communicator, GradReducer
+    from mpi4dl_amd.parallel.partition import model_generator
+    from mpi4dl_amd.parallel.pipeline import train_model
+
+    comm = Communicator(split_size=2, backend="gloo")
+    model = _build()
+    gen = model_generator(model, 2, input_size=(batch // parts, 3, 32, 32))
+    gen.get_output_shapes()
+    gen.ready_model(comm.local_rank, device=torch.device("cpu"))
+    opt = torch.optim.SGD(gen.models.parameters(), lr=lr, momentum=0.9)
+    eng = train_model(gen, comm.local_rank, batch, parts, comm, optimizer=opt,
+                      device=torch.device("cpu"))
+    red = GradReducer(comm)
+    xs, ys = _data(steps, 2 * batch)  # global batch; my replica takes half
+    lo = comm.replica * batch
+    losses = []
+    for x, y in zip(xs, ys):
+        loss, _, _ = eng.run_step(x[lo : lo + batch], y[lo : lo + batch])
+        red.apply_allreduce(eng.models)
+        eng.update()
+        losses.append(loss)
+    # the two replicas' weights must be identical after sync steps
+    flat = torch.cat([p.reshape(-1) for p in eng.models.parameters()])
+    peer = (comm.rank + comm.mp_size) % comm.world_size
+    other = torch.empty_like(flat)
+    from mpi4dl_amd import p2p
+
+    if comm.rank < peer:
+        p2p.send_tensors([flat], peer, tag_base=9000)
+        p2p.recv_tensors([other], peer, tag_base=9001)
+    else:
+        p2p.recv_tensors([other], peer, tag_base=9000)
+        p2p.send_tensors([flat], peer, tag_base=9001)
+    assert torch.allclose(flat, other, atol=1e-6)
+    return losses
+
+
+def test_outer_dp_parity():
+    steps, batch, parts, lr = 2, 4, 2, 0.01
+    # serial ground truth: batch 2B with grads = mean of the two halves
+    expected = _serial_losses(steps, 2 * batch, 2 * parts, lr)
+    got = run_distributed(_dp_body, 4, (steps, batch, parts, lr))
+    # loss on each replica's last stage covers its half; mean over the two
+    # replicas' losses equals the serial full-batch loss
+    for e, g0, g1 in zip(expected, got[1], got[3]):
+        assert abs(e - (g0 + g1) / 2) < 2e-4, (expected, got[1], got[3])
+
+
+def _eval_body(rank, world):
+    from mpi4dl_amd.comm import Communicator
+    from mpi4dl_amd.parallel.partition import model_generator
+    from mpi4dl_amd.parallel.pipeline import train_model
+
+    comm = Communicator(split_size=world, backend="gloo")
+    model = _build()
+    gen = model_generator(model, world, input_size=(2, 3, 32, 32))
+    gen.get_output_shapes()
+    gen.ready_model(comm.local_rank, device=torch.device("cpu"))
+    eng = train_model(gen, comm.local_rank, 4, 2, comm, device=torch.device("cpu"))
+    torch.manual_seed(5)
+    x = torch.randn(4, 3, 32, 32)
+    y = torch.randint(0, 10, (4,))
+    loss, corr, seen = eng.run_eval(x, y)
+    if rank == world - 1:
+        model.eval()
+        with torch.no_grad():
+            ref = torch.nn.functional.cross_entropy(model(x), y)
+        assert abs(loss - float(ref)) < 1e-4, (loss, float(ref))
+        assert seen == 4
+    return True
+
+
+def test_run_eval_matches_serial():
+    run_distributed(_eval_body, 2, ())
