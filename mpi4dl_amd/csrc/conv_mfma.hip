@@ -232,6 +232,192 @@ __global__ __launch_bounds__(256) void conv_fwd_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// Forward v2 (stride-1, wide-C shapes): R passes over kernel rows with
+// weights pre-permuted to [K][R][C*S] on the host, so
+//  * the A (weight) tile stays dwordx4-contiguous per pass, and
+//  * the B (pixel) loader loads ONE row segment per (c, px-chunk) and
+//    writes its S shifted copies — S x fewer global loads than v1 and
+//    16-byte vector loads instead of u16 scalars (guide G13).
+// Accumulators persist across the R passes.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
+    const bf16* __restrict__ x, const bf16* __restrict__ w2,
+    const float* __restrict__ bias, bf16* __restrict__ out, ConvGeom g) {
+  const int m_tiles = (g.K + BM - 1) / BM;
+  const int nwg = gridDim.x;
+  const int q8 = nwg >> 3, r8 = nwg & 7;
+  const int xcd = blockIdx.x & 7, sub = blockIdx.x >> 3;
+  int bid = (xcd < r8 ? xcd * (q8 + 1) : r8 * (q8 + 1) + (xcd - r8) * q8) + sub;
+  const int mt = bid % m_tiles;
+  bid /= m_tiles;
+  const int owt = bid % g.row_tiles;
+  bid /= g.row_tiles;
+  const int oh = bid % g.OH;
+  const int n = bid / g.OH;
+
+  const int k0out = mt * BM;
+  const int ow0 = owt * BN;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = wid >> 1, wn = wid & 1;
+
+  __shared__ __attribute__((aligned(16))) short lds[2 * (BM * (BK + 8)) +
+                                                    2 * (BN * BK)];
+  auto ldsA = [&](int buf) { return lds + buf * (BM * (BK + 8)); };
+  auto ldsB = [&](int buf) {
+    return lds + 2 * (BM * (BK + 8)) + buf * (BN * BK);
+  };
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  const int64_t HW = (int64_t)g.H * g.W;
+  const int64_t in_n = (int64_t)n * g.C * HW;
+  const int CS = g.C * g.S;
+  const int KT = (CS + BK - 1) / BK;       // k'-tiles per row pass
+  const int total_it = g.R * KT;
+  const int w_lo = ow0 - g.pw;             // sw == 1
+
+  auto stage = [&](int buf, int it) {
+    const int r = it / KT;
+    const int kk0 = (it % KT) * BK;
+    const int ih = oh * g.sh - g.ph + r;
+    const bool row_ok = (ih >= 0 && ih < g.H);
+    // ---- A: w2[K][R][CS], contiguous in k' ----
+#pragma unroll
+    for (int pass = 0; pass < 2; ++pass) {
+      const int idx = pass * 256 + tid;
+      const int row = idx >> 2;
+      const int kc = (idx & 3) * 8;
+      const int kout = k0out + row;
+      short v[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (kout < g.K) {
+        const int64_t base = ((int64_t)kout * g.R + r) * CS + kk0 + kc;
+        const int rem = CS - (kk0 + kc);
+        if (rem >= 8) {
+          *(s16x8*)v = *(const s16x8*)((const short*)w2 + base);
+        } else {
+          for (int e = 0; e < 8; ++e)
+            if (e < rem) v[e] = ((const short*)w2)[base + e];
+        }
+      }
+      short* dst = ldsA(buf) + row * (BK + 8) + kc;
+      *(s16x8*)dst = *(const s16x8*)v;
+    }
+    // ---- B: one segment per (c, 8-px chunk), S shifted LDS writes ----
+    // unit u -> (ci = u / 16, pxc = u % 16); c = c_lo + ci
+    const int c_lo = kk0 / g.S;
+    const int c_hi = (kk0 + BK - 1) / g.S;  // inclusive
+    const int nci = c_hi - c_lo + 1;
+    for (int u = tid; u < nci * 16; u += 256) {
+      const int ci = u >> 4;
+      const int pxc = u & 15;
+      const int c = c_lo + ci;
+      const int px0 = pxc * 8;
+      // segment covers global cols [w_lo+px0, w_lo+px0+8+S); read from a
+      // 16-byte-ALIGNED floor (3 x 8-short loads cover offset + 15 span)
+      short seg[24];
+      const int segn = 8 + g.S - 1;
+      const int a0 = w_lo + px0;
+      const int a0a = a0 & ~7;  // aligned floor
+      int d = a0 - a0a;         // 0..7
+      if (row_ok && c < g.C) {
+        const bf16* src = x + in_n + (int64_t)c * HW + (int64_t)ih * g.W;
+        if (a0a >= 0 && a0a + 24 <= g.W) {
+          *(s16x8*)seg = *(const s16x8*)((const short*)src + a0a);
+          *(s16x8*)(seg + 8) = *(const s16x8*)((const short*)src + a0a + 8);
+          *(s16x8*)(seg + 16) = *(const s16x8*)((const short*)src + a0a + 16);
+        } else {
+          d = 0;
+          for (int j = 0; j < segn; ++j) {
+            const int col = a0 + j;
+            seg[j] = (col >= 0 && col < g.W) ? ((const short*)src)[col] : 0;
+          }
+        }
+      } else {
+        d = 0;
+#pragma unroll
+        for (int j = 0; j < 24; ++j) seg[j] = 0;
+      }
+      // write the S shifts that fall inside this k'-tile
+      for (int ss = 0; ss < g.S; ++ss) {
+        const int kprime = c * g.S + ss;
+        const int kk = kprime - kk0;
+        if (kk < 0 || kk >= BK) continue;
+        const int ow_px0 = ow0 + px0;
+        short v[8];
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          v[e] = (ow_px0 + e < g.OW) ? seg[d + ss + e] : (short)0;
+        const int base = (px0 >> 4) * 512 + ((kk >> 3) << 7) +
+                         (((kk >> 2) & 1) << 6) + ((kk & 3) << 4) + (px0 & 15);
+        *(s16x8*)(ldsB(buf) + base) = *(const s16x8*)v;
+      }
+    }
+  };
+
+  stage(0, 0);
+  __syncthreads();
+  const int a_row0 = wm * 64;
+  const int b_px0 = wn * 64;
+
+  for (int it = 0; it < total_it; ++it) {
+    const int buf = it & 1;
+    if (it + 1 < total_it) stage(buf ^ 1, it + 1);
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf) {
+      const short* arow =
+          ldsA(buf) + (a_row0 + mf * 16 + (lane & 15)) * (BK + 8) +
+          ((lane >> 4) << 3);
+      s16x8 afrag = *(const s16x8*)arow;
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        const int pb = (b_px0 >> 4) + nf;
+        __attribute__((address_space(3))) short* bbase =
+            (__attribute__((address_space(3))) short*)(ldsB(buf)) + pb * 512 +
+            ((lane >> 4) << 7) + ((lane & 15) << 2);
+        s16x4 b0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (__attribute__((address_space(3))) s16x4*)bbase);
+        s16x4 b1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (__attribute__((address_space(3))) s16x4*)(bbase + 64));
+        s16x8 bfrag;
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          bfrag[e] = b0[e];
+          bfrag[e + 4] = b1[e];
+        }
+        acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag, bfrag, acc[mf][nf], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  const int64_t out_n = ((int64_t)n * g.K) * g.OH * g.OW;
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int kout = k0out + a_row0 + mf * 16 + ((lane >> 4) << 2) + reg;
+      if (kout >= g.K) continue;
+      const float b = bias ? bias[kout] : 0.f;
+      const int64_t orow =
+          out_n + (int64_t)kout * g.OH * g.OW + (int64_t)oh * g.OW;
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        const int ow = ow0 + b_px0 + nf * 16 + (lane & 15);
+        if (ow < g.OW) out[orow + ow] = (bf16)(acc[mf][nf][reg] + b);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Backward-weight kernel: GW[Kout][CRS] += GO[Kout][px] * X[CRS][px]^T
 // Both operands are pixel-contiguous rows -> plain b128 fragments.
 // Tile 64x64, BKpx = 64 pixels per step, one (n, oh-row-chunk) per
@@ -414,6 +600,18 @@ torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor w,
   const int64_t blocks = (int64_t)m_tiles * g.row_tiles * g.OH * g.N;
   TORCH_CHECK(blocks < (1LL << 31), "grid too large");
   auto stream = at::cuda::getCurrentCUDAStream();
+  if (g.sh == 1 && g.sw == 1 && g.C * g.S >= BK) {
+    // v2: row-pass kernel with [K][R][C*S]-permuted weights
+    auto w2 = w.view({g.K, g.C, g.R, g.S})
+                  .permute({0, 2, 1, 3})
+                  .reshape({g.K, g.R, (int64_t)g.C * g.S})
+                  .contiguous();
+    hipLaunchKernelGGL(conv_fwd_v2_kernel, dim3((uint32_t)blocks), dim3(256),
+                       0, stream.stream(), (const bf16*)x.data_ptr(),
+                       (const bf16*)w2.data_ptr(), bptr,
+                       (bf16*)out.data_ptr(), g);
+    return out;
+  }
   hipLaunchKernelGGL(conv_fwd_kernel, dim3((uint32_t)blocks), dim3(256), 0,
                      stream.stream(), (const bf16*)x.data_ptr(),
                      (const bf16*)w.data_ptr(), bptr, (bf16*)out.data_ptr(),
