@@ -520,6 +520,92 @@ class GradReducer:
             )
         self.allreduce_grads(module, self.comm.outer_dp_group)
 
+    # -- bucketed overlap (DDP-equivalent, reference mp_pipeline.py:92-124
+    # wraps partitions in torch DDP; here the same flat buffer the
+    # optimizer uses is reduced bucket-by-bucket as backward fills it) --
+
+    def prepare_overlap(
+        self,
+        module: torch.nn.Module,
+        group,
+        divide_by: Optional[float] = None,
+        bucket_mb: float = 25.0,
+    ):
+        """Register post-accumulate-grad hooks that all-reduce flat-buffer
+        buckets as soon as backward has produced them. The engine toggles
+        ``sync_enabled`` so only the LAST micro-batch triggers reduction
+        (the reference relies on DDP no_sync the same way,
+        train_spatial.py:1299-1308). Call ``finish_overlap`` before the
+        optimizer step."""
+        if group is None:
+            return None
+        fg = self.flat(module)
+        state = {
+            "fg": fg,
+            "group": group,
+            "divide": divide_by
+            if divide_by is not None
+            else float(dist.get_world_size(group=group)),
+            "handles": [],
+            "sync_enabled": False,
+        }
+        # bucket boundaries over the flat buffer, walked from the END
+        # (backward fills roughly reverse parameter order)
+        elems = int(bucket_mb * 1024 * 1024 / fg.buffer.element_size())
+        offsets = []
+        off = 0
+        for p in fg.params:
+            offsets.append((p, off, p.numel()))
+            off += p.numel()
+        buckets = []
+        cur = []
+        cur_elems = 0
+        for item in reversed(offsets):
+            cur.append(item)
+            cur_elems += item[2]
+            if cur_elems >= elems:
+                buckets.append(cur)
+                cur, cur_elems = [], 0
+        if cur:
+            buckets.append(cur)
+        state["buckets"] = []
+        for b in buckets:
+            lo = min(o for _, o, _ in b)
+            hi = max(o + n for _, o, n in b)
+            binfo = {"lo": lo, "hi": hi, "pending": len(b), "total": len(b)}
+            state["buckets"].append(binfo)
+            for p, _, _ in b:
+                p.register_post_accumulate_grad_hook(
+                    self._mk_overlap_hook(state, binfo)
+                )
+        module._mpi4dl_overlap = state
+        return state
+
+    @staticmethod
+    def _mk_overlap_hook(state, binfo):
+        def hook(param):
+            if not state["sync_enabled"]:
+                return
+            binfo["pending"] -= 1
+            if binfo["pending"] == 0:
+                binfo["pending"] = binfo["total"]
+                sl = state["fg"].buffer[binfo["lo"] : binfo["hi"]]
+                state["handles"].append(
+                    dist.all_reduce(sl, group=state["group"], async_op=True)
+                )
+
+        return hook
+
+    def finish_overlap(self, module: torch.nn.Module):
+        state = getattr(module, "_mpi4dl_overlap", None)
+        if state is None:
+            return
+        for h in state["handles"]:
+            h.wait()
+        state["handles"] = []
+        state["sync_enabled"] = False
+        state["fg"].rescale_(state["divide"])
+
     def apply_allreduce_master(self, module1, module2):
         """GEMS: average the two mirrored engines' grads over the 2-rank pair
         group, in a rank-ordered sequence so both members of a pair issue the

@@ -240,7 +240,12 @@ class train_model:
             parts_y = list(labels.chunk(self.parts, dim=0))
         for part in range(self.parts):
             self.forward_pass(parts_x[part], parts_y[part], part)
+        overlap = getattr(self.models, "_mpi4dl_overlap", None)
         for part in range(self.parts):
+            if overlap is not None:
+                # only the LAST micro-batch's backward triggers the
+                # bucketed allreduce (grads accumulate until then)
+                overlap["sync_enabled"] = part == self.parts - 1
             self.backward_pass(part)
         self._drain()
         return self.loss_sum / max(self.parts, 1), self.correct_sum, self.seen

@@ -176,3 +176,39 @@ def _eval_body(rank, world):
 
 def test_run_eval_matches_serial():
     run_distributed(_eval_body, 2, ())
+
+
+def _dp_overlap_body(rank, world, steps, batch, parts, lr):
+    """Same as _dp_body but gradients reduce via the bucketed overlap
+    hooks during backward instead of one post-step allreduce."""
+    from mpi4dl_amd.comm import Communicator, GradReducer
+    from mpi4dl_amd.parallel.partition import model_generator
+    from mpi4dl_amd.parallel.pipeline import train_model
+
+    comm = Communicator(split_size=2, backend="gloo")
+    model = _build()
+    gen = model_generator(model, 2, input_size=(batch // parts, 3, 32, 32))
+    gen.get_output_shapes()
+    gen.ready_model(comm.local_rank, device=torch.device("cpu"))
+    opt = torch.optim.SGD(gen.models.parameters(), lr=lr, momentum=0.9)
+    eng = train_model(gen, comm.local_rank, batch, parts, comm, optimizer=opt,
+                      device=torch.device("cpu"))
+    red = GradReducer(comm)
+    red.prepare_overlap(gen.models, comm.outer_dp_group, bucket_mb=0.01)
+    xs, ys = _data(steps, 2 * batch)
+    lo = comm.replica * batch
+    losses = []
+    for x, y in zip(xs, ys):
+        loss, _, _ = eng.run_step(x[lo : lo + batch], y[lo : lo + batch])
+        red.finish_overlap(eng.models)
+        eng.update()
+        losses.append(loss)
+    return losses
+
+
+def test_outer_dp_bucketed_overlap_parity():
+    steps, batch, parts, lr = 2, 4, 2, 0.01
+    expected = _serial_losses(steps, 2 * batch, 2 * parts, lr)
+    got = run_distributed(_dp_overlap_body, 4, (steps, batch, parts, lr))
+    for e, g0, g1 in zip(expected, got[1], got[3]):
+        assert abs(e - (g0 + g1) / 2) < 2e-4, (expected, got[1], got[3])
