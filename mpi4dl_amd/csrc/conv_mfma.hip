@@ -319,44 +319,55 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
       const int pxc = u & 15;
       const int c = c_lo + ci;
       const int px0 = pxc * 8;
-      // segment covers global cols [w_lo+px0, w_lo+px0+8+S); read from a
-      // 16-byte-ALIGNED floor (3 x 8-short loads cover offset + 15 span)
-      short seg[24];
+      // segment covers global cols [w_lo+px0, w_lo+px0+8+S). In the
+      // d==0 regime (halo-padded convs: pw==0, plus interior-aligned
+      // tiles) every seg index is compile-time -> registers; the generic
+      // branch (plain padded convs' edges) has a runtime shift d and
+      // accepts a scratch-backed segment.
       const int segn = 8 + g.S - 1;
       const int a0 = w_lo + px0;
-      const int a0a = a0 & ~7;  // aligned floor
-      int d = a0 - a0a;         // 0..7
-      if (row_ok && c < g.C) {
-        const bf16* src = x + in_n + (int64_t)c * HW + (int64_t)ih * g.W;
-        if (a0a >= 0 && a0a + 24 <= g.W) {
-          *(s16x8*)seg = *(const s16x8*)((const short*)src + a0a);
-          *(s16x8*)(seg + 8) = *(const s16x8*)((const short*)src + a0a + 8);
-          *(s16x8*)(seg + 16) = *(const s16x8*)((const short*)src + a0a + 16);
-        } else {
-          d = 0;
+      const int ow_px0 = ow0 + px0;
+      const bool ok = row_ok && c < g.C;
+      const bf16* src = x + in_n + (int64_t)c * HW + (int64_t)ih * g.W;
+      if (ok && a0 >= 0 && (a0 & 7) == 0 && a0 + 16 <= g.W &&
+          ow_px0 + 8 <= g.OW) {
+        short seg[16];
+        *(s16x8*)seg = *(const s16x8*)((const short*)src + a0);
+        *(s16x8*)(seg + 8) = *(const s16x8*)((const short*)src + a0 + 8);
+        for (int ss = 0; ss < g.S; ++ss) {
+          const int kk = c * g.S + ss - kk0;
+          if (kk < 0 || kk >= BK) continue;
+          short v[8];
+#pragma unroll
+          for (int e = 0; e < 8; ++e) v[e] = seg[ss + e];
+          const int base = (px0 >> 4) * 512 + ((kk >> 3) << 7) +
+                           (((kk >> 2) & 1) << 6) + ((kk & 3) << 4) +
+                           (px0 & 15);
+          *(s16x8*)(ldsB(buf) + base) = *(const s16x8*)v;
+        }
+      } else {
+        short seg[16];
+        if (ok) {
           for (int j = 0; j < segn; ++j) {
             const int col = a0 + j;
             seg[j] = (col >= 0 && col < g.W) ? ((const short*)src)[col] : 0;
           }
+        } else {
+#pragma unroll
+          for (int j = 0; j < 16; ++j) seg[j] = 0;
         }
-      } else {
-        d = 0;
+        for (int ss = 0; ss < g.S; ++ss) {
+          const int kk = c * g.S + ss - kk0;
+          if (kk < 0 || kk >= BK) continue;
+          short v[8];
 #pragma unroll
-        for (int j = 0; j < 24; ++j) seg[j] = 0;
-      }
-      // write the S shifts that fall inside this k'-tile
-      for (int ss = 0; ss < g.S; ++ss) {
-        const int kprime = c * g.S + ss;
-        const int kk = kprime - kk0;
-        if (kk < 0 || kk >= BK) continue;
-        const int ow_px0 = ow0 + px0;
-        short v[8];
-#pragma unroll
-        for (int e = 0; e < 8; ++e)
-          v[e] = (ow_px0 + e < g.OW) ? seg[d + ss + e] : (short)0;
-        const int base = (px0 >> 4) * 512 + ((kk >> 3) << 7) +
-                         (((kk >> 2) & 1) << 6) + ((kk & 3) << 4) + (px0 & 15);
-        *(s16x8*)(ldsB(buf) + base) = *(const s16x8*)v;
+          for (int e = 0; e < 8; ++e)
+            v[e] = (ok && ow_px0 + e < g.OW) ? seg[ss + e] : (short)0;
+          const int base = (px0 >> 4) * 512 + ((kk >> 3) << 7) +
+                           (((kk >> 2) & 1) << 6) + ((kk & 3) << 4) +
+                           (px0 & 15);
+          *(s16x8*)(ldsB(buf) + base) = *(const s16x8*)v;
+        }
       }
     }
   };
