@@ -241,9 +241,11 @@ __global__ __launch_bounds__(256) void conv_fwd_kernel(
 // Accumulators persist across the R passes.
 // ---------------------------------------------------------------------------
 
+template <int SS>
 __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ w2,
     const float* __restrict__ bias, bf16* __restrict__ out, ConvGeom g) {
+  const int S = SS > 0 ? SS : g.S;
   const int m_tiles = (g.K + BM - 1) / BM;
   const int nwg = gridDim.x;
   const int q8 = nwg >> 3, r8 = nwg & 7;
@@ -278,7 +280,7 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
 
   const int64_t HW = (int64_t)g.H * g.W;
   const int64_t in_n = (int64_t)n * g.C * HW;
-  const int CS = g.C * g.S;
+  const int CS = g.C * S;
   const int KT = (CS + BK - 1) / BK;       // k'-tiles per row pass
   const int total_it = g.R * KT;
   const int w_lo = ow0 - g.pw;             // sw == 1
@@ -311,58 +313,81 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
     }
     // ---- B: one segment per (c, 8-px chunk), S shifted LDS writes ----
     // unit u -> (ci = u / 16, pxc = u % 16); c = c_lo + ci
-    const int c_lo = kk0 / g.S;
-    const int c_hi = (kk0 + BK - 1) / g.S;  // inclusive
+    const int c_lo = kk0 / S;
+    const int c_hi = (kk0 + BK - 1) / S;  // inclusive
     const int nci = c_hi - c_lo + 1;
     for (int u = tid; u < nci * 16; u += 256) {
       const int ci = u >> 4;
       const int pxc = u & 15;
       const int c = c_lo + ci;
       const int px0 = pxc * 8;
-      // segment covers global cols [w_lo+px0, w_lo+px0+8+S). In the
-      // d==0 regime (halo-padded convs: pw==0, plus interior-aligned
-      // tiles) every seg index is compile-time -> registers; the generic
-      // branch (plain padded convs' edges) has a runtime shift d and
-      // accepts a scratch-backed segment.
-      const int segn = 8 + g.S - 1;
+      // segment covers global cols [w_lo+px0, w_lo+px0+8+S). Aligned
+      // 24-short vector load; the runtime shift d (uniform: depends only
+      // on pw) is resolved by a scalar-branch switch so, for the
+      // templated S, every seg index is compile-time -> registers.
       const int a0 = w_lo + px0;
+      const int a0a = a0 & ~7;
+      const int d = a0 - a0a;
       const int ow_px0 = ow0 + px0;
       const bool ok = row_ok && c < g.C;
       const bf16* src = x + in_n + (int64_t)c * HW + (int64_t)ih * g.W;
-      if (ok && a0 >= 0 && (a0 & 7) == 0 && a0 + 16 <= g.W &&
-          ow_px0 + 8 <= g.OW) {
-        short seg[16];
-        *(s16x8*)seg = *(const s16x8*)((const short*)src + a0);
-        *(s16x8*)(seg + 8) = *(const s16x8*)((const short*)src + a0 + 8);
-        for (int ss = 0; ss < g.S; ++ss) {
-          const int kk = c * g.S + ss - kk0;
+      short raw[24];
+      if (ok && a0a >= 0 && a0a + 24 <= g.W) {
+        *(s16x8*)raw = *(const s16x8*)((const short*)src + a0a);
+        *(s16x8*)(raw + 8) = *(const s16x8*)((const short*)src + a0a + 8);
+        *(s16x8*)(raw + 16) = *(const s16x8*)((const short*)src + a0a + 16);
+      } else if (ok) {
+#pragma unroll
+        for (int j = 0; j < 24; ++j) {
+          const int col = a0a + j;
+          raw[j] = (col >= 0 && col < g.W) ? ((const short*)src)[col] : 0;
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < 24; ++j) raw[j] = 0;
+      }
+      // uniform-scalar switch removes the runtime shift
+      short seg[16];
+      switch (d) {
+#define SHIFT_CASE(D)                                                   \
+  case D:                                                               \
+    _Pragma("unroll") for (int j = 0; j < 16; ++j) seg[j] = raw[j + D]; \
+    break;
+        SHIFT_CASE(0)
+        SHIFT_CASE(1)
+        SHIFT_CASE(2)
+        SHIFT_CASE(3)
+        SHIFT_CASE(4)
+        SHIFT_CASE(5)
+        SHIFT_CASE(6)
+        SHIFT_CASE(7)
+#undef SHIFT_CASE
+        default:
+          break;
+      }
+      const bool full = (ow_px0 + 8 <= g.OW);
+      if (SS > 0) {
+#pragma unroll
+        for (int ss = 0; ss < (SS > 0 ? SS : 1); ++ss) {
+          const int kk = c * S + ss - kk0;
           if (kk < 0 || kk >= BK) continue;
           short v[8];
 #pragma unroll
-          for (int e = 0; e < 8; ++e) v[e] = seg[ss + e];
+          for (int e = 0; e < 8; ++e)
+            v[e] = (full || ow_px0 + e < g.OW) ? seg[ss + e] : (short)0;
           const int base = (px0 >> 4) * 512 + ((kk >> 3) << 7) +
                            (((kk >> 2) & 1) << 6) + ((kk & 3) << 4) +
                            (px0 & 15);
           *(s16x8*)(ldsB(buf) + base) = *(const s16x8*)v;
         }
       } else {
-        short seg[16];
-        if (ok) {
-          for (int j = 0; j < segn; ++j) {
-            const int col = a0 + j;
-            seg[j] = (col >= 0 && col < g.W) ? ((const short*)src)[col] : 0;
-          }
-        } else {
-#pragma unroll
-          for (int j = 0; j < 16; ++j) seg[j] = 0;
-        }
-        for (int ss = 0; ss < g.S; ++ss) {
-          const int kk = c * g.S + ss - kk0;
+        for (int ss = 0; ss < S; ++ss) {
+          const int kk = c * S + ss - kk0;
           if (kk < 0 || kk >= BK) continue;
           short v[8];
 #pragma unroll
           for (int e = 0; e < 8; ++e)
-            v[e] = (ok && ow_px0 + e < g.OW) ? seg[ss + e] : (short)0;
+            v[e] = (ow_px0 + e < g.OW) ? seg[ss + e] : (short)0;
           const int base = (px0 >> 4) * 512 + ((kk >> 3) << 7) +
                            (((kk >> 2) & 1) << 6) + ((kk & 3) << 4) +
                            (px0 & 15);
@@ -617,10 +642,20 @@ torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor w,
                   .permute({0, 2, 1, 3})
                   .reshape({g.K, g.R, (int64_t)g.C * g.S})
                   .contiguous();
-    hipLaunchKernelGGL(conv_fwd_v2_kernel, dim3((uint32_t)blocks), dim3(256),
-                       0, stream.stream(), (const bf16*)x.data_ptr(),
-                       (const bf16*)w2.data_ptr(), bptr,
-                       (bf16*)out.data_ptr(), g);
+    auto launch_v2 = [&](auto* kern) {
+      hipLaunchKernelGGL(kern, dim3((uint32_t)blocks), dim3(256), 0,
+                         stream.stream(), (const bf16*)x.data_ptr(),
+                         (const bf16*)w2.data_ptr(), bptr,
+                         (bf16*)out.data_ptr(), g);
+    };
+    if (g.S == 3)
+      launch_v2(conv_fwd_v2_kernel<3>);
+    else if (g.S == 7)
+      launch_v2(conv_fwd_v2_kernel<7>);
+    else if (g.S == 1)
+      launch_v2(conv_fwd_v2_kernel<1>);
+    else
+      launch_v2(conv_fwd_v2_kernel<0>);
     return out;
   }
   hipLaunchKernelGGL(conv_fwd_kernel, dim3((uint32_t)blocks), dim3(256), 0,
