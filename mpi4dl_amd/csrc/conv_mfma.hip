@@ -87,10 +87,10 @@ __global__ __launch_bounds__(256) void conv_fwd_kernel(
   // LDS: A image [BM][BK] bf16 (+8 pad per row vs bank conflicts),
   //      B image (BN/16) blocks x 512 elems, double buffered.
   __shared__ __attribute__((aligned(16))) short lds[2 * (BM * (BK + 8)) +
-                                                    2 * (BN * BK)];
+                                                    2 * ((BN / 16) * 520)];
   auto ldsA = [&](int buf) { return lds + buf * (BM * (BK + 8)); };
   auto ldsB = [&](int buf) {
-    return lds + 2 * (BM * (BK + 8)) + buf * (BN * BK);
+    return lds + 2 * (BM * (BK + 8)) + buf * ((BN / 16) * 520);
   };
 
   f32x4 acc[4][4];
@@ -159,7 +159,7 @@ __global__ __launch_bounds__(256) void conv_fwd_kernel(
       //   (k>>3)*128 + ((k>>2)&1)*64 + (k&3)*16 + px
       // so this 8-pixel chunk is 16 CONTIGUOUS bytes (one ds_write_b128),
       // and the fragment read is one ds_read_b64_tr_b16 pair per k=32.
-      const int base = (px0 >> 4) * 512 + ((kk >> 3) << 7) +
+      const int base = (px0 >> 4) * 520 + ((kk >> 3) << 7) +
                        (((kk >> 2) & 1) << 6) + ((kk & 3) << 4) + (px0 & 15);
       short* dstB = ldsB(buf) + base;
 #pragma unroll
@@ -191,7 +191,7 @@ __global__ __launch_bounds__(256) void conv_fwd_kernel(
         // per-lane tr address: group base (lane>>4)*128 + (lane&15)*4;
         // read pair covers k = (lane>>4)*8 .. +8 for pixel col lane&15
         __attribute__((address_space(3))) short* bbase =
-            (__attribute__((address_space(3))) short*)(ldsB(buf)) + pb * 512 +
+            (__attribute__((address_space(3))) short*)(ldsB(buf)) + pb * 520 +
             ((lane >> 4) << 7) + ((lane & 15) << 2);
         s16x4 b0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
             (__attribute__((address_space(3))) s16x4*)bbase);
@@ -266,10 +266,10 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
   const int wm = wid >> 1, wn = wid & 1;
 
   __shared__ __attribute__((aligned(16))) short lds[2 * (BM * (BK + 8)) +
-                                                    2 * (BN * BK)];
+                                                    2 * ((BN / 16) * 520)];
   auto ldsA = [&](int buf) { return lds + buf * (BM * (BK + 8)); };
   auto ldsB = [&](int buf) {
-    return lds + 2 * (BM * (BK + 8)) + buf * (BN * BK);
+    return lds + 2 * (BM * (BK + 8)) + buf * ((BN / 16) * 520);
   };
 
   f32x4 acc[4][4];
@@ -375,7 +375,7 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
 #pragma unroll
           for (int e = 0; e < 8; ++e)
             v[e] = (full || ow_px0 + e < g.OW) ? seg[ss + e] : (short)0;
-          const int base = (px0 >> 4) * 512 + ((kk >> 3) << 7) +
+          const int base = (px0 >> 4) * 520 + ((kk >> 3) << 7) +
                            (((kk >> 2) & 1) << 6) + ((kk & 3) << 4) +
                            (px0 & 15);
           *(s16x8*)(ldsB(buf) + base) = *(const s16x8*)v;
@@ -388,7 +388,7 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
 #pragma unroll
           for (int e = 0; e < 8; ++e)
             v[e] = (ow_px0 + e < g.OW) ? seg[ss + e] : (short)0;
-          const int base = (px0 >> 4) * 512 + ((kk >> 3) << 7) +
+          const int base = (px0 >> 4) * 520 + ((kk >> 3) << 7) +
                            (((kk >> 2) & 1) << 6) + ((kk & 3) << 4) +
                            (px0 & 15);
           *(s16x8*)(ldsB(buf) + base) = *(const s16x8*)v;
@@ -415,7 +415,7 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
       for (int nf = 0; nf < 4; ++nf) {
         const int pb = (b_px0 >> 4) + nf;
         __attribute__((address_space(3))) short* bbase =
-            (__attribute__((address_space(3))) short*)(ldsB(buf)) + pb * 512 +
+            (__attribute__((address_space(3))) short*)(ldsB(buf)) + pb * 520 +
             ((lane >> 4) << 7) + ((lane & 15) << 2);
         s16x4 b0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
             (__attribute__((address_space(3))) s16x4*)bbase);
