@@ -77,6 +77,10 @@ def main():
     ap.add_argument("--num-classes", type=int, default=1000)
     ap.add_argument("--slice-method", default="vertical")
     ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    ap.add_argument("--grad-mode", default="drop", choices=["drop", "exact"],
+                    help="halo backward: 'drop' = reference semantics + "
+                         "halo/compute overlap (benchmark default); "
+                         "'exact' = transposed-gradient exchange")
     args = ap.parse_args()
 
     from mpi4dl_amd.comm import Communicator, GradReducer, init_distributed
@@ -123,7 +127,7 @@ def main():
     balance = make_balance(ncells, topo["split_size"], topo["spatial_size"] if spatial else 0)
 
     plan = (
-        SpatialPlan(comm, balance, args.slice_method)
+        SpatialPlan(comm, balance, args.slice_method, grad_mode=args.grad_mode)
         if spatial
         else None
     )
@@ -148,7 +152,7 @@ def main():
     if spatial:
         eng = train_model_spatial(
             gen, comm.local_rank, B, parts, comm,
-            slice_method=args.slice_method, **eng_kw,
+            slice_method=args.slice_method, grad_mode=args.grad_mode, **eng_kw,
         )
     else:
         eng = train_model(gen, comm.local_rank, B, parts, comm, **eng_kw)

@@ -127,12 +127,20 @@ class HaloConv2d(_SpatialBase):
                 ph, pw = self.outer_pad
                 xp = F.pad(x, (pw, pw, ph, ph))
         else:
+            hh, hw = (
+                self.halo_len
+                if isinstance(self.halo_len, tuple)
+                else (self.halo_len, self.halo_len)
+            )
             if (
                 self.grad_mode == "drop"
                 and self.exchanger is not None
                 and not x.is_meta
                 and self.stride in (1, (1, 1))
-                and self.halo_len not in (0, (0, 0))
+                and (hh or hw)
+                # interior conv must be valid: tile bigger than the kernel
+                and x.shape[-2] > 2 * hh
+                and x.shape[-1] > 2 * hw
             ):
                 return self._forward_overlap(x)
             xp = halo_pad(x, self.halo_len, self.exchanger, self.grad_mode)
