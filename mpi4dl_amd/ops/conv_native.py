@@ -89,10 +89,14 @@ def _dispatchable(x, conv: nn.Conv2d) -> bool:
     if ow < _MIN_OW:
         return False
     # measured policy (profiles/r01_conv_microbench.txt): the implicit-GEMM
-    # kernel beats hipBLASLt on degenerate small-C shapes (the image stem,
-    # where the im2col path dominates); mid-size shapes stay on the vendor
-    # GEMM until the staging pipeline lands (round 2).
-    return x.shape[1] <= 8
+    # kernel beats hipBLASLt on degenerate small-C stems (im2col-dominated)
+    # and matches it on wide 1x7/7x1 rows (AmoebaNet's hot convs, 1.03x /
+    # 0.98x at 512^2); square 3x3+ shapes stay on the vendor GEMM until the
+    # staging pipeline lands (round 2).
+    if x.shape[1] <= 8:
+        return True
+    one_d = (kh == 1 and kw >= 7) or (kw == 1 and kh >= 7)
+    return one_d and ow >= 384 and x.shape[1] <= 256
 
 
 class NativeConv2d(nn.Conv2d):
