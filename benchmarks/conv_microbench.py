@@ -61,6 +61,33 @@ def main():
         rel = (y - ref).abs().max().item() / max(ref.abs().max().item(), 1e-3)
         tag = f"C{C}->K{K} {H}x{W} k{kh}x{kw} s{s}"
         print(f"{tag:<42} {tn:8.3f}ms {tm:8.3f}ms {tm/tn:5.2f}x  relerr {rel:.3g}")
+        # full autograd step (fwd+bwd) A/B
+        if s == 1:
+            from mpi4dl_amd.ops.conv_native import native_conv2d
+
+            wf = w.float().requires_grad_(True)
+            xg = x.clone().requires_grad_(True)
+            g = torch.randn_like(ref).to(torch.bfloat16)
+
+            def nat_step():
+                y = native_conv2d(xg, wf, None, s, (ph, pw))
+                y.backward(g)
+                xg.grad = None
+                wf.grad = None
+
+            x2 = x.clone().requires_grad_(True)
+            w2 = w.float().requires_grad_(True)
+
+            def ref_step():
+                y = torch.nn.functional.conv2d(x2, w2.to(torch.bfloat16),
+                                               stride=s, padding=(ph, pw))
+                y.backward(g)
+                x2.grad = None
+                w2.grad = None
+
+            tn2 = timeit(nat_step, max(args.iters // 2, 3))
+            tm2 = timeit(ref_step, max(args.iters // 2, 3))
+            print(f"{'':<42} fwd+bwd {tn2:8.3f}ms {tm2:8.3f}ms {tm2/tn2:5.2f}x")
 
 
 if __name__ == "__main__":
