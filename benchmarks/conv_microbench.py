@@ -62,7 +62,7 @@ def main():
         tag = f"C{C}->K{K} {H}x{W} k{kh}x{kw} s{s}"
         print(f"{tag:<42} {tn:8.3f}ms {tm:8.3f}ms {tm/tn:5.2f}x  relerr {rel:.3g}")
         # full autograd step (fwd+bwd) A/B
-        if s == 1:
+        if True:
             from mpi4dl_amd.ops.conv_native import native_conv2d
 
             wf = w.float().requires_grad_(True)
@@ -70,7 +70,7 @@ def main():
             g = torch.randn_like(ref).to(torch.bfloat16)
 
             def nat_step():
-                y = native_conv2d(xg, wf, None, s, (ph, pw))
+                y = native_conv2d(xg, wf, None, (s, s), (ph, pw))
                 y.backward(g)
                 xg.grad = None
                 wf.grad = None
