@@ -77,6 +77,9 @@ def main():
     ap.add_argument("--num-classes", type=int, default=1000)
     ap.add_argument("--slice-method", default="vertical")
     ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    ap.add_argument("--gems", action="store_true",
+                    help="GEMS bidirectional pipelines on top of SP "
+                         "(two mirrored engines per GPU, 2x batch/step)")
     ap.add_argument("--grad-mode", default="drop", choices=["drop", "exact"],
                     help="halo backward: 'drop' = reference semantics + "
                          "halo/compute overlap (benchmark default); "
@@ -115,6 +118,7 @@ def main():
         ENABLE_SPATIAL=spatial,
         num_spatial_parts=topo["nsp"],
         spatial_size=topo["spatial_size"] if spatial else 0,
+        ENABLE_GEMS=args.gems,
     )
 
     # model (full, cells) — probe cell count cheaply on meta
@@ -126,16 +130,32 @@ def main():
         ncells = len(build())
     balance = make_balance(ncells, topo["split_size"], topo["spatial_size"] if spatial else 0)
 
+    use_gems = args.gems and spatial and world > 1
+
+    def make_gen(plan, pos):
+        model = build(plan)
+        gen = model_generator(
+            model, topo["split_size"], (mb, 3, S, S), balance=balance
+        )
+        gen.get_output_shapes()
+        gen.ready_model(comm.get_split_rank(pos), device=device)
+        return gen
+
     plan = (
         SpatialPlan(comm, balance, args.slice_method, grad_mode=args.grad_mode)
         if spatial
         else None
     )
-    model = build(plan)
-    gen = model_generator(model, topo["split_size"], (mb, 3, S, S), balance=balance)
-    gen.get_output_shapes()
-    gen.ready_model(comm.split_rank, device=device)
-    del model  # free remote stages
+    if use_gems:
+        r = comm.rank % comm.mp_size
+        plan2 = SpatialPlan(
+            comm, balance, args.slice_method, grad_mode=args.grad_mode,
+            gems_inverse=True,
+        )
+        gen = make_gen(plan, r)
+        gen2 = make_gen(plan2, comm.mp_size - 1 - r)
+    else:
+        gen = make_gen(plan, comm.local_rank)
 
     if on_gpu:
         from mpi4dl_amd.optim import FusedSGD
@@ -149,7 +169,21 @@ def main():
         autocast_dtype=autocast_dtype,
         act_dtype=act_dtype,
     )
-    if spatial:
+    if use_gems:
+        from mpi4dl_amd.parallel.gems import train_spatial_model_master
+
+        if on_gpu:
+            from mpi4dl_amd.optim import FusedSGD
+
+            opt2 = FusedSGD(gen2.models, lr=0.01, momentum=0.9)
+        else:
+            opt2 = torch.optim.SGD(gen2.models.parameters(), lr=0.01, momentum=0.9)
+        eng = train_spatial_model_master(
+            gen, gen2, B, parts, comm, slice_method=args.slice_method,
+            grad_mode=args.grad_mode, **eng_kw,
+        )
+        eng.train_model2.optimizer = opt2
+    elif spatial:
         eng = train_model_spatial(
             gen, comm.local_rank, B, parts, comm,
             slice_method=args.slice_method, grad_mode=args.grad_mode, **eng_kw,
@@ -160,15 +194,22 @@ def main():
 
     # synthetic data of the benchmark shape
     torch.manual_seed(1234 + rank)
-    x = torch.randn(B, 3, S, S, device=device, dtype=torch.float32)
-    y = torch.randint(0, args.num_classes, (B,), device=device)
+    B_step = 2 * B if use_gems else B  # GEMS runs two replicas per step
+    x = torch.randn(B_step, 3, S, S, device=device, dtype=torch.float32)
+    y = torch.randint(0, args.num_classes, (B_step,), device=device)
 
-    def step():
-        loss, _, _ = eng.run_step(x, y)
-        if spatial or comm.dp_size > 1:
-            reducer.apply_allreduce(eng.models)
-        eng.update()
-        return loss
+    if use_gems:
+        def step():
+            loss, _, _ = eng.run_step(x, y)
+            eng.allreduce_and_update()
+            return loss
+    else:
+        def step():
+            loss, _, _ = eng.run_step(x, y)
+            if spatial or comm.dp_size > 1:
+                reducer.apply_allreduce(eng.models)
+            eng.update()
+            return loss
 
     def fence():
         if world > 1:
@@ -191,11 +232,12 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
     elapsed = float(t[0])
 
-    imgs_per_s = B * args.steps / elapsed
+    imgs_per_s = B_step * args.steps / elapsed
     if rank == 0:
         par = "single" if n == 1 else (
             f"pp{topo['split_size']}" if not spatial
-            else f"sp{topo['nsp']}+pp{topo['split_size']}"
+            else (f"sp{topo['nsp']}+gems+pp{topo['split_size']}" if use_gems
+                  else f"sp{topo['nsp']}+pp{topo['split_size']}")
         )
         print(
             json.dumps(
@@ -214,7 +256,7 @@ def main():
                     "data": "synthetic",
                     "config": {
                         "model": f"amoebanet-d(L{args.num_layers},F{args.num_filters})",
-                        "global_batch": B,
+                        "global_batch": B_step,
                         "parts": parts,
                         "seq_len": None,
                         "image_size": S,
