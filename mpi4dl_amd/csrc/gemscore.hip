@@ -370,57 +370,118 @@ __device__ __forceinline__ float avg_div(int oh, int ow, int s, int k,
   return cnt > 0 ? (float)cnt : 1.f;
 }
 
-template <typename T>
+// 8 outputs per thread; compile-time (k,s) keeps the row segment in
+// registers (KK=0 -> generic scalar path).
+template <typename T, int KK, int SS>
 __global__ void avgpool_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
                                    int64_t NC, int H, int W, int OH, int OW,
-                                   int k, int s, int p, long gr0, long gc0,
+                                   int k_, int s_, int p, long gr0, long gc0,
                                    long Hg, long Wg, int include_pad) {
-  const int64_t total = NC * OH * OW;
+  const int k = KK > 0 ? KK : k_;
+  const int s = KK > 0 ? SS : s_;
+  const int OW8 = (OW + 7) / 8;
+  const int64_t total = NC * OH * OW8;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += stride) {
-    const int ow = (int)(i % OW);
-    const int oh = (int)((i / OW) % OH);
-    const int64_t plane = i / ((int64_t)OW * OH);
+  for (int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; t < total;
+       t += stride) {
+    const int owc = (int)(t % OW8);
+    const int oh = (int)((t / OW8) % OH);
+    const int64_t plane = t / ((int64_t)OW8 * OH);
     const T* xp = x + plane * H * W;
-    const int h0 = oh * s - p, w0 = ow * s - p;
-    float acc = 0.f;
-    for (int kh = 0; kh < k; ++kh) {
+    const int ow0 = owc * 8;
+    const int h0 = oh * s - p;
+    const int w_lo = ow0 * s - p;
+    float acc[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) acc[e] = 0.f;
+    for (int kh = 0; kh < (KK > 0 ? KK : k); ++kh) {
       const int h = h0 + kh;
       if (h < 0 || h >= H) continue;
-      for (int kw = 0; kw < k; ++kw) {
-        const int w = w0 + kw;
-        if (w < 0 || w >= W) continue;
-        acc += (float)xp[h * W + w];
+      const T* row = xp + h * W;
+      if (KK > 0) {
+        constexpr int SPAN = KK > 0 ? 8 * SS + KK - SS : 1;
+        float seg[SPAN];
+#pragma unroll
+        for (int j = 0; j < SPAN; ++j) {
+          const int w = w_lo + j;
+          seg[j] = (w >= 0 && w < W) ? (float)row[w] : 0.f;
+        }
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+#pragma unroll
+          for (int kw = 0; kw < KK; ++kw) acc[e] += seg[e * SS + kw];
+      } else {
+        for (int e = 0; e < 8; ++e) {
+          const int wb = w_lo + e * s;
+          for (int kw = 0; kw < k; ++kw) {
+            const int w = wb + kw;
+            if (w >= 0 && w < W) acc[e] += (float)row[w];
+          }
+        }
       }
     }
-    y[i] = (T)(acc / avg_div(oh, ow, s, k, gr0, gc0, Hg, Wg, include_pad));
+    const int64_t obase = plane * (int64_t)OH * OW + (int64_t)oh * OW;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const int ow = ow0 + e;
+      if (ow < OW)
+        y[obase + ow] =
+            (T)(acc[e] / avg_div(oh, ow, s, k, gr0, gc0, Hg, Wg, include_pad));
+    }
   }
 }
 
-template <typename T>
+// 8 inputs per thread; interior windows (the overwhelming majority)
+// divide by the constant k*k, skipping the per-window divisor math.
+template <typename T, int KK, int SS>
 __global__ void avgpool_bwd_kernel(const T* __restrict__ go, T* __restrict__ gi,
                                    int64_t NC, int H, int W, int OH, int OW,
-                                   int k, int s, int p, long gr0, long gc0,
+                                   int k_, int s_, int p, long gr0, long gc0,
                                    long Hg, long Wg, int include_pad) {
-  const int64_t total = NC * H * W;
+  const int k = KK > 0 ? KK : k_;
+  const int s = KK > 0 ? SS : s_;
+  const int W8 = (W + 7) / 8;
+  const int64_t total = NC * H * W8;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += stride) {
-    const int w = (int)(i % W);
-    const int h = (int)((i / W) % H);
-    const int64_t plane = i / ((int64_t)W * H);
+  for (int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; t < total;
+       t += stride) {
+    const int wc = (int)(t % W8);
+    const int h = (int)((t / W8) % H);
+    const int64_t plane = t / ((int64_t)W8 * H);
+    const int w0 = wc * 8;
+    float acc[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) acc[e] = 0.f;
     int oh_lo = (h + p - k + 1 + s - 1) / s; if (oh_lo < 0) oh_lo = 0;
     int oh_hi = (h + p) / s; if (oh_hi > OH - 1) oh_hi = OH - 1;
-    int ow_lo = (w + p - k + 1 + s - 1) / s; if (ow_lo < 0) ow_lo = 0;
-    int ow_hi = (w + p) / s; if (ow_hi > OW - 1) ow_hi = OW - 1;
-    float acc = 0.f;
+    int ow_lo = (w0 + p - k + 1 + s - 1) / s; if (ow_lo < 0) ow_lo = 0;
+    int ow_hi = (w0 + 7 + p) / s; if (ow_hi > OW - 1) ow_hi = OW - 1;
     const T* gop = go + plane * OH * OW;
-    for (int oh = oh_lo; oh <= oh_hi; ++oh)
-      for (int ow = ow_lo; ow <= ow_hi; ++ow)
-        acc += (float)gop[oh * OW + ow] /
-               avg_div(oh, ow, s, k, gr0, gc0, Hg, Wg, include_pad);
-    gi[i] = (T)acc;
+    // interior iff every touched window lies fully inside the global image
+    const bool interior =
+        include_pad ||
+        ((gr0 + (long)oh_lo * s) >= 0 && (gr0 + (long)oh_hi * s + k) <= Hg &&
+         (gc0 + (long)ow_lo * s) >= 0 && (gc0 + (long)ow_hi * s + k) <= Wg);
+    const float inv_kk = 1.f / (float)(k * k);
+    for (int oh = oh_lo; oh <= oh_hi; ++oh) {
+      for (int ow = ow_lo; ow <= ow_hi; ++ow) {
+        const float g = (float)gop[oh * OW + ow];
+        const float gd =
+            interior ? g * inv_kk
+                     : g / avg_div(oh, ow, s, k, gr0, gc0, Hg, Wg, include_pad);
+        // which of my 8 inputs does window (oh, ow) cover?
+        const int wlo = ow * s - p;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const int w = w0 + e;
+          if (w >= wlo && w < wlo + k && w < W) acc[e] += gd;
+        }
+      }
+    }
+    const int64_t ibase = plane * (int64_t)H * W + (int64_t)h * W;
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+      if (w0 + e < W) gi[ibase + w0 + e] = (T)acc[e];
   }
 }
 
@@ -835,12 +896,16 @@ torch::Tensor avgpool_fwd(torch::Tensor x, int64_t k, int64_t s, int64_t p,
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::Half, at::ScalarType::BFloat16, x.scalar_type(),
       "avgpool_fwd", [&] {
-        hipLaunchKernelGGL(
-            (avgpool_fwd_kernel<scalar_t>), dim3(grid_for(total, 256)),
-            dim3(256), 0, stream.stream(), x.data_ptr<scalar_t>(),
-            y.data_ptr<scalar_t>(), N * C, H, W, OH, OW, (int)k, (int)s,
-            (int)p, (long)gr0, (long)gc0, (long)Hg, (long)Wg,
-            include_pad ? 1 : 0);
+        auto launch = [&](auto kern) {
+          hipLaunchKernelGGL(kern, dim3(grid_for(total, 256)), dim3(256), 0,
+                             stream.stream(), x.data_ptr<scalar_t>(),
+                             y.data_ptr<scalar_t>(), N * C, H, W, OH, OW,
+                             (int)k, (int)s, (int)p, (long)gr0, (long)gc0,
+                             (long)Hg, (long)Wg, include_pad ? 1 : 0);
+        };
+        if (k == 3 && s == 2) launch(avgpool_fwd_kernel<scalar_t, 3, 2>);
+        else if (k == 3 && s == 1) launch(avgpool_fwd_kernel<scalar_t, 3, 1>);
+        else launch(avgpool_fwd_kernel<scalar_t, 0, 0>);
       });
   return y;
 }
@@ -857,12 +922,17 @@ torch::Tensor avgpool_bwd(torch::Tensor go, int64_t H, int64_t W, int64_t k,
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::Half, at::ScalarType::BFloat16, go.scalar_type(),
       "avgpool_bwd", [&] {
-        hipLaunchKernelGGL(
-            (avgpool_bwd_kernel<scalar_t>), dim3(grid_for(total, 256)),
-            dim3(256), 0, stream.stream(), go.data_ptr<scalar_t>(),
-            gi.data_ptr<scalar_t>(), N * C, (int)H, (int)W, OH, OW, (int)k,
-            (int)s, (int)p, (long)gr0, (long)gc0, (long)Hg, (long)Wg,
-            include_pad ? 1 : 0);
+        auto launch = [&](auto kern) {
+          hipLaunchKernelGGL(kern, dim3(grid_for(total, 256)), dim3(256), 0,
+                             stream.stream(), go.data_ptr<scalar_t>(),
+                             gi.data_ptr<scalar_t>(), N * C, (int)H, (int)W,
+                             OH, OW, (int)k, (int)s, (int)p, (long)gr0,
+                             (long)gc0, (long)Hg, (long)Wg,
+                             include_pad ? 1 : 0);
+        };
+        if (k == 3 && s == 2) launch(avgpool_bwd_kernel<scalar_t, 3, 2>);
+        else if (k == 3 && s == 1) launch(avgpool_bwd_kernel<scalar_t, 3, 1>);
+        else launch(avgpool_bwd_kernel<scalar_t, 0, 0>);
       });
   return gi;
 }
