@@ -93,14 +93,16 @@ class BatchNormFn(torch.autograd.Function):
         go = go.contiguous()
         C = x.shape[1]
         stats = ge.bn_bwd_stats(go, x, y, mean, invstd, ctx.relu)  # fp64
-        gw = stats[C:].float()  # local sum(go*xhat)
-        gb = stats[:C].float()  # local sum(go)
+        sf = stats.float()  # ONE cast; gw/gb are views of it
+        gw = sf[C:]  # local sum(go*xhat)
+        gb = sf[:C]  # local sum(go)
         if ctx.training:
-            g_global = stats
             if ctx.group is not None:
                 g_global = stats.clone()
                 dist.all_reduce(g_global, group=ctx.group)
-            gf = g_global.float()
+                gf = g_global.float()
+            else:
+                gf = sf
             gi = ge.bn_bwd_apply(
                 go, x, y, mean, invstd, w32,
                 gf[:C].contiguous(), gf[C:].contiguous(),
