@@ -245,7 +245,13 @@ template <int SS>
 __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ w2,
     const float* __restrict__ bias, bf16* __restrict__ out, ConvGeom g) {
+  // Software-pipelined staging (guide T14): global loads for tile t+2
+  // are issued while tile t computes; the LDS write of tile t+1 sits
+  // right after the single per-iteration barrier, so its vmcnt wait is
+  // hidden under the previous iteration's MFMAs.
   const int S = SS > 0 ? SS : g.S;
+  constexpr int UNITS = (SS == 1) ? 2 : 1;  // S==1: 32 c's -> 512 units
+
   const int m_tiles = (g.K + BM - 1) / BM;
   const int nwg = gridDim.x;
   const int q8 = nwg >> 3, r8 = nwg & 7;
@@ -281,113 +287,162 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
   const int64_t HW = (int64_t)g.H * g.W;
   const int64_t in_n = (int64_t)n * g.C * HW;
   const int CS = g.C * S;
-  const int KT = (CS + BK - 1) / BK;       // k'-tiles per row pass
+  const int KT = (CS + BK - 1) / BK;
   const int total_it = g.R * KT;
-  const int w_lo = ow0 - g.pw;             // sw == 1
+  const int w_lo = ow0 - g.pw;  // sw == 1
 
-  auto stage = [&](int buf, int it) {
+  // pipelined register state
+  s16x8 aReg[2];
+  s16x8 bReg[UNITS][3];
+
+  // pure functions of (it, tid): recomputed identically at load and write
+  auto a_ok = [&](int it, int pass, int* base_out, int* rem_out) {
     const int r = it / KT;
     const int kk0 = (it % KT) * BK;
-    const int ih = oh * g.sh - g.ph + r;
-    const bool row_ok = (ih >= 0 && ih < g.H);
-    // ---- A: w2[K][R][CS], contiguous in k' ----
+    const int idx = pass * 256 + tid;
+    const int row = idx >> 2;
+    const int kc = (idx & 3) * 8;
+    const int kout = k0out + row;
+    *rem_out = CS - (kk0 + kc);
+    if (kout >= g.K) return false;
+    *base_out = 0;
+    const int64_t b = ((int64_t)kout * g.R + r) * CS + kk0 + kc;
+    *(int64_t*)base_out = 0;  // unused; base returned via pointer below
+    return b >= 0;  // always true; base recomputed at call sites
+  };
+  (void)a_ok;
+
+  auto stage_load = [&](int it) {
+    const int r = it / KT;
+    const int kk0 = (it % KT) * BK;
 #pragma unroll
     for (int pass = 0; pass < 2; ++pass) {
       const int idx = pass * 256 + tid;
       const int row = idx >> 2;
       const int kc = (idx & 3) * 8;
       const int kout = k0out + row;
-      short v[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (kout < g.K) {
-        const int64_t base = ((int64_t)kout * g.R + r) * CS + kk0 + kc;
-        const int rem = CS - (kk0 + kc);
-        if (rem >= 8) {
-          *(s16x8*)v = *(const s16x8*)((const short*)w2 + base);
-        } else {
-          for (int e = 0; e < 8; ++e)
-            if (e < rem) v[e] = ((const short*)w2)[base + e];
-        }
+      if (kout < g.K && CS - (kk0 + kc) >= 8) {
+        aReg[pass] = *(const s16x8*)((const short*)w2 +
+                                     ((int64_t)kout * g.R + r) * CS + kk0 + kc);
       }
-      short* dst = ldsA(buf) + row * (BK + 8) + kc;
-      *(s16x8*)dst = *(const s16x8*)v;
     }
-    // ---- B: one segment per (c, 8-px chunk), S shifted LDS writes ----
-    // unit u -> (ci = u / 16, pxc = u % 16); c = c_lo + ci
+    const int ih = oh * g.sh - g.ph + r;
+    const bool row_ok = (ih >= 0 && ih < g.H);
     const int c_lo = kk0 / S;
-    const int c_hi = (kk0 + BK - 1) / S;  // inclusive
-    const int nci = c_hi - c_lo + 1;
-    for (int u = tid; u < nci * 16; u += 256) {
+#pragma unroll
+    for (int u0 = 0; u0 < UNITS; ++u0) {
+      const int u = u0 * 256 + tid;
       const int ci = u >> 4;
       const int pxc = u & 15;
       const int c = c_lo + ci;
       const int px0 = pxc * 8;
-      // segment covers global cols [w_lo+px0, w_lo+px0+8+S). Aligned
-      // 24-short vector load; the runtime shift d (uniform: depends only
-      // on pw) is resolved by a scalar-branch switch so, for the
-      // templated S, every seg index is compile-time -> registers.
+      const int a0 = w_lo + px0;
+      const int a0a = a0 & ~7;  // aligned floor; shift resolved at write
+      if (row_ok && c < g.C && a0a >= 0 &&
+          a0a + 16 + ((SS > 1) ? 8 : 0) <= g.W) {
+        const short* src =
+            (const short*)(x + in_n + (int64_t)c * HW + (int64_t)ih * g.W);
+        bReg[u0][0] = *(const s16x8*)(src + a0a);
+        bReg[u0][1] = *(const s16x8*)(src + a0a + 8);
+        if (SS > 1) bReg[u0][2] = *(const s16x8*)(src + a0a + 16);
+      }
+    }
+  };
+
+  auto stage_write = [&](int it) {
+    const int buf = it & 1;
+    const int r = it / KT;
+    const int kk0 = (it % KT) * BK;
+#pragma unroll
+    for (int pass = 0; pass < 2; ++pass) {
+      const int idx = pass * 256 + tid;
+      const int row = idx >> 2;
+      const int kc = (idx & 3) * 8;
+      const int kout = k0out + row;
+      short v[8];
+      const int rem = CS - (kk0 + kc);
+      if (kout < g.K && rem >= 8) {
+        *(s16x8*)v = aReg[pass];
+      } else if (kout < g.K && rem > 0) {
+        const int64_t b = ((int64_t)kout * g.R + r) * CS + kk0 + kc;
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          v[e] = (e < rem) ? ((const short*)w2)[b + e] : (short)0;
+      } else {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) v[e] = 0;
+      }
+      *(s16x8*)(ldsA(buf) + row * (BK + 8) + kc) = *(const s16x8*)v;
+    }
+    const int ih = oh * g.sh - g.ph + r;
+    const bool row_ok = (ih >= 0 && ih < g.H);
+    const int c_lo = kk0 / S;
+#pragma unroll
+    for (int u0 = 0; u0 < UNITS; ++u0) {
+      const int u = u0 * 256 + tid;
+      const int ci = u >> 4;
+      const int pxc = u & 15;
+      const int c = c_lo + ci;
+      const int px0 = pxc * 8;
       const int a0 = w_lo + px0;
       const int a0a = a0 & ~7;
       const int d = a0 - a0a;
       const int ow_px0 = ow0 + px0;
-      const bool ok = row_ok && c < g.C;
-      const bf16* src = x + in_n + (int64_t)c * HW + (int64_t)ih * g.W;
-      short raw[24];
-      if (ok && a0a >= 0 && a0a + 24 <= g.W) {
-        *(s16x8*)raw = *(const s16x8*)((const short*)src + a0a);
-        *(s16x8*)(raw + 8) = *(const s16x8*)((const short*)src + a0a + 8);
-        *(s16x8*)(raw + 16) = *(const s16x8*)((const short*)src + a0a + 16);
-      } else if (ok) {
-#pragma unroll
-        for (int j = 0; j < 24; ++j) {
-          const int col = a0a + j;
-          raw[j] = (col >= 0 && col < g.W) ? ((const short*)src)[col] : 0;
-        }
-      } else {
-#pragma unroll
-        for (int j = 0; j < 24; ++j) raw[j] = 0;
-      }
-      // uniform-scalar switch removes the runtime shift
-      short seg[16];
-      switch (d) {
-#define SHIFT_CASE(D)                                                   \
-  case D:                                                               \
-    _Pragma("unroll") for (int j = 0; j < 16; ++j) seg[j] = raw[j + D]; \
+      const bool fast = row_ok && c < g.C && a0a >= 0 &&
+                        a0a + 16 + ((SS > 1) ? 8 : 0) <= g.W;
+      if (fast && ow_px0 + 8 <= g.OW) {
+        short raw24[24];
+        *(s16x8*)raw24 = bReg[u0][0];
+        *(s16x8*)(raw24 + 8) = bReg[u0][1];
+        if (SS > 1) *(s16x8*)(raw24 + 16) = bReg[u0][2];
+        // uniform-scalar switch turns the runtime alignment shift into
+        // compile-time indices (keeps raw/seg in registers)
+        short raw[16];
+        switch (d) {
+#define SHIFT_CASE(D)                                                      \
+  case D:                                                                  \
+    _Pragma("unroll") for (int j = 0; j < 16; ++j) raw[j] = raw24[j + D];  \
     break;
-        SHIFT_CASE(0)
-        SHIFT_CASE(1)
-        SHIFT_CASE(2)
-        SHIFT_CASE(3)
-        SHIFT_CASE(4)
-        SHIFT_CASE(5)
-        SHIFT_CASE(6)
-        SHIFT_CASE(7)
+          SHIFT_CASE(0)
+          SHIFT_CASE(1)
+          SHIFT_CASE(2)
+          SHIFT_CASE(3)
+          SHIFT_CASE(4)
+          SHIFT_CASE(5)
+          SHIFT_CASE(6)
+          SHIFT_CASE(7)
 #undef SHIFT_CASE
-        default:
-          break;
-      }
-      const bool full = (ow_px0 + 8 <= g.OW);
-      if (SS > 0) {
+          default:
+            break;
+        }
 #pragma unroll
         for (int ss = 0; ss < (SS > 0 ? SS : 1); ++ss) {
           const int kk = c * S + ss - kk0;
           if (kk < 0 || kk >= BK) continue;
           short v[8];
 #pragma unroll
-          for (int e = 0; e < 8; ++e)
-            v[e] = (full || ow_px0 + e < g.OW) ? seg[ss + e] : (short)0;
+          for (int e = 0; e < 8; ++e) v[e] = raw[ss + e];
           const int base = (px0 >> 4) * 520 + ((kk >> 3) << 7) +
                            (((kk >> 2) & 1) << 6) + ((kk & 3) << 4) +
                            (px0 & 15);
           *(s16x8*)(ldsB(buf) + base) = *(const s16x8*)v;
         }
       } else {
+        // edge path: synchronous scalar load+write (rare)
+        const short* src =
+            (const short*)(x + in_n + (int64_t)c * HW + (int64_t)ih * g.W);
+        const bool ok = row_ok && c < g.C;
         for (int ss = 0; ss < S; ++ss) {
           const int kk = c * S + ss - kk0;
           if (kk < 0 || kk >= BK) continue;
           short v[8];
 #pragma unroll
-          for (int e = 0; e < 8; ++e)
-            v[e] = (ow_px0 + e < g.OW) ? seg[ss + e] : (short)0;
+          for (int e = 0; e < 8; ++e) {
+            const int col = a0 + ss + e;
+            v[e] = (ok && ow_px0 + e < g.OW && col >= 0 && col < g.W)
+                       ? src[col]
+                       : (short)0;
+          }
           const int base = (px0 >> 4) * 520 + ((kk >> 3) << 7) +
                            (((kk >> 2) & 1) << 6) + ((kk & 3) << 4) +
                            (px0 & 15);
@@ -397,14 +452,18 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
     }
   };
 
-  stage(0, 0);
-  __syncthreads();
   const int a_row0 = wm * 64;
   const int b_px0 = wn * 64;
 
+  stage_load(0);
+  stage_write(0);
+  if (total_it > 1) stage_load(1);
+
   for (int it = 0; it < total_it; ++it) {
     const int buf = it & 1;
-    if (it + 1 < total_it) stage(buf ^ 1, it + 1);
+    __syncthreads();
+    if (it + 1 < total_it) stage_write(it + 1);
+    if (it + 2 < total_it) stage_load(it + 2);
 #pragma unroll
     for (int mf = 0; mf < 4; ++mf) {
       const short* arow =
@@ -431,7 +490,6 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
             afrag, bfrag, acc[mf][nf], 0, 0, 0);
       }
     }
-    __syncthreads();
   }
 
   const int64_t out_n = ((int64_t)n * g.K) * g.OH * g.OW;
