@@ -27,6 +27,9 @@ import os
 import sys
 import time
 
+os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")  # avoid minutes-long
+# per-shape conv solution search on first touch of big images
+
 import torch
 import torch.distributed as dist
 
