@@ -27,8 +27,17 @@ import os
 import sys
 import time
 
-os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")  # avoid minutes-long
-# per-shape conv solution search on first touch of big images
+# At >=4096^2 MIOpen's first-touch per-shape solution search takes minutes;
+# FAST (immediate) mode avoids it. Kept off at the default 2048^2 config so
+# steady-state algorithm choice is unaffected there.
+if any(a.startswith("--image-size") for a in sys.argv):
+    try:
+        _i = [i for i, a in enumerate(sys.argv) if a.startswith("--image-size")][0]
+        _v = sys.argv[_i].split("=")[1] if "=" in sys.argv[_i] else sys.argv[_i + 1]
+        if int(_v) > 2048:
+            os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
+    except (IndexError, ValueError):
+        pass
 
 import torch
 import torch.distributed as dist
