@@ -295,22 +295,6 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
   s16x8 aReg[2];
   s16x8 bReg[UNITS][3];
 
-  // pure functions of (it, tid): recomputed identically at load and write
-  auto a_ok = [&](int it, int pass, int* base_out, int* rem_out) {
-    const int r = it / KT;
-    const int kk0 = (it % KT) * BK;
-    const int idx = pass * 256 + tid;
-    const int row = idx >> 2;
-    const int kc = (idx & 3) * 8;
-    const int kout = k0out + row;
-    *rem_out = CS - (kk0 + kc);
-    if (kout >= g.K) return false;
-    *base_out = 0;
-    const int64_t b = ((int64_t)kout * g.R + r) * CS + kk0 + kc;
-    *(int64_t*)base_out = 0;  // unused; base returned via pointer below
-    return b >= 0;  // always true; base recomputed at call sites
-  };
-  (void)a_ok;
 
   auto stage_load = [&](int it) {
     const int r = it / KT;

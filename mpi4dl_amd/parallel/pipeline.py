@@ -152,13 +152,19 @@ class train_model:
                 return self.models(x)
         return self.models(x)
 
+    def _leaf(self, b):
+        """Received buffer -> autograd leaf. Without autocast the compute
+        dtype is fp32, so bf16 boundary messages are up-cast here (under
+        autocast the model handles mixed dtypes itself)."""
+        t = b.clone()
+        if self.autocast_dtype is None and t.dtype not in (torch.float32,):
+            t = t.float()
+        return t.requires_grad_(True)
+
     def receive_input(self, part: int):
         bufs = self.input_buffers[part]
         p2p.recv_tensors(bufs, self.prev_rank, tag_base=1000 + part * 16)
-        leaves = []
-        for b in bufs:
-            t = b.clone().requires_grad_(True)
-            leaves.append(t)
+        leaves = [self._leaf(b) for b in bufs]
         return leaves[0] if len(leaves) == 1 else tuple(leaves)
 
     def send_output(self, y, part: int):

@@ -285,7 +285,7 @@ class train_model_spatial(train_model):
         bufs = self.input_buffers[part]
         if self.in_layout is None:
             p2p.recv_tensors(bufs, self.prev_rank, tag_base=1000 + part * 16)
-            leaves = [b.clone().requires_grad_(True) for b in bufs]
+            leaves = [self._leaf(b) for b in bufs]
             return leaves[0] if len(leaves) == 1 else tuple(leaves)
         # gather tiles (joint / skewed / joint_dp receiving side);
         # bufs[i][j] = tensor j of the tuple from edge i
@@ -295,7 +295,7 @@ class train_model_spatial(train_model):
             for j, b in enumerate(bufs[i])
         ]
         p2p.exchange([], recvs).wait()
-        leaves = [[b.clone().requires_grad_(True) for b in eb] for eb in bufs]
+        leaves = [[self._leaf(b) for b in eb] for eb in bufs]
         self._seam_leaves = getattr(self, "_seam_leaves", [None] * self.parts)
         self._seam_leaves[part] = leaves
         # merge grid per tuple slot: cat cols within row, then rows

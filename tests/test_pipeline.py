@@ -212,3 +212,37 @@ def test_outer_dp_bucketed_overlap_parity():
     got = run_distributed(_dp_overlap_body, 4, (steps, batch, parts, lr))
     for e, g0, g1 in zip(expected, got[1], got[3]):
         assert abs(e - (g0 + g1) / 2) < 2e-4, (expected, got[1], got[3])
+
+
+def _bf16_act_body(rank, world, steps, batch, parts, lr):
+    """bf16 activation boundaries (the GPU message dtype) on gloo: the
+    pipeline must converge within bf16 tolerance of the fp32 run."""
+    from mpi4dl_amd.comm import Communicator
+    from mpi4dl_amd.parallel.partition import model_generator
+    from mpi4dl_amd.parallel.pipeline import train_model
+
+    comm = Communicator(split_size=world, backend="gloo")
+    model = _build()
+    gen = model_generator(model, world, input_size=(batch // parts, 3, 32, 32))
+    gen.get_output_shapes()
+    gen.ready_model(comm.local_rank, device=torch.device("cpu"))
+    opt = torch.optim.SGD(gen.models.parameters(), lr=lr, momentum=0.9)
+    eng = train_model(
+        gen, comm.local_rank, batch, parts, comm, optimizer=opt,
+        device=torch.device("cpu"), act_dtype=torch.bfloat16,
+    )
+    xs, ys = _data(steps, batch)
+    losses = []
+    for x, y in zip(xs, ys):
+        loss, _, _ = eng.run_step(x, y)
+        eng.update()
+        losses.append(loss)
+    return losses
+
+
+def test_bf16_activation_boundaries():
+    steps, batch, parts, lr = 3, 4, 2, 0.01
+    expected = _serial_losses(steps, batch, parts, lr)
+    got = run_distributed(_bf16_act_body, 2, (steps, batch, parts, lr))[-1]
+    for e, g in zip(expected, got):
+        assert abs(e - g) < 0.05, (expected, got)  # bf16 message tolerance
