@@ -89,6 +89,8 @@ def main():
     ap.add_argument("--num-classes", type=int, default=1000)
     ap.add_argument("--slice-method", default="vertical")
     ap.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    ap.add_argument("--schedule", default="gpipe", choices=["gpipe", "1f1b"],
+                    help="pipeline schedule (1f1b = PipeDream-flush memory profile)")
     ap.add_argument("--gems", action="store_true",
                     help="GEMS bidirectional pipelines on top of SP "
                          "(two mirrored engines per GPU, 2x batch/step)")
@@ -180,6 +182,7 @@ def main():
         device=device,
         autocast_dtype=autocast_dtype,
         act_dtype=act_dtype,
+        schedule=args.schedule,
     )
     if use_gems:
         from mpi4dl_amd.parallel.gems import train_spatial_model_master

@@ -113,6 +113,7 @@ def make_engines(args, mode):
         device=device,
         autocast_dtype=torch.bfloat16 if (args.dtype == "bf16" and on_gpu) else None,
         act_dtype=torch.bfloat16 if (args.dtype == "bf16" and on_gpu) else torch.float32,
+        schedule=getattr(args, "schedule", "gpipe"),
     )
     reducer = GradReducer(comm)
 

@@ -63,7 +63,10 @@ class train_model:
         act_dtype: Optional[torch.dtype] = None,
         autocast_dtype: Optional[torch.dtype] = None,
         device: Optional[torch.device] = None,
+        schedule: str = "gpipe",
     ):
+        assert schedule in ("gpipe", "1f1b")
+        self.schedule = schedule
         self.model_gen = model_gen
         self.comm = comm
         self.local_rank = local_rank
@@ -244,15 +247,38 @@ class train_model:
             parts_x = list(inputs.chunk(self.parts, dim=0))
         if labels is not None and self.last_stage:
             parts_y = list(labels.chunk(self.parts, dim=0))
-        for part in range(self.parts):
-            self.forward_pass(parts_x[part], parts_y[part], part)
         overlap = getattr(self.models, "_mpi4dl_overlap", None)
-        for part in range(self.parts):
+
+        def bwd(part):
             if overlap is not None:
                 # only the LAST micro-batch's backward triggers the
                 # bucketed allreduce (grads accumulate until then)
                 overlap["sync_enabled"] = part == self.parts - 1
             self.backward_pass(part)
+
+        if self.schedule == "1f1b" and self.split_size > 1:
+            # PipeDream-flush: stage r admits (split_size - split_rank)
+            # in-flight micro-batches, then strictly alternates 1F1B —
+            # same math as GPipe (all grads before the update), but peak
+            # live activations drop from `parts` to `stages - rank`.
+            warm = min(self.parts, self.split_size - self.split_rank)
+            fwd = bwd_i = 0
+            for _ in range(warm):
+                self.forward_pass(parts_x[fwd], parts_y[fwd], fwd)
+                fwd += 1
+            while fwd < self.parts:
+                bwd(bwd_i)
+                bwd_i += 1
+                self.forward_pass(parts_x[fwd], parts_y[fwd], fwd)
+                fwd += 1
+            while bwd_i < self.parts:
+                bwd(bwd_i)
+                bwd_i += 1
+        else:
+            for part in range(self.parts):
+                self.forward_pass(parts_x[part], parts_y[part], part)
+            for part in range(self.parts):
+                bwd(part)
         self._drain()
         return self.loss_sum / max(self.parts, 1), self.correct_sum, self.seen
 

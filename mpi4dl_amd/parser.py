@@ -60,6 +60,7 @@ def get_parser() -> argparse.ArgumentParser:
     p.add_argument("--weight-decay", type=float, default=1e-4)
     p.add_argument("--momentum", type=float, default=0.9)
     p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    p.add_argument("--schedule", default="gpipe", choices=["gpipe", "1f1b"])
     p.add_argument("--grad-mode", default="exact", choices=["exact", "drop"],
                    help="halo backward: exact transposed exchange or the "
                         "reference's drop semantics")

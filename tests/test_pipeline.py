@@ -246,3 +246,33 @@ def test_bf16_activation_boundaries():
     got = run_distributed(_bf16_act_body, 2, (steps, batch, parts, lr))[-1]
     for e, g in zip(expected, got):
         assert abs(e - g) < 0.05, (expected, got)  # bf16 message tolerance
+
+
+def _pipeline_1f1b_body(rank, world, steps, batch, parts, lr):
+    from mpi4dl_amd.comm import Communicator
+    from mpi4dl_amd.parallel.partition import model_generator
+    from mpi4dl_amd.parallel.pipeline import train_model
+
+    comm = Communicator(split_size=world, backend="gloo")
+    model = _build()
+    gen = model_generator(model, world, input_size=(batch // parts, 3, 32, 32))
+    gen.get_output_shapes()
+    gen.ready_model(comm.local_rank, device=torch.device("cpu"))
+    opt = torch.optim.SGD(gen.models.parameters(), lr=lr, momentum=0.9)
+    eng = train_model(gen, comm.local_rank, batch, parts, comm, optimizer=opt,
+                      device=torch.device("cpu"), schedule="1f1b")
+    xs, ys = _data(steps, batch)
+    losses = []
+    for x, y in zip(xs, ys):
+        loss, _, _ = eng.run_step(x, y)
+        eng.update()
+        losses.append(loss)
+    return losses
+
+
+def test_1f1b_schedule_parity():
+    steps, batch, parts, lr = 2, 8, 4, 0.01
+    expected = _serial_losses(steps, batch, parts, lr)
+    got = run_distributed(_pipeline_1f1b_body, 3, (steps, batch, parts, lr))[-1]
+    for e, g in zip(expected, got):
+        assert abs(e - g) < 1e-4, (expected, got)
