@@ -23,7 +23,7 @@ spatial seam handle multi-tensor activations.
 
 from __future__ import annotations
 
-from typing import List, Optional, Tuple, Union
+from typing import List, Optional
 
 import torch
 import torch.nn as nn

@@ -17,7 +17,6 @@ shape inference and for ranks that never materialise remote stages.
 
 from __future__ import annotations
 
-import math
 from typing import Optional
 
 import torch

@@ -11,13 +11,13 @@ exactly the single-GPU function (the reference's per-tile BN does not).
 
 from __future__ import annotations
 
-from typing import Callable, Optional
+from typing import Optional
 
 import torch.nn as nn
 
 from ..ops.plan import SpatialPlan
 from ..ops.spatial_conv import HaloConv2d, HaloPool2d
-from .resnet import BasicBlockV1, BottleneckV2, Head
+from .resnet import BottleneckV2, Head  # noqa: F401
 
 
 def _bn_relu(mknorm, ch):

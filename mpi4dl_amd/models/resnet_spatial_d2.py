@@ -31,7 +31,7 @@ import torch.nn as nn
 
 from ..ops.halo import halo_pad_d2
 from ..ops.plan import SpatialPlan
-from ..ops.spatial_conv import HaloConv2d, outer_pad_only
+from ..ops.spatial_conv import outer_pad_only
 from .resnet import Head
 from .resnet_spatial import BottleneckV2S, StemS
 

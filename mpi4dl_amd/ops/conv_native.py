@@ -19,7 +19,6 @@ import os
 
 import torch
 import torch.nn as nn
-import torch.nn.functional as F
 
 from . import backend
 

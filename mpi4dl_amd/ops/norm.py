@@ -18,12 +18,9 @@ the tile group.
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 import torch.distributed as dist
 import torch.nn as nn
-import torch.nn.functional as F
 
 
 class _AllReduceSum(torch.autograd.Function):

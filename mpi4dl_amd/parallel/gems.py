@@ -26,8 +26,6 @@ MI355X-native notes:
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 
 from .. import p2p
