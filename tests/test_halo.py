@@ -214,3 +214,29 @@ def test_halo_overlap_square():
 
 def test_halo_overlap_vertical():
     run_distributed(_overlap_body, 2, ("vertical",))
+
+
+def _exchange_layer_body(rank, world, h, d2):
+    """Standalone HaloExchangeLayer (C9): output = tile + ring, exact."""
+    from mpi4dl_amd.ops.halo import TileLayout
+    from mpi4dl_amd.ops.spatial_conv import HaloExchangeLayer
+
+    layout = TileLayout(world, "vertical")
+    H = W = 16
+    full = torch.arange(1.0 * 2 * 3 * H * W).reshape(2, 3, H, W)
+    tile = layout.slice_input(full, rank).contiguous()
+    layer = HaloExchangeLayer(
+        h, num_spatial_parts=world, slice_method="vertical",
+        spatial_local_rank=rank,
+    )
+    out = layer(tile)
+    r, c = layout.pos(rank)
+    tw = W // layout.cols
+    fullp = F.pad(full, (h, h, h, h))
+    expect = fullp[:, :, 0 : H + 2 * h, c * tw : c * tw + tw + 2 * h]
+    assert torch.equal(out, expect), f"rank {rank}"
+    return True
+
+
+def test_halo_exchange_layer():
+    run_distributed(_exchange_layer_body, 2, (3, False))
