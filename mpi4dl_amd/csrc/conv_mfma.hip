@@ -250,9 +250,7 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
   // right after the single per-iteration barrier, so its vmcnt wait is
   // hidden under the previous iteration's MFMAs.
   const int S = SS > 0 ? SS : g.S;
-  constexpr int BK2 = 64;           // v2 K-depth: 2 mfma k-steps/iter
-  constexpr int BSTRIDE = 1040;     // per-16px-block shorts (8*128 + pad)
-  constexpr int UNITS = (SS == 1) ? 4 : (SS == 3 ? 2 : 1);
+  constexpr int UNITS = (SS == 1) ? 2 : 1;  // S==1: 32 c's -> 512 units
 
   const int m_tiles = (g.K + BM - 1) / BM;
   const int nwg = gridDim.x;
@@ -273,11 +271,11 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
   const int wid = tid >> 6;
   const int wm = wid >> 1, wn = wid & 1;
 
-  __shared__ __attribute__((aligned(16))) short lds[2 * (BM * (BK2 + 8)) +
-                                                    2 * ((BN / 16) * BSTRIDE)];
-  auto ldsA = [&](int buf) { return lds + buf * (BM * (BK2 + 8)); };
+  __shared__ __attribute__((aligned(16))) short lds[2 * (BM * (BK + 8)) +
+                                                    2 * ((BN / 16) * 520)];
+  auto ldsA = [&](int buf) { return lds + buf * (BM * (BK + 8)); };
   auto ldsB = [&](int buf) {
-    return lds + 2 * (BM * (BK2 + 8)) + buf * ((BN / 16) * BSTRIDE);
+    return lds + 2 * (BM * (BK + 8)) + buf * ((BN / 16) * 520);
   };
 
   f32x4 acc[4][4];
@@ -289,23 +287,23 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
   const int64_t HW = (int64_t)g.H * g.W;
   const int64_t in_n = (int64_t)n * g.C * HW;
   const int CS = g.C * S;
-  const int KT = (CS + BK2 - 1) / BK2;
+  const int KT = (CS + BK - 1) / BK;
   const int total_it = g.R * KT;
   const int w_lo = ow0 - g.pw;  // sw == 1
 
   // pipelined register state
-  s16x8 aReg[4];
+  s16x8 aReg[2];
   s16x8 bReg[UNITS][3];
 
 
   auto stage_load = [&](int it) {
     const int r = it / KT;
-    const int kk0 = (it % KT) * BK2;
+    const int kk0 = (it % KT) * BK;
 #pragma unroll
-    for (int pass = 0; pass < 4; ++pass) {
+    for (int pass = 0; pass < 2; ++pass) {
       const int idx = pass * 256 + tid;
-      const int row = idx >> 3;
-      const int kc = (idx & 7) * 8;
+      const int row = idx >> 2;
+      const int kc = (idx & 3) * 8;
       const int kout = k0out + row;
       if (kout < g.K && CS - (kk0 + kc) >= 8) {
         aReg[pass] = *(const s16x8*)((const short*)w2 +
@@ -338,12 +336,12 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
   auto stage_write = [&](int it) {
     const int buf = it & 1;
     const int r = it / KT;
-    const int kk0 = (it % KT) * BK2;
+    const int kk0 = (it % KT) * BK;
 #pragma unroll
-    for (int pass = 0; pass < 4; ++pass) {
+    for (int pass = 0; pass < 2; ++pass) {
       const int idx = pass * 256 + tid;
-      const int row = idx >> 3;
-      const int kc = (idx & 7) * 8;
+      const int row = idx >> 2;
+      const int kc = (idx & 3) * 8;
       const int kout = k0out + row;
       short v[8];
       const int rem = CS - (kk0 + kc);
@@ -358,7 +356,7 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
 #pragma unroll
         for (int e = 0; e < 8; ++e) v[e] = 0;
       }
-      *(s16x8*)(ldsA(buf) + row * (BK2 + 8) + kc) = *(const s16x8*)v;
+      *(s16x8*)(ldsA(buf) + row * (BK + 8) + kc) = *(const s16x8*)v;
     }
     const int ih = oh * g.sh - g.ph + r;
     const bool row_ok = (ih >= 0 && ih < g.H);
@@ -404,11 +402,11 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
 #pragma unroll
         for (int ss = 0; ss < (SS > 0 ? SS : 1); ++ss) {
           const int kk = c * S + ss - kk0;
-          if (kk < 0 || kk >= BK2) continue;
+          if (kk < 0 || kk >= BK) continue;
           short v[8];
 #pragma unroll
           for (int e = 0; e < 8; ++e) v[e] = raw[ss + e];
-          const int base = (px0 >> 4) * BSTRIDE + ((kk >> 3) << 7) +
+          const int base = (px0 >> 4) * 520 + ((kk >> 3) << 7) +
                            (((kk >> 2) & 1) << 6) + ((kk & 3) << 4) +
                            (px0 & 15);
           *(s16x8*)(ldsB(buf) + base) = *(const s16x8*)v;
@@ -420,7 +418,7 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
         const bool ok = row_ok && c < g.C;
         for (int ss = 0; ss < S; ++ss) {
           const int kk = c * S + ss - kk0;
-          if (kk < 0 || kk >= BK2) continue;
+          if (kk < 0 || kk >= BK) continue;
           short v[8];
 #pragma unroll
           for (int e = 0; e < 8; ++e) {
@@ -429,7 +427,7 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
                        ? src[col]
                        : (short)0;
           }
-          const int base = (px0 >> 4) * BSTRIDE + ((kk >> 3) << 7) +
+          const int base = (px0 >> 4) * 520 + ((kk >> 3) << 7) +
                            (((kk >> 2) & 1) << 6) + ((kk & 3) << 4) +
                            (px0 & 15);
           *(s16x8*)(ldsB(buf) + base) = *(const s16x8*)v;
@@ -451,33 +449,29 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
     if (it + 1 < total_it) stage_write(it + 1);
     if (it + 2 < total_it) stage_load(it + 2);
 #pragma unroll
-    for (int kh = 0; kh < 2; ++kh) {
+    for (int mf = 0; mf < 4; ++mf) {
+      const short* arow =
+          ldsA(buf) + (a_row0 + mf * 16 + (lane & 15)) * (BK + 8) +
+          ((lane >> 4) << 3);
+      s16x8 afrag = *(const s16x8*)arow;
 #pragma unroll
-      for (int mf = 0; mf < 4; ++mf) {
-        const short* arow =
-            ldsA(buf) + (a_row0 + mf * 16 + (lane & 15)) * (BK2 + 8) +
-            kh * 32 + ((lane >> 4) << 3);
-        s16x8 afrag = *(const s16x8*)arow;
+      for (int nf = 0; nf < 4; ++nf) {
+        const int pb = (b_px0 >> 4) + nf;
+        __attribute__((address_space(3))) short* bbase =
+            (__attribute__((address_space(3))) short*)(ldsB(buf)) + pb * 520 +
+            ((lane >> 4) << 7) + ((lane & 15) << 2);
+        s16x4 b0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (__attribute__((address_space(3))) s16x4*)bbase);
+        s16x4 b1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (__attribute__((address_space(3))) s16x4*)(bbase + 64));
+        s16x8 bfrag;
 #pragma unroll
-        for (int nf = 0; nf < 4; ++nf) {
-          const int pb = (b_px0 >> 4) + nf;
-          __attribute__((address_space(3))) short* bbase =
-              (__attribute__((address_space(3))) short*)(ldsB(buf)) +
-              pb * BSTRIDE + kh * 512 + ((lane >> 4) << 7) +
-              ((lane & 15) << 2);
-          s16x4 b0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-              (__attribute__((address_space(3))) s16x4*)bbase);
-          s16x4 b1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-              (__attribute__((address_space(3))) s16x4*)(bbase + 64));
-          s16x8 bfrag;
-#pragma unroll
-          for (int e = 0; e < 4; ++e) {
-            bfrag[e] = b0[e];
-            bfrag[e + 4] = b1[e];
-          }
-          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afrag, bfrag, acc[mf][nf], 0, 0, 0);
+        for (int e = 0; e < 4; ++e) {
+          bfrag[e] = b0[e];
+          bfrag[e + 4] = b1[e];
         }
+        acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag, bfrag, acc[mf][nf], 0, 0, 0);
       }
     }
   }
