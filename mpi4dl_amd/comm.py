@@ -491,11 +491,14 @@ class GradReducer:
 
     def allreduce_grads(self, module: torch.nn.Module, group, divide_by: float = None):
         """Average grads over ``group`` (no-op for group=None / size 1)."""
+        from .utils import GLOBAL_TIMER
+
         fg = self.flat(module)
         if group is None or fg.buffer.numel() == 0:
             return
         n = dist.get_world_size(group=group)
-        dist.all_reduce(fg.buffer, group=group)
+        with GLOBAL_TIMER.phase("grad/allreduce"):
+            dist.all_reduce(fg.buffer, group=group)
         fg.rescale_(divide_by if divide_by is not None else float(n))
 
     def apply_allreduce(self, module: torch.nn.Module):

@@ -177,6 +177,8 @@ class HaloExchanger:
         region. Blocks until the ring is filled (async overlap is
         handled by HaloConv2d's interior/boundary split, not here).
         """
+        from ..utils import GLOBAL_TIMER  # noqa: F811 (cheap; cached import)
+
         hh, hw = _hpair(h)
         if (hh == 0 and hw == 0) or not self.neigh:
             return
@@ -191,7 +193,8 @@ class HaloExchanger:
             "same constraint, it just corrupts silently)"
         )
         if xp.is_cuda:
-            return self._exchange_padded_gpu(xp, h, off, nominal)
+            with GLOBAL_TIMER.phase("halo/exchange"):
+                return self._exchange_padded_gpu(xp, h, off, nominal)
         sends, recvs = [], []
         for d, t in self.neigh:
             if (d[0] != 0 and hh == 0) or (d[1] != 0 and hw == 0):

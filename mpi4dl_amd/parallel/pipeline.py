@@ -29,6 +29,7 @@ import torch.nn as nn
 
 from .. import p2p
 from ..comm import Communicator, current_device
+from ..utils import GLOBAL_TIMER
 
 log = logging.getLogger(__name__)
 
@@ -166,7 +167,8 @@ class train_model:
 
     def receive_input(self, part: int):
         bufs = self.input_buffers[part]
-        p2p.recv_tensors(bufs, self.prev_rank, tag_base=1000 + part * 16)
+        with GLOBAL_TIMER.phase("pipeline/recv_act"):
+            p2p.recv_tensors(bufs, self.prev_rank, tag_base=1000 + part * 16)
         leaves = [self._leaf(b) for b in bufs]
         return leaves[0] if len(leaves) == 1 else tuple(leaves)
 
@@ -182,7 +184,8 @@ class train_model:
         else:
             x = self.receive_input(part)
         self.inputs[part] = x
-        y = self._run_stage(x)
+        with GLOBAL_TIMER.phase("pipeline/forward"):
+            y = self._run_stage(x)
         self.outputs[part] = y
         if self.last_stage:
             yl = data_y.to(self.device, non_blocking=True)
@@ -219,12 +222,15 @@ class train_model:
         """One micro-batch backward (mp_pipeline.py:475-507)."""
         y = self.outputs[part]
         if self.last_stage:
-            y.backward()
+            with GLOBAL_TIMER.phase("pipeline/backward"):
+                y.backward()
         else:
-            gbufs = self.receive_output_grad(part)
+            with GLOBAL_TIMER.phase("pipeline/recv_grad"):
+                gbufs = self.receive_output_grad(part)
             ys = list(y) if isinstance(y, tuple) else [y]
             grads = [g.to(t.dtype) for g, t in zip(gbufs, ys)]
-            torch.autograd.backward(ys, grads)
+            with GLOBAL_TIMER.phase("pipeline/backward"):
+                torch.autograd.backward(ys, grads)
         if not self.first_stage:
             self.send_input_grad(part)
         # free graph state for this part
