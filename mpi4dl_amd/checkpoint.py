@@ -158,3 +158,12 @@ def consolidate(path: str) -> dict:
         for k, v in shard["model"].items():
             full[k] = v
     return full
+
+
+def load_from_consolidated(full_state: dict, module: torch.nn.Module):
+    """Load a consolidated (cell-global) state_dict into a LOCAL stage of
+    ANY partitioning: nn.Sequential slices keep global child names, so the
+    stage's keys select its subset directly — this is the re-shard path
+    (train on split=4, consolidate, resume on split=2)."""
+    subset = {k: full_state[k] for k in module.state_dict().keys()}
+    module.load_state_dict(subset)

@@ -102,3 +102,40 @@ def test_consolidate(tmp_path):
     )
     for k in ref_sd:
         assert torch.equal(full[k], ref_sd[k]), k
+
+
+def _reshard_body(rank, world, tmpdir):
+    """Resume shards saved at split=3 into a split=2 run via consolidate."""
+    from mpi4dl_amd.checkpoint import consolidate, load_from_consolidated
+    from mpi4dl_amd.comm import Communicator
+    from mpi4dl_amd.models.resnet import get_resnet_v1
+    from mpi4dl_amd.parallel.partition import model_generator
+    from mpi4dl_amd.parallel.pipeline import train_model
+
+    comm = Communicator(split_size=world, backend="gloo")
+    full = consolidate(tmpdir)  # written by the 3-way run
+    torch.manual_seed(99)  # deliberately different init
+    model = get_resnet_v1((2, 3, IMG, IMG), NCLS, n=1, num_filters=8)
+    gen = model_generator(model, world, input_size=(2, 3, IMG, IMG))
+    gen.get_output_shapes()
+    gen.ready_model(comm.local_rank, device=torch.device("cpu"))
+    load_from_consolidated(full, gen.models)
+    eng = train_model(gen, comm.local_rank, 2, 1, comm, device=torch.device("cpu"))
+    torch.manual_seed(5)
+    x = torch.randn(2, 3, IMG, IMG)
+    y = torch.randint(0, NCLS, (2,))
+    loss, _, _ = eng.run_eval(x, y)
+    if rank == world - 1:
+        # reference: the consolidated full model on one process
+        ref = get_resnet_v1((2, 3, IMG, IMG), NCLS, n=1, num_filters=8)
+        ref.load_state_dict(full)
+        ref.eval()
+        with torch.no_grad():
+            rl = torch.nn.functional.cross_entropy(ref(x), y)
+        assert abs(loss - float(rl)) < 1e-4, (loss, float(rl))
+    return True
+
+
+def test_reshard_via_consolidate(tmp_path):
+    run_distributed(_consolidate_body, 3, (str(tmp_path),))
+    run_distributed(_reshard_body, 2, (str(tmp_path),))
