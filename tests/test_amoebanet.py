@@ -145,3 +145,52 @@ def test_amoebanet_sp_parity():
     got = run_distributed(_sp_body, 3, (steps, batch, parts, lr))[-1]
     for e, g in zip(expected, got):
         assert abs(e - g) < 2e-4, (expected, got)
+
+
+def _sp_square_body(rank, world, steps, batch, parts, lr):
+    from mpi4dl_amd.comm import Communicator, GradReducer
+    from mpi4dl_amd.models.amoebanet import amoebanetd as build
+    from mpi4dl_amd.ops.plan import SpatialPlan
+    from mpi4dl_amd.parallel.partition import model_generator
+    from mpi4dl_amd.parallel.spatial import train_model_spatial
+
+    split = 2
+    comm = Communicator(
+        split_size=split, ENABLE_SPATIAL=True, num_spatial_parts=4,
+        spatial_size=1, backend="gloo",
+    )
+    probe = _build()
+    balance = [4, len(probe) - 4]
+    plan = SpatialPlan(comm, balance, "square")
+    torch.manual_seed(0)
+    model = build(NCLS, LAYERS, FILTERS, plan=plan)
+    gen = model_generator(
+        model, split, input_size=(batch // parts, 3, IMG, IMG), balance=balance
+    )
+    gen.get_output_shapes()
+    gen.ready_model(comm.split_rank, device=torch.device("cpu"))
+    opt = torch.optim.SGD(gen.models.parameters(), lr=lr, momentum=0.9)
+    eng = train_model_spatial(
+        gen, comm.local_rank, batch, parts, comm,
+        slice_method="square", optimizer=opt, device=torch.device("cpu"),
+    )
+    red = GradReducer(comm)
+    torch.manual_seed(42)
+    losses = []
+    for _ in range(steps):
+        x = torch.randn(batch, 3, IMG, IMG)
+        y = torch.randint(0, NCLS, (batch,))
+        loss, _, _ = eng.run_step(x, y)
+        red.apply_allreduce(eng.models)
+        eng.update()
+        losses.append(loss)
+    return losses
+
+
+def test_amoebanet_sp_square_parity():
+    steps, batch, parts, lr = 2, 2, 1, 0.01
+    expected = _serial_losses(steps, batch, parts, lr)
+    # 4 square tiles (8-neighbour halos incl. corners) + 1 LP rank
+    got = run_distributed(_sp_square_body, 5, (steps, batch, parts, lr))[-1]
+    for e, g in zip(expected, got):
+        assert abs(e - g) < 1e-3, (expected, got)  # fp32 reduction-order noise

@@ -143,3 +143,13 @@ def test_sp_local_dp():
     # their mean must match the serial loss
     for e, g0, g1 in zip(expected, got[2], got[3]):
         assert abs(e - (g0 + g1) / 2) < 2e-4, (expected, got[2], got[3])
+
+
+def test_sp_horizontal_2tiles_plus_lp():
+    steps, batch, parts, lr = 2, 2, 1, 0.01
+    expected = _serial_losses(steps, batch, parts, lr)
+    got = run_distributed(
+        _spatial_body, 3, (steps, batch, parts, lr, "horizontal", 2, 1, 2, 1)
+    )[-1]
+    for e, g in zip(expected, got):
+        assert abs(e - g) < 2e-4, (expected, got)
