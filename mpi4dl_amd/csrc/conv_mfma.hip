@@ -87,10 +87,10 @@ __global__ __launch_bounds__(256) void conv_fwd_kernel(
   // LDS: A image [BM][BK] bf16 (+8 pad per row vs bank conflicts),
   //      B image (BN/16) blocks x 512 elems, double buffered.
   __shared__ __attribute__((aligned(16))) short lds[2 * (BM * (BK + 8)) +
-                                                    2 * ((BN / 16) * 528)];
+                                                    2 * ((BN / 16) * 520)];
   auto ldsA = [&](int buf) { return lds + buf * (BM * (BK + 8)); };
   auto ldsB = [&](int buf) {
-    return lds + 2 * (BM * (BK + 8)) + buf * ((BN / 16) * 528);
+    return lds + 2 * (BM * (BK + 8)) + buf * ((BN / 16) * 520);
   };
 
   f32x4 acc[4][4];
@@ -159,7 +159,7 @@ __global__ __launch_bounds__(256) void conv_fwd_kernel(
       //   (k>>3)*128 + ((k>>2)&1)*64 + (k&3)*16 + px
       // so this 8-pixel chunk is 16 CONTIGUOUS bytes (one ds_write_b128),
       // and the fragment read is one ds_read_b64_tr_b16 pair per k=32.
-      const int base = (px0 >> 4) * 528 + ((kk >> 3) << 7) +
+      const int base = (px0 >> 4) * 520 + ((kk >> 3) << 7) +
                        (((kk >> 2) & 1) << 6) + ((kk & 3) << 4) + (px0 & 15);
       short* dstB = ldsB(buf) + base;
 #pragma unroll
@@ -191,7 +191,7 @@ __global__ __launch_bounds__(256) void conv_fwd_kernel(
         // per-lane tr address: group base (lane>>4)*128 + (lane&15)*4;
         // read pair covers k = (lane>>4)*8 .. +8 for pixel col lane&15
         __attribute__((address_space(3))) short* bbase =
-            (__attribute__((address_space(3))) short*)(ldsB(buf)) + pb * 528 +
+            (__attribute__((address_space(3))) short*)(ldsB(buf)) + pb * 520 +
             ((lane >> 4) << 7) + ((lane & 15) << 2);
         s16x4 b0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
             (__attribute__((address_space(3))) s16x4*)bbase);
@@ -279,10 +279,10 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
   const int wm = wid >> 1, wn = wid & 1;
 
   __shared__ __attribute__((aligned(16))) short lds[2 * (BMT * (BK + 8)) +
-                                                    2 * (PBLK * 528)];
+                                                    2 * (PBLK * 520)];
   auto ldsA = [&](int buf) { return lds + buf * (BMT * (BK + 8)); };
   auto ldsB = [&](int buf) {
-    return lds + 2 * (BMT * (BK + 8)) + buf * (PBLK * 528);
+    return lds + 2 * (BMT * (BK + 8)) + buf * (PBLK * 520);
   };
 
   f32x4 acc[MF][NF];
@@ -414,7 +414,7 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
           short v[8];
 #pragma unroll
           for (int e = 0; e < 8; ++e) v[e] = raw[ss + e];
-          const int base = (px0 >> 4) * 528 + ((kk >> 3) << 7) +
+          const int base = (px0 >> 4) * 520 + ((kk >> 3) << 7) +
                            (((kk >> 2) & 1) << 6) + ((kk & 3) << 4) +
                            (px0 & 15);
           *(s16x8*)(ldsB(buf) + base) = *(const s16x8*)v;
@@ -435,7 +435,7 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
                        ? src[col]
                        : (short)0;
           }
-          const int base = (px0 >> 4) * 528 + ((kk >> 3) << 7) +
+          const int base = (px0 >> 4) * 520 + ((kk >> 3) << 7) +
                            (((kk >> 2) & 1) << 6) + ((kk & 3) << 4) +
                            (px0 & 15);
           *(s16x8*)(ldsB(buf) + base) = *(const s16x8*)v;
@@ -446,23 +446,6 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
 
   const int a_row0 = wm * (BMT / 2);
   const int b_px0 = wn * (BNT / 2);
-
-  // fragment LDS addresses are loop-invariant per buffer — precompute
-  // (the per-iteration VALU was ~12 insts/MFMA, PMC run 20)
-  const short* aAddr[2][MF];
-  __attribute__((address_space(3))) short* bAddr[2][NF];
-#pragma unroll
-  for (int b2 = 0; b2 < 2; ++b2) {
-#pragma unroll
-    for (int mf = 0; mf < MF; ++mf)
-      aAddr[b2][mf] = ldsA(b2) + (a_row0 + mf * 16 + (lane & 15)) * (BK + 8) +
-                      ((lane >> 4) << 3);
-#pragma unroll
-    for (int nf = 0; nf < NF; ++nf)
-      bAddr[b2][nf] = (__attribute__((address_space(3))) short*)(ldsB(b2)) +
-                      ((b_px0 >> 4) + nf) * 528 + ((lane >> 4) << 7) +
-                      ((lane & 15) << 2);
-  }
 
   stage_load(0);
   stage_write(0);
@@ -475,10 +458,16 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
     if (it + 2 < total_it) stage_load(it + 2);
 #pragma unroll
     for (int mf = 0; mf < MF; ++mf) {
-      s16x8 afrag = *(const s16x8*)aAddr[buf][mf];
+      const short* arow =
+          ldsA(buf) + (a_row0 + mf * 16 + (lane & 15)) * (BK + 8) +
+          ((lane >> 4) << 3);
+      s16x8 afrag = *(const s16x8*)arow;
 #pragma unroll
       for (int nf = 0; nf < NF; ++nf) {
-        __attribute__((address_space(3))) short* bbase = bAddr[buf][nf];
+        const int pb = (b_px0 >> 4) + nf;
+        __attribute__((address_space(3))) short* bbase =
+            (__attribute__((address_space(3))) short*)(ldsB(buf)) + pb * 520 +
+            ((lane >> 4) << 7) + ((lane & 15) << 2);
         s16x4 b0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
             (__attribute__((address_space(3))) s16x4*)bbase);
         s16x4 b1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
