@@ -241,7 +241,7 @@ __global__ __launch_bounds__(256) void conv_fwd_kernel(
 // Accumulators persist across the R passes.
 // ---------------------------------------------------------------------------
 
-template <int SS, int BMT, int BNT>
+template <int SS>
 __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ w2,
     const float* __restrict__ bias, bf16* __restrict__ out, ConvGeom g) {
@@ -250,46 +250,39 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
   // right after the single per-iteration barrier, so its vmcnt wait is
   // hidden under the previous iteration's MFMAs.
   const int S = SS > 0 ? SS : g.S;
-  // tile geometry: 4 waves as 2x2; wave tile = (BMT/2) x (BNT/2)
-  constexpr int MF = BMT / 32;   // 16-row fragments per wave (4 or 2)
-  constexpr int NF = BNT / 32;   // 16-col fragments per wave (4 or 2)
-  constexpr int PBLK = BNT / 16; // 16-px blocks in the B image
-  // staging units = nci c-columns x (BNT/8) px chunks; worst-case nci
-  // per BK=32 tile: 32 (S=1), 12 (S=3), 6 (S=7)
-  constexpr int UNITS = (SS == 1 && BNT == 128) ? 2 : 1;
+  constexpr int UNITS = (SS == 1) ? 2 : 1;  // S==1: 32 c's -> 512 units
 
-  const int m_tiles = (g.K + BMT - 1) / BMT;
+  const int m_tiles = (g.K + BM - 1) / BM;
   const int nwg = gridDim.x;
   const int q8 = nwg >> 3, r8 = nwg & 7;
   const int xcd = blockIdx.x & 7, sub = blockIdx.x >> 3;
   int bid = (xcd < r8 ? xcd * (q8 + 1) : r8 * (q8 + 1) + (xcd - r8) * q8) + sub;
   const int mt = bid % m_tiles;
   bid /= m_tiles;
-  const int row_tiles = (g.OW + BNT - 1) / BNT;
-  const int owt = bid % row_tiles;
-  bid /= row_tiles;
+  const int owt = bid % g.row_tiles;
+  bid /= g.row_tiles;
   const int oh = bid % g.OH;
   const int n = bid / g.OH;
 
-  const int k0out = mt * BMT;
-  const int ow0 = owt * BNT;
+  const int k0out = mt * BM;
+  const int ow0 = owt * BN;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
   const int wm = wid >> 1, wn = wid & 1;
 
-  __shared__ __attribute__((aligned(16))) short lds[2 * (BMT * (BK + 8)) +
-                                                    2 * (PBLK * 520)];
-  auto ldsA = [&](int buf) { return lds + buf * (BMT * (BK + 8)); };
+  __shared__ __attribute__((aligned(16))) short lds[2 * (BM * (BK + 8)) +
+                                                    2 * ((BN / 16) * 520)];
+  auto ldsA = [&](int buf) { return lds + buf * (BM * (BK + 8)); };
   auto ldsB = [&](int buf) {
-    return lds + 2 * (BMT * (BK + 8)) + buf * (PBLK * 520);
+    return lds + 2 * (BM * (BK + 8)) + buf * ((BN / 16) * 520);
   };
 
-  f32x4 acc[MF][NF];
+  f32x4 acc[4][4];
 #pragma unroll
-  for (int i = 0; i < MF; ++i)
+  for (int i = 0; i < 4; ++i)
 #pragma unroll
-    for (int j = 0; j < NF; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
   const int64_t HW = (int64_t)g.H * g.W;
   const int64_t in_n = (int64_t)n * g.C * HW;
@@ -299,8 +292,7 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
   const int w_lo = ow0 - g.pw;  // sw == 1
 
   // pipelined register state
-  constexpr int APASS = (BMT * BK) / 2048;  // 2 for BMT=128, 1 for 64
-  s16x8 aReg[APASS];
+  s16x8 aReg[2];
   s16x8 bReg[UNITS][3];
 
 
@@ -308,7 +300,7 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
     const int r = it / KT;
     const int kk0 = (it % KT) * BK;
 #pragma unroll
-    for (int pass = 0; pass < APASS; ++pass) {
+    for (int pass = 0; pass < 2; ++pass) {
       const int idx = pass * 256 + tid;
       const int row = idx >> 2;
       const int kc = (idx & 3) * 8;
@@ -324,8 +316,8 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
 #pragma unroll
     for (int u0 = 0; u0 < UNITS; ++u0) {
       const int u = u0 * 256 + tid;
-      const int ci = u / (BNT / 8);
-      const int pxc = u % (BNT / 8);
+      const int ci = u >> 4;
+      const int pxc = u & 15;
       const int c = c_lo + ci;
       const int px0 = pxc * 8;
       const int a0 = w_lo + px0;
@@ -346,7 +338,7 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
     const int r = it / KT;
     const int kk0 = (it % KT) * BK;
 #pragma unroll
-    for (int pass = 0; pass < APASS; ++pass) {
+    for (int pass = 0; pass < 2; ++pass) {
       const int idx = pass * 256 + tid;
       const int row = idx >> 2;
       const int kc = (idx & 3) * 8;
@@ -372,8 +364,8 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
 #pragma unroll
     for (int u0 = 0; u0 < UNITS; ++u0) {
       const int u = u0 * 256 + tid;
-      const int ci = u / (BNT / 8);
-      const int pxc = u % (BNT / 8);
+      const int ci = u >> 4;
+      const int pxc = u & 15;
       const int c = c_lo + ci;
       const int px0 = pxc * 8;
       const int a0 = w_lo + px0;
@@ -444,8 +436,8 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
     }
   };
 
-  const int a_row0 = wm * (BMT / 2);
-  const int b_px0 = wn * (BNT / 2);
+  const int a_row0 = wm * 64;
+  const int b_px0 = wn * 64;
 
   stage_load(0);
   stage_write(0);
@@ -457,13 +449,13 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
     if (it + 1 < total_it) stage_write(it + 1);
     if (it + 2 < total_it) stage_load(it + 2);
 #pragma unroll
-    for (int mf = 0; mf < MF; ++mf) {
+    for (int mf = 0; mf < 4; ++mf) {
       const short* arow =
           ldsA(buf) + (a_row0 + mf * 16 + (lane & 15)) * (BK + 8) +
           ((lane >> 4) << 3);
       s16x8 afrag = *(const s16x8*)arow;
 #pragma unroll
-      for (int nf = 0; nf < NF; ++nf) {
+      for (int nf = 0; nf < 4; ++nf) {
         const int pb = (b_px0 >> 4) + nf;
         __attribute__((address_space(3))) short* bbase =
             (__attribute__((address_space(3))) short*)(ldsB(buf)) + pb * 520 +
@@ -486,7 +478,7 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
 
   const int64_t out_n = ((int64_t)n * g.K) * g.OH * g.OW;
 #pragma unroll
-  for (int mf = 0; mf < MF; ++mf) {
+  for (int mf = 0; mf < 4; ++mf) {
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg) {
       const int kout = k0out + a_row0 + mf * 16 + ((lane >> 4) << 2) + reg;
@@ -495,7 +487,7 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
       const int64_t orow =
           out_n + (int64_t)kout * g.OH * g.OW + (int64_t)oh * g.OW;
 #pragma unroll
-      for (int nf = 0; nf < NF; ++nf) {
+      for (int nf = 0; nf < 4; ++nf) {
         const int ow = ow0 + b_px0 + nf * 16 + (lane & 15);
         if (ow < g.OW) out[orow + ow] = (bf16)(acc[mf][nf][reg] + b);
       }
@@ -692,28 +684,20 @@ torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor w,
                   .permute({0, 2, 1, 3})
                   .reshape({g.K, g.R, (int64_t)g.C * g.S})
                   .contiguous();
-    // grid sizing (guide G1/G11): small-HW shapes with 128x128 tiles
-    // launch <1024 blocks = 1 wave/SIMD (every stall exposed) — switch
-    // to 64x64 tiles, 4x the blocks
-    const int64_t blocks128 =
-        (int64_t)((g.K + 127) / 128) * ((g.OW + 127) / 128) * g.OH * g.N;
-    const bool small = blocks128 < 1024;
-    const int64_t blocks_v2 =
-        small ? (int64_t)((g.K + 63) / 64) * ((g.OW + 63) / 64) * g.OH * g.N
-              : blocks128;
     auto launch_v2 = [&](auto* kern) {
-      hipLaunchKernelGGL(kern, dim3((uint32_t)blocks_v2), dim3(256), 0,
+      hipLaunchKernelGGL(kern, dim3((uint32_t)blocks), dim3(256), 0,
                          stream.stream(), (const bf16*)x.data_ptr(),
                          (const bf16*)w2.data_ptr(), bptr,
                          (bf16*)out.data_ptr(), g);
     };
-    if (g.S == 3 && small) launch_v2(conv_fwd_v2_kernel<3, 64, 64>);
-    else if (g.S == 3) launch_v2(conv_fwd_v2_kernel<3, 128, 128>);
-    else if (g.S == 7 && small) launch_v2(conv_fwd_v2_kernel<7, 64, 64>);
-    else if (g.S == 7) launch_v2(conv_fwd_v2_kernel<7, 128, 128>);
-    else if (g.S == 1 && small) launch_v2(conv_fwd_v2_kernel<1, 64, 64>);
-    else if (g.S == 1) launch_v2(conv_fwd_v2_kernel<1, 128, 128>);
-    else launch_v2(conv_fwd_v2_kernel<0, 128, 128>);
+    if (g.S == 3)
+      launch_v2(conv_fwd_v2_kernel<3>);
+    else if (g.S == 7)
+      launch_v2(conv_fwd_v2_kernel<7>);
+    else if (g.S == 1)
+      launch_v2(conv_fwd_v2_kernel<1>);
+    else
+      launch_v2(conv_fwd_v2_kernel<0>);
     return out;
   }
   hipLaunchKernelGGL(conv_fwd_kernel, dim3((uint32_t)blocks), dim3(256), 0,
