@@ -208,10 +208,27 @@ def run_training(args, mode):
     B = extras["batch_per_step"]
     torch.manual_seed(1405 + comm.rank)  # reference seeds a fixed value too
     S = args.image_size
-    x = torch.randn(B, 3, S, S)
-    y = torch.randint(0, args.num_classes, (B,))
-    if on_gpu:
-        x, y = x.cuda(), y.cuda()
+
+    loader = None
+    if args.app in (1, 2):
+        # real data: ImageFolder / CIFAR-10 (reference APP wiring,
+        # benchmark_resnet_lp.py:183-208); every rank iterates the same
+        # deterministic batch sequence
+        from mpi4dl_amd.data import make_dataloader
+
+        loader, n = make_dataloader(
+            args.app, args.datapath, B, S, args.num_classes,
+            num_workers=args.num_workers,
+        )
+        log.info("dataset: app=%d size=%d (%d batches/epoch)", args.app, n,
+                 len(loader))
+        batches = iter(loader)
+    else:
+        # synthetic: one fixed random batch (perf benchmarking path)
+        x = torch.randn(B, 3, S, S)
+        y = torch.randint(0, args.num_classes, (B,))
+        if on_gpu:
+            x, y = x.cuda(), y.cuda()
 
     if args.resume and args.checkpoint_dir:
         ckpt.load_checkpoint(args.checkpoint_dir, extras["gen"].models,
@@ -220,6 +237,14 @@ def run_training(args, mode):
     times = []
     for epoch in range(args.num_epochs):
         for it in range(args.num_steps):
+            if loader is not None:
+                try:
+                    x, y = next(batches)
+                except StopIteration:
+                    batches = iter(loader)
+                    x, y = next(batches)
+                if on_gpu:
+                    x, y = x.cuda(non_blocking=True), y.cuda(non_blocking=True)
             t0 = time.perf_counter()
             loss, corr, seen = step(x, y)
             if on_gpu:
