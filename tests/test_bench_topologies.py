@@ -1,0 +1,75 @@
+"""bench.py end-to-end at the driver's SCALE topologies, on gloo.
+
+The round-end driver launches bench.py at N=1/2/4/8 (one rank per GPU
+over RCCL). These tests run the SAME main() — same topology table, same
+engine construction, same JSON contract — at world 2 (pp2), 4
+(sp2+pp3) and 8 (sp4+pp5) on CPU, so a rendezvous/topology/balance bug
+can't first appear on the 8-GPU box."""
+
+import io
+import json
+import sys
+
+from dist_util import run_distributed
+
+
+def _bench_body(rank, world, extra_argv):
+    import bench
+
+    argv = [
+        "bench.py", "--gpus", str(world), "--steps", "1", "--warmup", "0",
+        "--image-size", "64", "--batch", "8", "--num-layers", "6",
+        "--num-filters", "16", "--num-classes", "10",
+    ] + list(extra_argv)
+    if "--image-size" in extra_argv:
+        argv = [a for a in argv]  # extra_argv overrides the 64 default
+        i = argv.index("--image-size")
+        del argv[i:i + 2]
+    old_argv, old_out = sys.argv, sys.stdout
+    sys.argv = argv
+    sys.stdout = cap = io.StringIO()
+    try:
+        bench.main()
+    finally:
+        sys.argv, sys.stdout = old_argv, old_out
+    lines = [ln for ln in cap.getvalue().splitlines() if ln.startswith("{")]
+    return json.loads(lines[-1]) if lines else None
+
+
+def _check(out, world, parallelism):
+    rank0 = out[0]
+    assert rank0 is not None, "rank 0 printed no JSON"
+    assert rank0["n_gpus"] == world
+    assert rank0["config"]["parallelism"] == parallelism
+    assert rank0["value"] > 0 and rank0["ms_per_step"] > 0
+    for k in ("metric", "unit", "steps", "warmup", "higher_is_better",
+              "scaling", "vs_baseline", "dtype", "data"):
+        assert k in rank0, k
+    # only rank 0 prints
+    assert all(o is None for o in out[1:])
+
+
+def test_bench_n2_pp():
+    out = run_distributed(_bench_body, 2, ((),), timeout=300)
+    _check(out, 2, "pp2")
+
+
+def test_bench_n4_sp_pp():
+    out = run_distributed(_bench_body, 4, ((),), timeout=300)
+    _check(out, 4, "sp2+pp3")
+
+
+def test_bench_n8_sp_pp():
+    # 128² so the deepest spatial cell's 4-way tile (W=4) still fits the
+    # 1×7 halo of 3 — at 64² the tile is 2 wide and the loud halo-bounds
+    # assertion fires (by design; the real config is 2048²)
+    out = run_distributed(
+        _bench_body, 8, (("--image-size", "128"),), timeout=600
+    )
+    _check(out, 8, "sp4+pp5")
+
+
+def test_bench_n4_gems():
+    out = run_distributed(_bench_body, 4, (("--gems",),), timeout=300)
+    _check(out, 4, "sp2+gems+pp3")
+    assert out[0]["config"]["global_batch"] == 16  # two replicas per step
