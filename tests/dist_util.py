@@ -37,7 +37,9 @@ def _worker(rank, world_size, port, fn, args, q):
         dist.barrier()
         dist.destroy_process_group()
         q.put((rank, "ok", result))
-    except Exception:
+    except BaseException:
+        # BaseException: argparse's SystemExit must reach the queue too,
+        # else the parent only sees a timeout
         q.put((rank, "err", traceback.format_exc()))
         raise
 
