@@ -39,8 +39,10 @@ def get_parser() -> argparse.ArgumentParser:
     p.add_argument("--num-classes", type=int, default=10)
     p.add_argument("--balance", type=str, default=None,
                    help="csv: cells per partition")
-    p.add_argument("--halo-d2", action="store_true",
-                   help="use the D2 fused-halo model variant")
+    p.add_argument("--halo-d2", "--halo-D2", action="store_true",
+                   dest="halo_d2",
+                   help="use the D2 fused-halo model variant "
+                        "(--halo-D2 = the reference's spelling)")
     p.add_argument("--fused-layers", type=int, default=4,
                    help="blocks per fused D2 halo exchange")
     p.add_argument("--local-DP", type=int, default=1,
