@@ -259,6 +259,40 @@ def run_training(args, mode):
     if args.checkpoint_dir and not args.resume:
         ckpt.save_checkpoint(args.checkpoint_dir, extras["gen"].models,
                              extras.get("optimizer"), comm)
+
+    if args.enable_evaluation:
+        # forward-only pass over held-out batches (CIFAR test split when
+        # --app 2, else fresh synthetic data); top-1 acc on the last rank
+        eng = extras["engine"]
+        Be = args.batch_size  # GEMS evals one replica: plain batch, not 2x
+        if args.app == 2:
+            from mpi4dl_amd.data import make_dataloader
+
+            ev_loader, _ = make_dataloader(
+                2, args.datapath, Be, S, args.num_classes,
+                num_workers=args.num_workers, train=False,
+            )
+            ev_batches = list(ev_loader)[: max(args.num_steps, 1)]
+        else:
+            g = torch.Generator().manual_seed(7)
+            ev_batches = [
+                (torch.randn(Be, 3, S, S, generator=g),
+                 torch.randint(0, args.num_classes, (Be,), generator=g))
+                for _ in range(max(args.num_steps, 1))
+            ]
+        tot_loss, tot_corr, tot_seen = 0.0, 0, 0
+        for x, y in ev_batches:
+            if on_gpu:
+                x, y = x.cuda(), y.cuda()
+            loss, corr, seen = eng.run_eval(x, y)
+            tot_loss += loss
+            tot_corr += corr
+            tot_seen += seen
+        if comm.rank == comm.world_size - 1 and tot_seen:
+            print(
+                f"Eval loss {tot_loss / len(ev_batches):.4f} "
+                f"acc {tot_corr / tot_seen:.3f} over {tot_seen} samples"
+            )
     if comm.rank == 0:
         steady = times[1:] or times
         img_s = B / statistics.median(steady)

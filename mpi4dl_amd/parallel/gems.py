@@ -148,6 +148,14 @@ class train_model_master:
             seen += s1 + s2
         return loss / self.replications, corr, seen
 
+    @torch.no_grad()
+    def run_eval(self, inputs, labels):
+        """Forward-only pass on replica 1 only — after
+        allreduce_and_update both replicas hold identical weights, so
+        one engine's eval is the model's eval (inputs/labels: a plain
+        batch_size batch, not the 2x training layout)."""
+        return self.train_model1.run_eval(inputs, labels)
+
     def allreduce_and_update(self):
         """Pair the two replicas' grads, then step both optimizers
         (reference apply_allreduce_master_and_update, comm.py:516-523)."""
@@ -258,6 +266,11 @@ class train_spatial_model_master:
             corr += c1 + c2
             seen += s1 + s2
         return loss / self.replications, corr, seen
+
+    @torch.no_grad()
+    def run_eval(self, inputs, labels):
+        """Forward-only pass on replica 1 (see train_model_master.run_eval)."""
+        return self.train_model1.run_eval(inputs, labels)
 
     def allreduce_and_update(self):
         """Spatial tile reduction per engine (engine2 over mirror groups),

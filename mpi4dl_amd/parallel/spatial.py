@@ -402,9 +402,10 @@ class train_model_spatial(train_model):
     # step
     # ------------------------------------------------------------------
 
-    def run_step(self, inputs, labels):
+    def _slice_io(self, inputs, labels):
         """First-stage tile ranks slice their tile from the full input
-        (reference split_input, train_spatial.py:241)."""
+        (reference split_input, train_spatial.py:241); local-DP last
+        stages shard the labels."""
         if inputs is not None and self.first_stage and self.is_tile_rank:
             inputs = self.layout.slice_input(inputs, self.tile_idx).contiguous()
         L = self.comm.LOCAL_DP_LP
@@ -419,4 +420,13 @@ class train_model_spatial(train_model):
                 py.chunk(L, dim=0)[self.my_dp] for py in labels.chunk(self.parts, 0)
             ]
             labels = torch.cat(shards, dim=0)
+        return inputs, labels
+
+    def run_step(self, inputs, labels):
+        inputs, labels = self._slice_io(inputs, labels)
         return super().run_step(inputs, labels)
+
+    @torch.no_grad()
+    def run_eval(self, inputs, labels):
+        inputs, labels = self._slice_io(inputs, labels)
+        return super().run_eval(inputs, labels)

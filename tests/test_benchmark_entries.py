@@ -32,8 +32,9 @@ def _run_mode(rank, world, mode, model, extra):
 
 def test_entry_lp_resnet():
     # 2 LP stages (benchmark_resnet_lp.py equivalent)
-    got = run_distributed(_run_mode, 2,
-                          ("lp", "resnet", ("--split-size", "2")))
+    got = run_distributed(
+        _run_mode, 2,
+        ("lp", "resnet", ("--split-size", "2", "--enable-evaluation")))
     assert got[0] == 2
 
 
@@ -54,8 +55,9 @@ def test_entry_sp_amoebanet():
 
 def test_entry_gems_resnet():
     # two mirrored LP engines on 2 ranks (benchmark_resnet_gems_master.py)
-    got = run_distributed(_run_mode, 2,
-                          ("gems", "resnet", ("--split-size", "2")))
+    got = run_distributed(
+        _run_mode, 2,
+        ("gems", "resnet", ("--split-size", "2", "--enable-evaluation")))
     assert got[0] == 2
 
 
@@ -66,7 +68,8 @@ def test_entry_gems_sp_resnet():
         _run_mode, 4,
         ("gems_sp", "resnet",
          ("--split-size", "3", "--num-spatial-parts", "2",
-          "--spatial-size", "1", "--slice-method", "vertical")),
+          "--spatial-size", "1", "--slice-method", "vertical",
+          "--enable-evaluation")),
         timeout=300,
     )
     assert got[0] == 2

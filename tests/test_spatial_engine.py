@@ -208,3 +208,61 @@ def test_sp_horizontal_2tiles_plus_lp():
     )[-1]
     for e, g in zip(expected, got):
         assert abs(e - g) < 2e-4, (expected, got)
+
+
+def _sp_eval_body(rank, world, batch):
+    from mpi4dl_amd.comm import Communicator
+    from mpi4dl_amd.models import resnet_spatial
+    from mpi4dl_amd.ops.plan import SpatialPlan
+    from mpi4dl_amd.parallel.partition import model_generator
+    from mpi4dl_amd.parallel.spatial import train_model_spatial
+
+    comm = Communicator(
+        split_size=2, ENABLE_SPATIAL=True, num_spatial_parts=2,
+        spatial_size=1, backend="gloo",
+    )
+    torch.manual_seed(0)
+    probe = resnet_spatial.get_resnet_v1((1, 3, IMG, IMG), NCLS, n=1, num_filters=8)
+    base, rem = divmod(len(probe), 2)
+    balance = [base + (1 if i < rem else 0) for i in range(2)]
+    plan = SpatialPlan(comm, balance, "vertical")
+    torch.manual_seed(0)
+    model = resnet_spatial.get_resnet_v1(
+        (batch, 3, IMG, IMG), NCLS, n=1, num_filters=8, plan=plan
+    )
+    gen = model_generator(model, 2, input_size=(batch, 3, IMG, IMG),
+                          balance=balance)
+    gen.get_output_shapes()
+    gen.ready_model(comm.split_rank, device=torch.device("cpu"))
+    eng = train_model_spatial(
+        gen, comm.local_rank, batch, 1, comm, slice_method="vertical",
+        device=torch.device("cpu"),
+    )
+    torch.manual_seed(9)
+    x = torch.randn(batch, 3, IMG, IMG)
+    y = torch.randint(0, NCLS, (batch,))
+    loss, corr, seen = eng.run_eval(x, y)
+    return loss, corr, seen
+
+
+def test_sp_eval_parity():
+    """Spatial run_eval (forward-only, eval-mode BN) must match serial
+    model.eval() on the full image — the spatial input slicing now runs
+    in run_eval too."""
+    from mpi4dl_amd.models.resnet import get_resnet_v1
+
+    batch = 2
+    torch.manual_seed(0)
+    ref = get_resnet_v1((batch, 3, IMG, IMG), num_classes=NCLS, n=1, num_filters=8)
+    ref.eval()
+    torch.manual_seed(9)
+    x = torch.randn(batch, 3, IMG, IMG)
+    y = torch.randint(0, NCLS, (batch,))
+    with torch.no_grad():
+        logits = ref(x).float()
+        rl = float(nn.functional.cross_entropy(logits, y))
+        rcorr = int((logits.argmax(1) == y).sum())
+    got = run_distributed(_sp_eval_body, 3, (batch,))
+    loss, corr, seen = got[-1]
+    assert seen == batch and corr == rcorr
+    assert abs(loss - rl) < 1e-4, (loss, rl)
