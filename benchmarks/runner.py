@@ -115,7 +115,7 @@ def make_engines(args, mode):
         act_dtype=torch.bfloat16 if (args.dtype == "bf16" and on_gpu) else torch.float32,
         schedule=getattr(args, "schedule", "gpipe"),
     )
-    reducer = GradReducer(comm)
+    reducer = GradReducer(comm, fp16_allreduce=args.fp16_allreduce)
 
     if not gems:
         plan = (

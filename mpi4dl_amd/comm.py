@@ -445,9 +445,14 @@ class GradReducer:
     rank-dependent order (comm.py:460-504).
     """
 
-    def __init__(self, comm: Communicator):
+    def __init__(self, comm: Communicator, fp16_allreduce: bool = False):
         self.comm = comm
         self._flat = {}  # id(module) -> FlatGrads
+        # --fp16-allreduce: reduce a bf16 copy of the flat grads (halves
+        # bytes on the xGMI links; bf16 keeps fp32's exponent range so
+        # gradient magnitudes survive — fp16 would overflow). Opt-in:
+        # trades ~3 bits of mantissa in the reduced gradient.
+        self.fp16_allreduce = fp16_allreduce
 
     # -- weights -------------------------------------------------------------
 
@@ -498,7 +503,12 @@ class GradReducer:
             return
         n = dist.get_world_size(group=group)
         with GLOBAL_TIMER.phase("grad/allreduce"):
-            dist.all_reduce(fg.buffer, group=group)
+            if self.fp16_allreduce:
+                comp = fg.buffer.to(torch.bfloat16)
+                dist.all_reduce(comp, group=group)
+                fg.buffer.copy_(comp)
+            else:
+                dist.all_reduce(fg.buffer, group=group)
         fg.rescale_(divide_by if divide_by is not None else float(n))
 
     def apply_allreduce(self, module: torch.nn.Module):

@@ -7,6 +7,7 @@ GEMS inversion (comm.py:77-80).
 
 import pytest
 
+from dist_util import run_distributed
 from mpi4dl_amd.comm import Communicator, compute_mp_size, normalize_spatial_parts
 
 
@@ -149,3 +150,39 @@ def test_outer_dp_groups_gloo():
     from dist_util import run_distributed
 
     run_distributed(_outer_dp_body, world_size=4)
+
+
+def _fp16_allreduce_body(rank, world):
+    import torch
+
+    from mpi4dl_amd.comm import Communicator, GradReducer
+
+    comm = Communicator(split_size=1, backend="gloo")  # world ranks = DP replicas
+    torch.manual_seed(rank)
+    m = torch.nn.Linear(8, 4)
+    (m(torch.randn(2, 8)).sum()).backward()
+    exact = GradReducer(comm)
+    comp = GradReducer(comm, fp16_allreduce=True)
+    ref = [p.grad.clone() for p in m.parameters()]
+    comp.apply_allreduce(m)
+    got = [p.grad.clone() for p in m.parameters()]
+    # reconstruct expectation: bf16-rounded local grads averaged in bf16
+    return [g.tolist() for g in got], [r.tolist() for r in ref]
+
+
+def test_fp16_allreduce_close():
+    """--fp16-allreduce reduces a bf16 copy: result must equal the fp32
+    average within bf16 rounding (~3 decimal digits)."""
+    import torch
+
+    got = run_distributed(_fp16_allreduce_body, 2)
+    g0 = [torch.tensor(t) for t in got[0][0]]
+    g1 = [torch.tensor(t) for t in got[1][0]]
+    # both ranks converge to the same reduced gradient
+    for a, b in zip(g0, g1):
+        assert torch.allclose(a, b)
+    # and it is close to the true fp32 mean of the two local grads
+    r0 = [torch.tensor(t) for t in got[0][1]]
+    r1 = [torch.tensor(t) for t in got[1][1]]
+    for a, x, y in zip(g0, r0, r1):
+        assert torch.allclose(a, (x + y) / 2, rtol=2e-2, atol=2e-2)
