@@ -161,7 +161,6 @@ def _fp16_allreduce_body(rank, world):
     torch.manual_seed(rank)
     m = torch.nn.Linear(8, 4)
     (m(torch.randn(2, 8)).sum()).backward()
-    exact = GradReducer(comm)
     comp = GradReducer(comm, fp16_allreduce=True)
     ref = [p.grad.clone() for p in m.parameters()]
     comp.apply_allreduce(m)
