@@ -96,6 +96,9 @@ def main():
     ap.add_argument("--gems", action="store_true",
                     help="GEMS bidirectional pipelines on top of SP "
                          "(two mirrored engines per GPU, 2x batch/step)")
+    ap.add_argument("--act-ckpt", action="store_true",
+                    help="recompute cell forwards in backward (fits larger "
+                         "global batches in HBM at ~1 extra forward cost)")
     ap.add_argument("--grad-mode", default="drop", choices=["drop", "exact"],
                     help="halo backward: 'drop' = reference semantics + "
                          "halo/compute overlap (benchmark default); "
@@ -189,6 +192,7 @@ def main():
         autocast_dtype=autocast_dtype,
         act_dtype=act_dtype,
         schedule=args.schedule,
+        act_ckpt=args.act_ckpt,
     )
     if use_gems:
         from mpi4dl_amd.parallel.gems import train_spatial_model_master

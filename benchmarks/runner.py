@@ -114,6 +114,7 @@ def make_engines(args, mode):
         autocast_dtype=torch.bfloat16 if (args.dtype == "bf16" and on_gpu) else None,
         act_dtype=torch.bfloat16 if (args.dtype == "bf16" and on_gpu) else torch.float32,
         schedule=getattr(args, "schedule", "gpipe"),
+        act_ckpt=getattr(args, "act_ckpt", False),
     )
     reducer = GradReducer(comm, fp16_allreduce=args.fp16_allreduce)
 

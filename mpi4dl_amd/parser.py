@@ -66,6 +66,9 @@ def get_parser() -> argparse.ArgumentParser:
     p.add_argument("--grad-mode", default="exact", choices=["exact", "drop"],
                    help="halo backward: exact transposed exchange or the "
                         "reference's drop semantics")
+    p.add_argument("--act-ckpt", action="store_true",
+                   help="activation checkpointing: recompute cell forwards "
+                        "in backward (cuts GPipe peak activation memory)")
     p.add_argument("--checkpoint-dir", default=None)
     p.add_argument("--resume", action="store_true")
     p.add_argument("--verbose", action="store_true")

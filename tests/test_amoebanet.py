@@ -194,3 +194,39 @@ def test_amoebanet_sp_square_parity():
     got = run_distributed(_sp_square_body, 5, (steps, batch, parts, lr))[-1]
     for e, g in zip(expected, got):
         assert abs(e - g) < 1e-3, (expected, got)  # fp32 reduction-order noise
+
+
+def _lp_ckpt_body(rank, world, steps, batch, parts, lr):
+    from mpi4dl_amd.comm import Communicator
+    from mpi4dl_amd.parallel.partition import model_generator
+    from mpi4dl_amd.parallel.pipeline import train_model
+
+    comm = Communicator(split_size=world, backend="gloo")
+    model = _build()
+    gen = model_generator(model, world, input_size=(batch // parts, 3, IMG, IMG))
+    gen.get_output_shapes()
+    gen.ready_model(comm.local_rank, device=torch.device("cpu"))
+    opt = torch.optim.SGD(gen.models.parameters(), lr=lr, momentum=0.9)
+    eng = train_model(
+        gen, comm.local_rank, batch, parts, comm, optimizer=opt,
+        device=torch.device("cpu"), act_ckpt=True,
+    )
+    torch.manual_seed(42)
+    losses = []
+    for _ in range(steps):
+        x = torch.randn(batch, 3, IMG, IMG)
+        y = torch.randint(0, NCLS, (batch,))
+        loss, _, _ = eng.run_step(x, y)
+        eng.update()
+        losses.append(loss)
+    return losses
+
+
+def test_amoebanet_act_ckpt_parity():
+    """Tuple (x, skip) activations through per-cell non-reentrant
+    checkpoint: trajectory must match serial."""
+    steps, batch, parts, lr = 2, 2, 1, 0.01
+    expected = _serial_losses(steps, batch, parts, lr)
+    got = run_distributed(_lp_ckpt_body, 3, (steps, batch, parts, lr))[-1]
+    for e, g in zip(expected, got):
+        assert abs(e - g) < 2e-4, (expected, got)

@@ -276,3 +276,58 @@ def test_1f1b_schedule_parity():
     got = run_distributed(_pipeline_1f1b_body, 3, (steps, batch, parts, lr))[-1]
     for e, g in zip(expected, got):
         assert abs(e - g) < 1e-4, (expected, got)
+
+
+def _act_ckpt_body(rank, world, steps, batch, parts, lr):
+    from mpi4dl_amd.comm import Communicator
+    from mpi4dl_amd.models.resnet import get_resnet_v1
+    from mpi4dl_amd.parallel.partition import model_generator
+    from mpi4dl_amd.parallel.pipeline import train_model
+
+    comm = Communicator(split_size=world, backend="gloo")
+    torch.manual_seed(0)
+    model = get_resnet_v1((batch // parts, 3, 32, 32), 10, n=1, num_filters=8)
+    gen = model_generator(model, world,
+                          input_size=(batch // parts, 3, 32, 32))
+    gen.get_output_shapes()
+    gen.ready_model(comm.local_rank, device=torch.device("cpu"))
+    opt = torch.optim.SGD(gen.models.parameters(), lr=lr, momentum=0.9)
+    eng = train_model(gen, comm.local_rank, batch, parts, comm, optimizer=opt,
+                      device=torch.device("cpu"), act_ckpt=True)
+    torch.manual_seed(42)
+    losses = []
+    for _ in range(steps):
+        x = torch.randn(batch, 3, 32, 32)
+        y = torch.randint(0, 10, (batch,))
+        loss, _, _ = eng.run_step(x, y)
+        eng.update()
+        losses.append(loss)
+    return losses
+
+
+def test_act_ckpt_parity():
+    """Activation checkpointing (recompute in backward) must not change
+    the training trajectory — same losses as plain serial training."""
+    from mpi4dl_amd.models.resnet import get_resnet_v1
+
+    steps, batch, parts, lr = 3, 4, 2, 0.01
+    torch.manual_seed(0)
+    model = get_resnet_v1((batch // parts, 3, 32, 32), 10, n=1, num_filters=8)
+    opt = torch.optim.SGD(model.parameters(), lr=lr, momentum=0.9)
+    crit = nn.CrossEntropyLoss()
+    torch.manual_seed(42)
+    expected = []
+    for _ in range(steps):
+        x = torch.randn(batch, 3, 32, 32)
+        y = torch.randint(0, 10, (batch,))
+        total = 0.0
+        for px, py in zip(x.chunk(parts), y.chunk(parts)):
+            loss = crit(model(px).float(), py)
+            (loss / parts).backward()
+            total += float(loss.detach())
+        opt.step()
+        opt.zero_grad(set_to_none=False)
+        expected.append(total / parts)
+    got = run_distributed(_act_ckpt_body, 2, (steps, batch, parts, lr))[-1]
+    for e, g in zip(expected, got):
+        assert abs(e - g) < 2e-4, (expected, got)
