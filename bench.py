@@ -258,6 +258,9 @@ def main():
     elapsed = float(t[0])
 
     imgs_per_s = B_step * args.steps / elapsed
+    peak_gb = (
+        round(torch.cuda.max_memory_allocated() / 2**30, 2) if on_gpu else None
+    )
     if rank == 0:
         par = "single" if n == 1 else (
             f"pp{topo['split_size']}" if not spatial
@@ -287,6 +290,8 @@ def main():
                         "image_size": S,
                         "parallelism": par,
                         "slice_method": args.slice_method if spatial else None,
+                        "peak_hbm_gb": peak_gb,
+                        "act_ckpt": args.act_ckpt,
                     },
                 }
             ),
