@@ -13,7 +13,7 @@ N = 3
 F = 8
 
 
-def _d2_eval_body(rank, world, fused):
+def _d2_eval_body(rank, world, fused, method="vertical"):
     from mpi4dl_amd.comm import Communicator
     from mpi4dl_amd.models import resnet, resnet_spatial_d2
     from mpi4dl_amd.ops.halo import TileLayout
@@ -29,7 +29,7 @@ def _d2_eval_body(rank, world, fused):
     torch.manual_seed(0)
     serial = resnet.get_resnet_v2((1, 3, IMG, IMG), NCLS, n=N, num_filters=F)
     ncells = len(serial)
-    plan = SpatialPlan(comm, [ncells], "vertical")
+    plan = SpatialPlan(comm, [ncells], method)
     torch.manual_seed(0)
     d2 = resnet_spatial_d2.get_resnet_v2(
         (1, 3, IMG, IMG), NCLS, n=N, num_filters=F, plan=plan, fused_layers=fused
@@ -38,7 +38,7 @@ def _d2_eval_body(rank, world, fused):
     d2.eval()
     torch.manual_seed(9)
     x = torch.randn(2, 3, IMG, IMG)
-    layout = TileLayout(world, "vertical")
+    layout = TileLayout(world, method)
     with torch.no_grad():
         y_ser = x
         for cell in list(serial)[:-1]:
@@ -60,6 +60,12 @@ def test_d2_eval_exact_fused2():
 
 def test_d2_eval_exact_fused4():
     run_distributed(_d2_eval_body, 2, (4,))
+
+
+def test_d2_eval_exact_square():
+    """D2 fused halo on a 2x2 square grid: both axes pad interior sides
+    only and the corner strips route through the fused exchange."""
+    run_distributed(_d2_eval_body, 4, (2, "square"))
 
 
 def _d2_train_body(rank, world, steps):
