@@ -457,6 +457,34 @@ def halo_pad(x, h, exchanger: HaloExchanger, grad_mode: str = "exact", fill: flo
     return _HaloPadFn.apply(x, h, exchanger, grad_mode, fill)
 
 
+class _OverlapExactPadFn(torch.autograd.Function):
+    """F.pad only (the async exchange fills the ring outside autograd,
+    overlapped with the interior conv — spatial_conv._forward_overlap);
+    backward performs the exact transposed halo-gradient exchange, so
+    overlap mode and grad_mode='exact' compose.
+
+    Correctness: the band/interior decomposition computes the same
+    outputs as one conv over the padded tile, so d(loss)/d(pad) is
+    identical by linearity; the interior conv's input grad flows to x
+    directly (autograd sums it with this Function's cropped output)."""
+
+    @staticmethod
+    def forward(ctx, x, h, exchanger):
+        ctx.h = h
+        ctx.exchanger = exchanger
+        hh, hw = _hpair(h)
+        return F.pad(x, (hw, hw, hh, hh))
+
+    @staticmethod
+    def backward(ctx, gp):
+        if gp.is_meta:
+            hh, hw = _hpair(ctx.h)
+            H, W = gp.shape[-2] - 2 * hh, gp.shape[-1] - 2 * hw
+            return gp[:, :, hh : hh + H, hw : hw + W], None, None
+        g = ctx.exchanger.exchange_grad_padded(gp.contiguous(), ctx.h)
+        return g, None, None
+
+
 class _HaloPadD2Fn(torch.autograd.Function):
     """D2 pad: interior sides only (boundary sides stay unpadded — each
     conv re-applies its own zero pad there, reference spatial.py:67-111)."""

@@ -133,8 +133,7 @@ class HaloConv2d(_SpatialBase):
                 else (self.halo_len, self.halo_len)
             )
             if (
-                self.grad_mode == "drop"
-                and self.exchanger is not None
+                self.exchanger is not None
                 and not x.is_meta
                 and self.stride in (1, (1, 1))
                 and (hh or hw)
@@ -153,17 +152,24 @@ class HaloConv2d(_SpatialBase):
         ring is still on the wire; border bands are computed after the
         ring lands and the pieces are concatenated.
 
-        Reference ('drop') gradient semantics: the received ring is a
-        constant — weight grads still include the ring pixels (identical
-        to the blocking drop path); the exact transposed-grad mode uses
-        the blocking path instead.
+        Gradient semantics per grad_mode: 'drop' treats the received
+        ring as a constant (weight grads still include ring pixels —
+        identical to the blocking drop path); 'exact' routes the pad
+        through _OverlapExactPadFn, whose backward performs the
+        transposed halo-gradient exchange — same trajectory as the
+        blocking exact path, with the forward ring overlapped.
         """
         hh, hw = (
             self.halo_len
             if isinstance(self.halo_len, tuple)
             else (self.halo_len, self.halo_len)
         )
-        xp = F.pad(x, (hw, hw, hh, hh))
+        if self.grad_mode == "exact":
+            from .halo import _OverlapExactPadFn
+
+            xp = _OverlapExactPadFn.apply(x, (hh, hw), self.exchanger)
+        else:
+            xp = F.pad(x, (hw, hw, hh, hh))
         with torch.no_grad():
             finish = self.exchanger.exchange_padded_async(xp, (hh, hw))
         # interior while the halo is in flight
