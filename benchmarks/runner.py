@@ -101,12 +101,28 @@ def make_engines(args, mode):
         return gen
 
     def mkopt(module):
+        name = getattr(args, "optimizer", "sgd").lower()
+        if name == "adam":
+            return torch.optim.Adam(
+                module.parameters(), lr=args.learning_rate,
+                weight_decay=args.weight_decay,
+            )
+        if name == "adamw":
+            return torch.optim.AdamW(
+                module.parameters(), lr=args.learning_rate,
+                weight_decay=args.weight_decay,
+            )
+        if name != "sgd":
+            raise ValueError(f"--optimizer {name}: use sgd / adam / adamw")
         if on_gpu:
             from mpi4dl_amd.optim import FusedSGD
 
-            return FusedSGD(module, lr=args.learning_rate, momentum=args.momentum)
+            return FusedSGD(module, lr=args.learning_rate,
+                            momentum=args.momentum,
+                            weight_decay=args.weight_decay)
         return torch.optim.SGD(
-            module.parameters(), lr=args.learning_rate, momentum=args.momentum
+            module.parameters(), lr=args.learning_rate,
+            momentum=args.momentum, weight_decay=args.weight_decay,
         )
 
     eng_kw = dict(

@@ -57,7 +57,7 @@ def get_parser() -> argparse.ArgumentParser:
     p.add_argument("--num-gpus-mp", type=int, default=1,
                    help="compat flag (topology comes from WORLD_SIZE)")
     p.add_argument("--num-workers", type=int, default=0)
-    p.add_argument("--optimizer", default="sgd")
+    p.add_argument("--optimizer", default="sgd", choices=["sgd", "adam", "adamw"])
     p.add_argument("--learning-rate", type=float, default=0.001)
     p.add_argument("--weight-decay", type=float, default=1e-4)
     p.add_argument("--momentum", type=float, default=0.9)

@@ -99,3 +99,13 @@ def test_entry_sp_d2_resnet():
         timeout=300,
     )
     assert got[0] == 2
+
+
+def test_entry_lp_adamw():
+    # --optimizer / --weight-decay are honoured (adamw path)
+    got = run_distributed(
+        _run_mode, 2,
+        ("lp", "resnet",
+         ("--split-size", "2", "--optimizer", "adamw",
+          "--weight-decay", "0.01")))
+    assert got[0] == 2
