@@ -1,0 +1,56 @@
+"""Winograd F(2x2,3x3) oracle vs direct convolution — the numerical
+reference the round-2 MFMA Winograd kernel will be tested against."""
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+from mpi4dl_amd.ops.winograd_ref import filter_transform, winograd_conv2d_ref
+
+
+@pytest.mark.parametrize(
+    "N,C,K,H,W,pad",
+    [
+        (1, 1, 1, 4, 4, 1),
+        (2, 3, 8, 16, 16, 1),
+        (1, 4, 4, 15, 17, 1),   # odd sizes: tile-pad + crop
+        (1, 8, 16, 9, 9, 0),    # valid conv
+        (3, 16, 32, 32, 32, 1),
+    ],
+)
+def test_matches_direct_conv(N, C, K, H, W, pad):
+    torch.manual_seed(0)
+    x = torch.randn(N, C, H, W)
+    w = torch.randn(K, C, 3, 3) * 0.2
+    b = torch.randn(K)
+    ref = F.conv2d(x, w, b, stride=1, padding=pad)
+    got = winograd_conv2d_ref(x, w, b, padding=pad)
+    assert got.shape == ref.shape, (got.shape, ref.shape)
+    err = (got - ref).abs().max() / ref.abs().max()
+    assert err < 1e-5, float(err)
+
+
+def test_bf16_inputs_fp32_transforms():
+    """The kernel's numeric plan: bf16 data, fp32 transforms/accum —
+    error must stay within direct bf16 conv tolerance."""
+    torch.manual_seed(1)
+    x = torch.randn(2, 16, 32, 32)
+    w = torch.randn(32, 16, 3, 3) * 0.1
+    ref = F.conv2d(x, w, None, stride=1, padding=1)
+    got = winograd_conv2d_ref(
+        x.to(torch.bfloat16).float(), w.to(torch.bfloat16).float(), None, 1
+    )
+    rel = (got - ref).abs().max() / ref.abs().max()
+    assert rel < 0.03, float(rel)
+
+
+def test_filter_transform_shape_and_linearity():
+    torch.manual_seed(2)
+    w1 = torch.randn(4, 3, 3, 3)
+    w2 = torch.randn(4, 3, 3, 3)
+    U = filter_transform(w1)
+    assert U.shape == (4, 3, 4, 4)
+    # transform is linear: U(w1 + w2) = U(w1) + U(w2)
+    assert torch.allclose(
+        filter_transform(w1 + w2), U + filter_transform(w2), atol=1e-6
+    )
