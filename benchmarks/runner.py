@@ -40,6 +40,11 @@ log = logging.getLogger("mpi4dl_amd.benchmark")
 def build_model(args, plan, mb):
     shape = (mb, 3, args.image_size, args.image_size)
     torch.manual_seed(0)
+    if args.model == "resnet101":
+        # BASELINE config 4: ResNet-101 SP+PP at high resolution
+        from mpi4dl_amd.models import resnet_spatial as M
+
+        return M.get_resnet101_cells(shape, args.num_classes, plan=plan)
     if args.model == "resnet":
         if args.halo_d2:
             from mpi4dl_amd.models import resnet_spatial_d2 as M

@@ -109,3 +109,16 @@ def test_entry_lp_adamw():
          ("--split-size", "2", "--optimizer", "adamw",
           "--weight-decay", "0.01")))
     assert got[0] == 2
+
+
+def test_entry_sp_resnet101():
+    # BASELINE config 4 shape: ResNet-101 cells with SP+PP (tiny size)
+    got = run_distributed(
+        _run_mode, 3,
+        ("sp", "resnet101",
+         ("--split-size", "2", "--num-spatial-parts", "2",
+          "--spatial-size", "1", "--slice-method", "vertical",
+          "--image-size", "64", "--batch-size", "2", "--parts", "1")),
+        timeout=300,
+    )
+    assert got[0] == 2
