@@ -40,6 +40,13 @@ log = logging.getLogger("mpi4dl_amd.benchmark")
 def build_model(args, plan, mb):
     shape = (mb, 3, args.image_size, args.image_size)
     torch.manual_seed(0)
+    if args.model == "resnet18":
+        # BASELINE config 1: ResNet-18 layer parallelism (no spatial variant
+        # — LP/PP/GEMS/DP only)
+        assert plan is None, "resnet18 is LP-only; use resnet/resnet101 for SP"
+        from mpi4dl_amd.models.resnet import get_resnet18_cells
+
+        return get_resnet18_cells(shape, args.num_classes)
     if args.model == "resnet101":
         # BASELINE config 4: ResNet-101 SP+PP at high resolution
         from mpi4dl_amd.models import resnet_spatial as M

@@ -169,6 +169,30 @@ def get_resnet_v2(
     return model
 
 
+def get_resnet18_cells(
+    input_shape,
+    num_classes: int = 1000,
+    width: int = 64,
+    device: Optional[str] = None,
+) -> nn.Sequential:
+    """ImageNet-style ResNet-18 as flat cells (BASELINE config 1:
+    ResNet-18 layer parallelism at 224²). Basic-block counts 2-2-2-2."""
+    _, in_ch, H, W = input_shape
+    cells = [_stem(in_ch, width, min(H, W))]
+    ch = width
+    for group, blocks in enumerate([2, 2, 2, 2]):
+        out = width * (2**group)
+        for b in range(blocks):
+            stride = 2 if (group > 0 and b == 0) else 1
+            cells.append(BasicBlockV1(ch, out, stride))
+            ch = out
+    cells.append(Head(ch, num_classes, final_bn=False))
+    model = nn.Sequential(*cells)
+    if device is not None:
+        model = model.to(device)
+    return model
+
+
 def get_resnet101_cells(
     input_shape,
     num_classes: int = 1000,

@@ -19,7 +19,7 @@ def get_parser() -> argparse.ArgumentParser:
     p.add_argument("--fp16-allreduce", action="store_true",
                    help="compat flag (gradient allreduce already bucketed)")
     p.add_argument("--model", default="resnet",
-                   choices=["resnet", "resnet101", "amoebanet"])
+                   choices=["resnet", "resnet18", "resnet101", "amoebanet"])
     p.add_argument("--batch-size", type=int, default=8)
     p.add_argument("--parts", type=int, default=4,
                    help="pipeline micro-batches per step")
