@@ -48,3 +48,20 @@ def test_config3_amoebanet_gems_8stages():
         timeout=600,
     )
     assert got[0] == 2
+
+
+def test_config5_amoebanet_sp_gems_8ranks():
+    """SP+GEMS+PP at the config-5 topology: 4 spatial tiles + LP stages,
+    mp = 5 + 4 - 1 = 8 >= 2*4 tiles (mirror disjointness), two GEMS
+    engines per rank. 8192^2 sizing is the GPU-scale variant; the
+    schedule/seam/mirror wiring is what this validates."""
+    got = run_distributed(
+        _run_mode, 8,
+        ("gems_sp",
+         ("--model", "resnet", "--image-size", "128", "--batch-size", "4",
+          "--parts", "2", "--split-size", "5", "--num-spatial-parts", "4",
+          "--spatial-size", "1", "--slice-method", "vertical",
+          "--num-layers", "9", "--num-filters", "4", "--num-classes", "10")),
+        timeout=600,
+    )
+    assert got[0] == 2
