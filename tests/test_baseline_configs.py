@@ -58,10 +58,10 @@ def test_config5_amoebanet_sp_gems_8ranks():
     got = run_distributed(
         _run_mode, 8,
         ("gems_sp",
-         ("--model", "resnet", "--image-size", "128", "--batch-size", "4",
+         ("--model", "amoebanet", "--image-size", "128", "--batch-size", "4",
           "--parts", "2", "--split-size", "5", "--num-spatial-parts", "4",
           "--spatial-size", "1", "--slice-method", "vertical",
-          "--num-layers", "9", "--num-filters", "4", "--num-classes", "10")),
+          "--num-layers", "6", "--num-filters", "8", "--num-classes", "10")),
         timeout=600,
     )
     assert got[0] == 2
