@@ -38,6 +38,12 @@ if any(a.startswith("--image-size") for a in sys.argv):
             os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
     except (IndexError, ValueError):
         pass
+# Multi-rank runs: 8 processes running the exhaustive find concurrently
+# contend on the shared find-db file locks (serialized minutes-long finds
+# can outlive the warmup). FAST keeps the multi-GPU warmup bounded; the
+# N=1 headline still runs the full find.
+if int(os.environ.get("WORLD_SIZE", "1")) > 1:
+    os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
 
 import torch
 import torch.distributed as dist
