@@ -122,3 +122,11 @@ def test_entry_sp_resnet101():
         timeout=300,
     )
     assert got[0] == 2
+
+
+def test_entry_gems_times2():
+    # --times 2: two replica pairs per step (4x batch consumed)
+    got = run_distributed(
+        _run_mode, 2,
+        ("gems", "resnet", ("--split-size", "2", "--times", "2")))
+    assert got[0] == 2

@@ -114,6 +114,16 @@ def test_make_dataloader_wiring(tmp_path):
     assert xb.shape == (2, 3, 8, 8)
 
 
+def test_dataloader_num_workers(tmp_path):
+    from mpi4dl_amd.data import make_dataloader
+
+    root, _, _ = _make_cifar(str(tmp_path), n=12)
+    loader, n = make_dataloader(2, root, batch_size=4, image_size=32,
+                                num_classes=10, num_workers=2)
+    batches = list(loader)
+    assert len(batches) == 3 and batches[0][0].shape == (4, 3, 32, 32)
+
+
 def _runner_body(rank, world, root):
     import sys
 
