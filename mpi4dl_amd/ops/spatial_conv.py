@@ -24,6 +24,8 @@ from __future__ import annotations
 
 from typing import Optional
 
+import os
+
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
@@ -140,6 +142,8 @@ class HaloConv2d(_SpatialBase):
                 # interior conv must be valid: tile bigger than the kernel
                 and x.shape[-2] > 2 * hh
                 and x.shape[-1] > 2 * hw
+                # debugging knob: force the blocking exchange path
+                and os.environ.get("MPI4DL_NO_OVERLAP", "0") != "1"
             ):
                 return self._forward_overlap(x)
             xp = halo_pad(x, self.halo_len, self.exchanger, self.grad_mode)

@@ -320,3 +320,60 @@ def test_sp_act_ckpt_parity():
     got = run_distributed(_sp_act_ckpt_body, 3, (steps, batch, parts, lr))[-1]
     for e, g in zip(expected, got):
         assert abs(e - g) < 2e-4, (expected, got)
+
+
+def _sp_drop_body(rank, world, steps, batch, parts, lr, no_overlap):
+    import os
+
+    os.environ["MPI4DL_NO_OVERLAP"] = "1" if no_overlap else "0"
+    from mpi4dl_amd.comm import Communicator, GradReducer
+    from mpi4dl_amd.models import resnet_spatial
+    from mpi4dl_amd.ops.plan import SpatialPlan
+    from mpi4dl_amd.parallel.partition import model_generator
+    from mpi4dl_amd.parallel.spatial import train_model_spatial
+
+    comm = Communicator(
+        split_size=2, ENABLE_SPATIAL=True, num_spatial_parts=2,
+        spatial_size=1, backend="gloo",
+    )
+    torch.manual_seed(0)
+    probe = resnet_spatial.get_resnet_v1((1, 3, IMG, IMG), NCLS, n=1, num_filters=8)
+    base, rem = divmod(len(probe), 2)
+    balance = [base + (1 if i < rem else 0) for i in range(2)]
+    plan = SpatialPlan(comm, balance, "vertical", grad_mode="drop")
+    torch.manual_seed(0)
+    model = resnet_spatial.get_resnet_v1(
+        (batch // parts, 3, IMG, IMG), NCLS, n=1, num_filters=8, plan=plan
+    )
+    gen = model_generator(
+        model, 2, input_size=(batch // parts, 3, IMG, IMG), balance=balance
+    )
+    gen.get_output_shapes()
+    gen.ready_model(comm.split_rank, device=torch.device("cpu"))
+    opt = torch.optim.SGD(gen.models.parameters(), lr=lr, momentum=0.9)
+    eng = train_model_spatial(
+        gen, comm.local_rank, batch, parts, comm, slice_method="vertical",
+        optimizer=opt, grad_mode="drop", device=torch.device("cpu"),
+    )
+    red = GradReducer(comm)
+    torch.manual_seed(42)
+    losses = []
+    for _ in range(steps):
+        x = torch.randn(batch, 3, IMG, IMG)
+        y = torch.randint(0, NCLS, (batch,))
+        loss, _, _ = eng.run_step(x, y)
+        red.apply_allreduce(eng.models)
+        eng.update()
+        losses.append(loss)
+    return losses
+
+
+def test_drop_overlap_equals_blocking():
+    """grad_mode='drop': the halo/compute-overlap trajectory must be
+    identical to the blocking-exchange trajectory (MPI4DL_NO_OVERLAP=1)
+    — overlap is a pure scheduling change."""
+    args = (2, 4, 2, 0.01)
+    overlap = run_distributed(_sp_drop_body, 3, args + (False,))[-1]
+    blocking = run_distributed(_sp_drop_body, 3, args + (True,))[-1]
+    for a, b in zip(overlap, blocking):
+        assert abs(a - b) < 1e-6, (overlap, blocking)
