@@ -262,6 +262,11 @@ def run_training(args, mode):
     if args.resume and args.checkpoint_dir:
         ckpt.load_checkpoint(args.checkpoint_dir, extras["gen"].models,
                              extras.get("optimizer"), comm)
+        eng = extras["engine"]
+        if hasattr(eng, "sync_models"):
+            # GEMS: the checkpoint restored replica 1 only — mirror it
+            # onto replica 2 (otherwise the replicas diverge from step 0)
+            eng.sync_models()
 
     times = []
     for epoch in range(args.num_epochs):

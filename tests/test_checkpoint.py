@@ -139,3 +139,64 @@ def _reshard_body(rank, world, tmpdir):
 def test_reshard_via_consolidate(tmp_path):
     run_distributed(_consolidate_body, 3, (str(tmp_path),))
     run_distributed(_reshard_body, 2, (str(tmp_path),))
+
+
+def _gems_ckpt_body(rank, world, tmpdir, phase):
+    """phase 'save': train 1 step, checkpoint. phase 'resume': restore,
+    verify BOTH replica engines hold the checkpointed weights."""
+    import sys
+
+    sys.path.insert(0, os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "benchmarks"))
+    from runner import make_engines
+
+    from mpi4dl_amd import checkpoint as ckpt
+    from mpi4dl_amd.parser import get_parser
+
+    args = get_parser().parse_args([
+        "--model", "resnet", "--batch-size", "2", "--parts", "1",
+        "--split-size", "2", "--image-size", "32", "--num-layers", "9",
+        "--num-filters", "4", "--num-classes", "10", "--backend", "gloo",
+    ])
+    step, comm, extras = make_engines(args, "gems")
+    eng = extras["engine"]
+    if phase == "save":
+        torch.manual_seed(42)
+        x = torch.randn(4, 3, 32, 32)  # 2x batch for the two replicas
+        y = torch.randint(0, 10, (4,))
+        step(x, y)
+        ckpt.save_checkpoint(tmpdir, extras["gen"].models, None, comm)
+        # numpy: torch tensors ride the queue via shm FDs that die with
+        # the child process
+        return [p.detach().numpy().copy()
+                for p in eng.train_model1.models.parameters()]
+    # resume path (fresh random init on purpose: different seed)
+    ckpt.load_checkpoint(tmpdir, extras["gen"].models, None, comm)
+    eng.sync_models()
+    p1 = [p.detach().numpy().copy()
+          for p in eng.train_model1.models.parameters()]
+    p2 = [p.detach().numpy().copy()
+          for p in eng.train_model2.models.parameters()]
+    return p1, p2
+
+
+def _gems_save(rank, world, tmpdir):
+    return _gems_ckpt_body(rank, world, tmpdir, "save")
+
+
+def _gems_resume(rank, world, tmpdir):
+    return _gems_ckpt_body(rank, world, tmpdir, "resume")
+
+
+def test_gems_resume_restores_both_replicas(tmp_path):
+    saved = run_distributed(_gems_save, 2, (str(tmp_path),))
+    out = run_distributed(_gems_resume, 2, (str(tmp_path),))
+    for rank in range(2):
+        p1, p2 = out[rank]
+        # replica 1 == what rank `rank` checkpointed
+        for a, b in zip(p1, saved[rank]):
+            assert (a == b).all()
+        # replica 2 on rank r == replica 1's stage from the mirror rank
+        mirror = 1 - rank
+        for a, b in zip(p2, saved[mirror]):
+            assert (a == b).all()
