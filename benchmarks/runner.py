@@ -302,10 +302,18 @@ def run_training(args, mode):
         if args.app == 2:
             from mpi4dl_amd.data import make_dataloader
 
-            ev_loader, _ = make_dataloader(
-                2, args.datapath, Be, S, args.num_classes,
-                num_workers=args.num_workers, train=False,
-            )
+            try:
+                ev_loader, _ = make_dataloader(
+                    2, args.datapath, Be, S, args.num_classes,
+                    num_workers=args.num_workers, train=False,
+                )
+            except FileNotFoundError:
+                log.warning("no CIFAR test_batch on disk; evaluating on "
+                            "the train split")
+                ev_loader, _ = make_dataloader(
+                    2, args.datapath, Be, S, args.num_classes,
+                    num_workers=args.num_workers, train=True,
+                )
             ev_batches = list(ev_loader)[: max(args.num_steps, 1)]
         else:
             g = torch.Generator().manual_seed(7)

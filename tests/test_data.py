@@ -138,6 +138,7 @@ def _runner_body(rank, world, root):
         "--batch-size", "4", "--parts", "2", "--split-size", str(world),
         "--image-size", "32", "--num-layers", "9", "--num-filters", "4",
         "--num-epochs", "1", "--num-steps", "2", "--backend", "gloo",
+        "--enable-evaluation",  # no test_batch on disk -> train-split fallback
     ])
     times = run_training(args, "lp")
     return len(times)
