@@ -202,6 +202,10 @@ class train_model_spatial(train_model):
         """My stage's micro-batch size (LP stages under local-DP see 1/L)."""
         L = self.comm.LOCAL_DP_LP
         if not self.is_tile_rank and L > 1:
+            assert self.mb % L == 0, (
+                f"micro-batch {self.mb} not divisible by LOCAL_DP_LP {L}: "
+                "raise batch_size or lower parts/local-DP"
+            )
             return self.mb // L
         return self.mb
 
