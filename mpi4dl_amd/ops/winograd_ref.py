@@ -100,3 +100,46 @@ def winograd_conv2d_ref(
     if bias is not None:
         out = out + bias.float().view(1, -1, 1, 1)
     return out
+
+
+def winograd_bmm_conv2d(
+    x: torch.Tensor,
+    U: torch.Tensor,
+    bias: torch.Tensor = None,
+    padding: int = 1,
+    out_dtype: torch.dtype = None,
+) -> torch.Tensor:
+    """GPU-capable F(2x2,3x3) conv: transforms in torch, the 16 GEMMs as
+    ONE batched matmul (rides hipBLASLt on ROCm — a plain library GEMM).
+
+    U is the precomputed filter transform (filter_transform(w), shape
+    [K, C, 4, 4]) — amortised across steps in inference / recomputed
+    once per step in training. The round-2 plan fuses the input/output
+    transforms into HIP kernels around the same bmm; this version is the
+    drop-in correctness/perf staging point (and the A/B baseline).
+
+    Accumulation: matmul in x.dtype (bf16 on GPU -> MFMA with fp32
+    accumulate inside hipBLASLt); transforms in fp32.
+    """
+    N, C, H, W = x.shape
+    K = U.shape[0]
+    OH, OW = H + 2 * padding - 2, W + 2 * padding - 2
+    TH, TW = (OH + 1) // 2, (OW + 1) // 2
+    xin = F.pad(
+        x, (padding, 2 * TW + 2 - W - padding, padding, 2 * TH + 2 - H - padding)
+    )
+    d = xin.unfold(2, 4, 2).unfold(3, 4, 2).float()  # [N,C,TH,TW,4,4]
+    bt = BT.to(d.device)
+    V = torch.einsum("ir,ncturs,js->ijnctu", bt, d, bt)  # [4,4,N,C,TH,TW]
+    T = N * TH * TW
+    Vm = V.reshape(16, N, C, TH * TW).permute(0, 2, 1, 3).reshape(16, C, T)
+    Um = U.reshape(K, C, 16).permute(2, 0, 1)  # [16, K, C]
+    cd = x.dtype if x.is_cuda else torch.float32
+    M = torch.bmm(Um.to(cd), Vm.to(cd)).float()  # [16, K, T]
+    M = M.reshape(4, 4, K, N, TH, TW)
+    at = AT.to(M.device)
+    Y = torch.einsum("pi,ijkntu,qj->kntupq", at, M, at)
+    out = Y.permute(1, 0, 2, 4, 3, 5).reshape(N, K, 2 * TH, 2 * TW)[:, :, :OH, :OW]
+    if bias is not None:
+        out = out + bias.float().view(1, -1, 1, 1)
+    return out.to(out_dtype) if out_dtype is not None else out

@@ -316,3 +316,21 @@ def test_act_ckpt_memory_and_parity():
     # recompute must shrink held activations (whole-step peak incl.
     # weights/grads: expect at least ~20% lower)
     assert peak_b < peak_a * 0.8, (peak_a, peak_b)
+
+
+@gpu
+@requires_gpu
+def test_winograd_bmm_gpu_bf16():
+    """F(2x2,3x3) via batched hipBLASLt GEMM on MFMA: bf16 compute,
+    fp32 transforms — within bf16 direct-conv tolerance (the round-2
+    fused-kernel path A/Bs against this)."""
+    from mpi4dl_amd.ops.winograd_ref import filter_transform, winograd_bmm_conv2d
+
+    torch.manual_seed(0)
+    x = torch.randn(2, 64, 128, 128, device="cuda", dtype=torch.bfloat16)
+    w = (torch.randn(64, 64, 3, 3, device="cuda") * 0.1)
+    ref = torch.nn.functional.conv2d(x.float(), w, None, stride=1, padding=1)
+    U = filter_transform(w)
+    got = winograd_bmm_conv2d(x, U, None, padding=1)
+    rel = (got - ref).abs().max() / ref.abs().max()
+    assert rel < 0.05, float(rel)

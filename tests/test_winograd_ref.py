@@ -54,3 +54,23 @@ def test_filter_transform_shape_and_linearity():
     assert torch.allclose(
         filter_transform(w1 + w2), U + filter_transform(w2), atol=1e-6
     )
+
+
+@pytest.mark.parametrize(
+    "N,C,K,H,W,pad", [(2, 8, 16, 17, 15, 1), (1, 16, 8, 32, 32, 1),
+                      (2, 4, 4, 10, 10, 0)]
+)
+def test_bmm_variant_matches_direct(N, C, K, H, W, pad):
+    """The batched-GEMM formulation (the one that rides hipBLASLt on
+    GPU) must match direct conv too."""
+    from mpi4dl_amd.ops.winograd_ref import winograd_bmm_conv2d
+
+    torch.manual_seed(3)
+    x = torch.randn(N, C, H, W)
+    w = torch.randn(K, C, 3, 3) * 0.2
+    b = torch.randn(K)
+    ref = F.conv2d(x, w, b, stride=1, padding=pad)
+    got = winograd_bmm_conv2d(x, filter_transform(w), b, padding=pad)
+    assert got.shape == ref.shape
+    err = (got - ref).abs().max() / ref.abs().max()
+    assert err < 1e-5, float(err)
