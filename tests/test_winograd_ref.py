@@ -74,3 +74,32 @@ def test_bmm_variant_matches_direct(N, C, K, H, W, pad):
     assert got.shape == ref.shape
     err = (got - ref).abs().max() / ref.abs().max()
     assert err < 1e-5, float(err)
+
+
+def test_winograd_dispatch_env(monkeypatch):
+    """MPI4DL_WINOGRAD=1 routes NativeConv2d 3x3/s1 through the
+    batched-GEMM winograd path — forward AND backward must match
+    nn.Conv2d (backward falls out of autograd over the transforms)."""
+    monkeypatch.setenv("MPI4DL_WINOGRAD", "1")
+    from mpi4dl_amd.ops.conv_native import NativeConv2d
+
+    torch.manual_seed(0)
+    m = NativeConv2d(8, 16, 3, padding=1)
+    ref = torch.nn.Conv2d(8, 16, 3, padding=1)
+    ref.load_state_dict(m.state_dict())
+    x = torch.randn(2, 8, 16, 16, requires_grad=True)
+    x2 = x.detach().clone().requires_grad_(True)
+    y, y2 = m(x), ref(x2)
+    assert (y - y2).abs().max() < 1e-5
+    g = torch.randn_like(y)
+    y.backward(g)
+    y2.backward(g)
+    assert (x.grad - x2.grad).abs().max() < 1e-5
+    assert (m.weight.grad - ref.weight.grad).abs().max() < 1e-3
+    assert (m.bias.grad - ref.bias.grad).abs().max() < 1e-3
+    # 5x5 kernels must NOT take the winograd path
+    m5 = NativeConv2d(4, 4, 5, padding=2)
+    r5 = torch.nn.Conv2d(4, 4, 5, padding=2)
+    r5.load_state_dict(m5.state_dict())
+    xx = torch.randn(1, 4, 12, 12)
+    assert (m5(xx) - r5(xx)).abs().max() < 1e-6
