@@ -38,11 +38,44 @@ def timeit(fn, iters, warmup=5):
     return (time.perf_counter() - t0) / iters * 1e3
 
 
+def winograd_ab(args):
+    """F(2x2,3x3) batched-GEMM staging vs MIOpen at the bench 3x3 shapes."""
+    from mpi4dl_amd.ops.winograd_ref import filter_transform, winograd_bmm_conv2d
+
+    print(f"{'shape':<42} {'winograd':>9} {'miopen':>9} {'ratio':>6}")
+    for C, K, H, W, kh, kw, s, ph, pw in SHAPES:
+        if (kh, kw) != (3, 3) or s != 1:
+            continue
+        x = torch.randn(args.batch, C, H, W, device="cuda", dtype=torch.bfloat16)
+        w = torch.randn(K, C, 3, 3, device="cuda") * 0.05
+        U = filter_transform(w)
+        tw = timeit(lambda: winograd_bmm_conv2d(x, U, None, padding=ph),
+                    args.iters)
+        tm = timeit(
+            lambda: torch.nn.functional.conv2d(
+                x, w.to(torch.bfloat16), stride=1, padding=(ph, pw)
+            ),
+            args.iters,
+        )
+        y = winograd_bmm_conv2d(x, U, None, padding=ph)
+        ref = torch.nn.functional.conv2d(
+            x.float(), w, stride=1, padding=(ph, pw)
+        )
+        rel = (y - ref).abs().max().item() / max(ref.abs().max().item(), 1e-3)
+        tag = f"C{C}->K{K} {H}x{W} k3x3"
+        print(f"{tag:<42} {tw:8.3f}ms {tm:8.3f}ms {tm/tw:5.2f}x  relerr {rel:.3g}")
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--iters", type=int, default=20)
     ap.add_argument("--batch", type=int, default=1)
+    ap.add_argument("--winograd", action="store_true",
+                    help="A/B winograd_bmm_conv2d vs MIOpen on the 3x3 "
+                         "stride-1 shapes (round-2 staging)")
     args = ap.parse_args()
+    if args.winograd:
+        return winograd_ab(args)
     from mpi4dl_amd.ops import backend
 
     ge = backend.ext()
