@@ -5,6 +5,8 @@ import pytest
 import torch
 import torch.nn.functional as F
 
+from dist_util import run_distributed
+
 from mpi4dl_amd.ops.winograd_ref import filter_transform, winograd_conv2d_ref
 
 
@@ -103,3 +105,27 @@ def test_winograd_dispatch_env(monkeypatch):
     r5.load_state_dict(m5.state_dict())
     xx = torch.randn(1, 4, 12, 12)
     assert (m5(xx) - r5(xx)).abs().max() < 1e-6
+
+
+def _sp_winograd_body(rank, world, steps, batch, parts, lr):
+    import os
+
+    os.environ["MPI4DL_WINOGRAD"] = "1"
+    import test_spatial_engine as T
+
+    return T._spatial_body(
+        rank, world, steps, batch, parts, lr, "vertical", 2, 1, 2, 1
+    )
+
+
+def test_winograd_composes_with_spatial_engine():
+    """MPI4DL_WINOGRAD=1 under the full SP engine (halo-padded 3x3
+    convs route through the batched-GEMM path): trajectory still tracks
+    serial training."""
+    import test_spatial_engine as T
+
+    steps, batch, parts, lr = 2, 2, 1, 0.01
+    expected = T._serial_losses(steps, batch, parts, lr)
+    got = run_distributed(_sp_winograd_body, 3, (steps, batch, parts, lr))[-1]
+    for e, g in zip(expected, got):
+        assert abs(e - g) < 5e-4, (expected, got)
