@@ -172,40 +172,44 @@ class train_model_master:
         self.train_model2.update()
 
     def _paired_swap(self):
-        """MASTER-OPT: flat-grad P2P swap with the mirror rank instead of
-        a pair allreduce (reference send_recv_grads,
-        train_spatial_master.py:296-325)."""
-        comm = self.comm
-        r = comm.rank % comm.mp_size
-        mirror = comm.mp_size - 1 - r
-        fg1 = self.reducer.flat(self.train_model1.models)
-        fg2 = self.reducer.flat(self.train_model2.models)
-        if mirror == r:
-            if fg1.buffer.numel():
-                mean = (fg1.buffer + fg2.buffer) / 2
-                fg1.buffer.copy_(mean)
-                fg2.buffer.copy_(mean)
-            return
-        peer = comm.global_rank(mirror)
-        # my model1 grads pair with mirror's model2 grads (same stage)
-        r1 = torch.empty_like(fg1.buffer)
-        r2 = torch.empty_like(fg2.buffer)
-        # tags keyed by STAGE index (my model1 holds stage r, my model2
-        # stage `mirror`): the pair exchanges stage min(r,mirror) first on
-        # both sides, so gloo tags AND RCCL issue order both line up.
-        first = r < mirror
-        seq = (
-            [(fg1.buffer, r1, 7100), (fg2.buffer, r2, 7101)]
-            if first
-            else [(fg2.buffer, r2, 7100), (fg1.buffer, r1, 7101)]
-        )
-        # both sides order the two swaps by stage index (min stage first),
-        # so RCCL's order-based pairing matches: my model1<->mirror model2
-        # then my model2<->mirror model1.
-        for send_buf, recv_buf, tag in seq:
-            p2p.exchange([(send_buf, peer, tag)], [(recv_buf, peer, tag)]).wait()
-        fg1.buffer.add_(r1).mul_(0.5)
-        fg2.buffer.add_(r2).mul_(0.5)
+        _paired_grad_swap(self.comm, self.reducer,
+                          self.train_model1.models, self.train_model2.models)
+
+
+def _paired_grad_swap(comm, reducer, models1, models2):
+    """MASTER-OPT: flat-grad P2P swap with the mirror rank instead of
+    a pair allreduce (reference send_recv_grads,
+    train_spatial_master.py:296-325)."""
+    r = comm.rank % comm.mp_size
+    mirror = comm.mp_size - 1 - r
+    fg1 = reducer.flat(models1)
+    fg2 = reducer.flat(models2)
+    if mirror == r:
+        if fg1.buffer.numel():
+            mean = (fg1.buffer + fg2.buffer) / 2
+            fg1.buffer.copy_(mean)
+            fg2.buffer.copy_(mean)
+        return
+    peer = comm.global_rank(mirror)
+    # my model1 grads pair with mirror's model2 grads (same stage)
+    r1 = torch.empty_like(fg1.buffer)
+    r2 = torch.empty_like(fg2.buffer)
+    # tags keyed by STAGE index (my model1 holds stage r, my model2
+    # stage `mirror`): the pair exchanges stage min(r,mirror) first on
+    # both sides, so gloo tags AND RCCL issue order both line up.
+    first = r < mirror
+    seq = (
+        [(fg1.buffer, r1, 7100), (fg2.buffer, r2, 7101)]
+        if first
+        else [(fg2.buffer, r2, 7100), (fg1.buffer, r1, 7101)]
+    )
+    # both sides order the two swaps by stage index (min stage first),
+    # so RCCL's order-based pairing matches: my model1<->mirror model2
+    # then my model2<->mirror model1.
+    for send_buf, recv_buf, tag in seq:
+        p2p.exchange([(send_buf, peer, tag)], [(recv_buf, peer, tag)]).wait()
+    fg1.buffer.add_(r1).mul_(0.5)
+    fg2.buffer.add_(r2).mul_(0.5)
 
 
 def _slc(t, off, n):
@@ -291,9 +295,15 @@ class train_spatial_model_master:
                 self.train_model2.split_rank
             )
             red.allreduce_grads(self.train_model2.models, g, divide_by=1.0)
-        red.apply_allreduce_master(
-            self.train_model1.models, self.train_model2.models
-        )
+        if self.enable_comm_opt:
+            # MASTER-OPT: the mirror-pair exchange as a flat P2P swap
+            # (reference run_step_allreduce, train_spatial_master.py:327)
+            _paired_grad_swap(comm, red, self.train_model1.models,
+                              self.train_model2.models)
+        else:
+            red.apply_allreduce_master(
+                self.train_model1.models, self.train_model2.models
+            )
         red.allreduce_grads(self.train_model1.models, comm.outer_dp_group)
         red.allreduce_grads(self.train_model2.models, comm.outer_dp_group)
         self.train_model1.update()

@@ -100,7 +100,7 @@ def test_gems_lp_parity_comm_opt():
         assert abs(e - g) < 2e-4, (expected, combined)
 
 
-def _gems_sp_body(rank, world, steps, B, parts, lr):
+def _gems_sp_body(rank, world, steps, B, parts, lr, comm_opt=False):
     from mpi4dl_amd.comm import Communicator
     from mpi4dl_amd.models import resnet_spatial
     from mpi4dl_amd.ops.plan import SpatialPlan
@@ -138,7 +138,8 @@ def _gems_sp_body(rank, world, steps, B, parts, lr):
     gen1 = mkgen(r, False)
     gen2 = mkgen(comm.mp_size - 1 - r, True)
     eng = train_spatial_model_master(
-        gen1, gen2, B, parts, comm, slice_method="vertical", lr=lr
+        gen1, gen2, B, parts, comm, slice_method="vertical", lr=lr,
+        enable_comm_opt=comm_opt,
     )
     torch.manual_seed(42)
     losses = []
@@ -156,6 +157,18 @@ def test_gems_sp_parity():
     expected = _serial_losses(steps, B, parts, lr)
     # split 3, 2 tiles -> mp = 4; engine2 tiles on mirrored ranks 3,2
     got = run_distributed(_gems_sp_body, 4, (steps, B, parts, lr))
+    combined = [sum(g[s] for g in got) for s in range(steps)]
+    for e, g in zip(expected, combined):
+        assert abs(e - g) < 2e-4, (expected, combined)
+
+
+def test_gems_sp_parity_comm_opt():
+    """MASTER-OPT in the SPATIAL master: the flat-grad mirror swap (now
+    actually taken when --enable-master-comm-opt) must produce the same
+    trajectory as the pair allreduce."""
+    steps, B, parts, lr = 2, 2, 1, 0.01
+    expected = _serial_losses(steps, B, parts, lr)
+    got = run_distributed(_gems_sp_body, 4, (steps, B, parts, lr, True))
     combined = [sum(g[s] for g in got) for s in range(steps)]
     for e, g in zip(expected, combined):
         assert abs(e - g) < 2e-4, (expected, combined)
