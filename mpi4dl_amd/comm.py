@@ -281,6 +281,11 @@ class Communicator:
                     a, b = j, self.mp_size - 1 - j
                     if a > b:
                         continue
+                    if a == b:
+                        # odd mp_size: the middle rank hosts BOTH engines'
+                        # copy of its stage — no group; the reducer
+                        # averages the pair locally (group=None path)
+                        continue
                     g = self._new_group([rep * self.mp_size + a, rep * self.mp_size + b])
                     if rep == self.replica:
                         self.master_pair_groups[a] = g

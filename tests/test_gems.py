@@ -172,3 +172,15 @@ def test_gems_sp_parity_comm_opt():
     combined = [sum(g[s] for g in got) for s in range(steps)]
     for e, g in zip(expected, combined):
         assert abs(e - g) < 2e-4, (expected, combined)
+
+
+def test_gems_lp_parity_odd_stages():
+    """GEMS with an ODD stage count (mp=3): the middle rank hosts both
+    replicas' copy of its stage and must average the pair locally (a
+    {r, r} process group is invalid). Trajectory == DP over 2x batch."""
+    steps, B, parts, lr = 2, 2, 1, 0.01
+    expected = _serial_losses(steps, B, parts, lr)
+    got = run_distributed(_gems_body, 3, (steps, B, parts, lr, False))
+    combined = [sum(g[s] for g in got) for s in range(steps)]
+    for e, g in zip(expected, combined):
+        assert abs(e - g) < 2e-4, (expected, combined)
