@@ -334,3 +334,38 @@ def test_winograd_bmm_gpu_bf16():
     got = winograd_bmm_conv2d(x, U, None, padding=1)
     rel = (got - ref).abs().max() / ref.abs().max()
     assert rel < 0.05, float(rel)
+
+
+@gpu
+@requires_gpu
+def test_fused_sgd_matches_torch():
+    """One-kernel momentum SGD (with weight decay) over the flat
+    buffers == torch.optim.SGD over 3 steps."""
+    from mpi4dl_amd.optim import FusedSGD
+
+    torch.manual_seed(0)
+    m1 = torch.nn.Sequential(
+        torch.nn.Conv2d(3, 7, 3, padding=1), torch.nn.Linear(5, 11)
+    ).cuda()
+    m2 = torch.nn.Sequential(
+        torch.nn.Conv2d(3, 7, 3, padding=1), torch.nn.Linear(5, 11)
+    ).cuda()
+    m2.load_state_dict(m1.state_dict())
+    o1 = FusedSGD(m1, lr=0.05, momentum=0.9, weight_decay=0.01)
+    o2 = torch.optim.SGD(m2.parameters(), lr=0.05, momentum=0.9,
+                         weight_decay=0.01)
+    for step in range(3):
+        torch.manual_seed(10 + step)
+        for p1, p2 in zip(m1.parameters(), m2.parameters()):
+            g = torch.randn_like(p1)
+            # FusedSGD made every .grad a view into the flat buffer
+            assert p1.grad is not None
+            p1.grad.copy_(g)
+            p2.grad = g.clone()
+        o1.step()
+        o2.step()
+        o2.zero_grad(set_to_none=False)
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        assert torch.allclose(p1, p2, rtol=1e-5, atol=1e-6), (
+            (p1 - p2).abs().max()
+        )
