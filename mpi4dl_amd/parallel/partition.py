@@ -8,9 +8,9 @@ mp_pipeline.py:41-168). Differences by design:
   reference's trick of a dummy CUDA forward at a small "seq" size whose
   shapes are then rescaled (mp_pipeline.py:126-168, train_spatial.py:61-238),
   which silently breaks on resolution-inhomogeneous layers.
-* ``ready_model`` materialises ONLY the local partition's weights on the
-  target device; remote partitions stay meta (the reference builds the
-  full model on every rank).
+* ``ready_model`` moves ONLY the local partition's weights to the
+  target device (and drops the host-side copy of remote partitions);
+  the reference materialises the full model on every rank's GPU.
 
 Spatial/halo modules in this package are meta-aware: on meta tensors
 they skip communication and produce correctly-shaped outputs, so the
@@ -118,6 +118,9 @@ class model_generator:
         local = local.to(device=device, dtype=dtype)
         self.models = local
         self.ready_rank = split_rank
+        # free the host copy of remote stages (shape inference is done;
+        # at F416-scale the full model is ~0.5 GB/process of host RAM)
+        self.model = None
         return local
 
 
