@@ -31,8 +31,10 @@ def conv_bn_act(
     act: bool = True,
     bn: bool = True,
 ):
+    # bias=True even under BN: the reference's resnet_layer uses the
+    # nn.Conv2d default everywhere (resnet.py:40-46)
     pad = kernel // 2
-    layers = [nn.Conv2d(in_ch, out_ch, kernel, stride=stride, padding=pad, bias=not bn)]
+    layers = [nn.Conv2d(in_ch, out_ch, kernel, stride=stride, padding=pad)]
     if bn:
         layers.append(nn.BatchNorm2d(out_ch))
     if act:
@@ -60,31 +62,36 @@ class BasicBlockV1(nn.Module):
 
 
 class BottleneckV2(nn.Module):
-    """v2 cell: pre-activation bottleneck (BN-ReLU-conv ×3) + skip."""
+    """v2 cell (reference make_cell_v2, resnet.py:181-231): pre-activation
+    3x3(stride) -> 3x3 -> 1x1 with an un-normalised 1x1 projection on the
+    first block of a stage; no activation after the add. ``preact=False``
+    reproduces the reference's first-block-of-first-stage case (r1 has
+    neither BN nor ReLU before its conv)."""
 
-    expansion = 4
-
-    def __init__(self, in_ch: int, mid_ch: int, stride: int = 1):
+    def __init__(self, in_ch: int, mid_ch: int, out_ch: int,
+                 stride: int = 1, preact: bool = True):
         super().__init__()
-        out_ch = mid_ch * self.expansion
-        self.pre = nn.Sequential(nn.BatchNorm2d(in_ch), nn.ReLU(inplace=True))
-        self.body = nn.Sequential(
-            nn.Conv2d(in_ch, mid_ch, 1, bias=False),
-            nn.BatchNorm2d(mid_ch),
-            nn.ReLU(inplace=True),
-            nn.Conv2d(mid_ch, mid_ch, 3, stride=stride, padding=1, bias=False),
-            nn.BatchNorm2d(mid_ch),
-            nn.ReLU(inplace=True),
-            nn.Conv2d(mid_ch, out_ch, 1, bias=False),
+        self.pre1 = (
+            nn.Sequential(nn.BatchNorm2d(in_ch), nn.ReLU(inplace=True))
+            if preact
+            else nn.Identity()
         )
+        self.conv1 = nn.Conv2d(in_ch, mid_ch, 3, stride=stride, padding=1)
+        self.pre2 = nn.Sequential(nn.BatchNorm2d(mid_ch), nn.ReLU(inplace=True))
+        self.conv2 = nn.Conv2d(mid_ch, mid_ch, 3, padding=1)
+        self.pre3 = nn.Sequential(nn.BatchNorm2d(mid_ch), nn.ReLU(inplace=True))
+        self.conv3 = nn.Conv2d(mid_ch, out_ch, 1)
         self.proj = None
         if stride != 1 or in_ch != out_ch:
-            self.proj = nn.Conv2d(in_ch, out_ch, 1, stride=stride, bias=False)
+            # reference r4: plain conv on the RAW input (resnet.py:212-219)
+            self.proj = nn.Conv2d(in_ch, out_ch, 1, stride=stride)
 
     def forward(self, x):
-        h = self.pre(x)
-        s = x if self.proj is None else self.proj(h)
-        return self.body(h) + s
+        y = self.conv1(self.pre1(x))
+        y = self.conv2(self.pre2(y))
+        y = self.conv3(self.pre3(y))
+        s = x if self.proj is None else self.proj(x)
+        return s + y
 
 
 class Head(nn.Module):
@@ -156,12 +163,20 @@ def get_resnet_v2(
     _, in_ch, H, W = input_shape
     cells = [_stem(in_ch, num_filters, min(H, W))]
     ch = num_filters
+    mid = num_filters
     for group in range(3):
-        mid = num_filters * (2**group)
+        # reference width schedule (resnet.py:288-300): stage 0 expands
+        # 4x (mids 16->outs 64), later stages 2x (64->128, 128->256);
+        # stage 0 never downsamples
+        out = mid * (4 if group == 0 else 2)
         for block in range(n):
             stride = 2 if (group > 0 and block == 0) else 1
-            cells.append(BottleneckV2(ch, mid, stride))
-            ch = mid * BottleneckV2.expansion
+            cells.append(BottleneckV2(
+                ch, mid, out, stride,
+                preact=not (group == 0 and block == 0),
+            ))
+            ch = out
+        mid = out
     cells.append(Head(ch, num_classes, final_bn=True))
     model = nn.Sequential(*cells)
     if device is not None:
@@ -208,8 +223,8 @@ def get_resnet101_cells(
         mid = width * (2**group)
         for b in range(blocks):
             stride = 2 if (group > 0 and b == 0) else 1
-            cells.append(BottleneckV2(ch, mid, stride))
-            ch = mid * BottleneckV2.expansion
+            cells.append(BottleneckV2(ch, mid, mid * 4, stride))
+            ch = mid * 4
     cells.append(Head(ch, num_classes, final_bn=True))
     model = nn.Sequential(*cells)
     if device is not None:

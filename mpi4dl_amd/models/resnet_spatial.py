@@ -17,7 +17,7 @@ import torch.nn as nn
 
 from ..ops.plan import SpatialPlan
 from ..ops.spatial_conv import HaloConv2d, HaloPool2d
-from .resnet import BottleneckV2, Head  # noqa: F401
+from .resnet import Head  # noqa: F401
 
 
 def _bn_relu(mknorm, ch):
@@ -42,10 +42,10 @@ class BasicBlockV1S(nn.Module):
     def __init__(self, in_ch, out_ch, stride, ctx, mknorm=nn.BatchNorm2d):
         super().__init__()
         self.body = nn.Sequential(
-            sconv(in_ch, out_ch, 3, stride, ctx),
+            sconv(in_ch, out_ch, 3, stride, ctx, bias=True),
             _bn_relu(mknorm, out_ch),
             nn.Identity(),
-            sconv(out_ch, out_ch, 3, 1, ctx),
+            sconv(out_ch, out_ch, 3, 1, ctx, bias=True),
             mknorm(out_ch),
         )
         self.proj = None
@@ -60,31 +60,33 @@ class BasicBlockV1S(nn.Module):
 
 
 class BottleneckV2S(nn.Module):
-    """Spatial variant of BottleneckV2 (reference make_cell_v2_spatial :375)."""
+    """Spatial variant of BottleneckV2 (reference make_cell_v2_spatial
+    :375): pre-act 3x3(s) -> 3x3 -> 1x1; both 3x3s are halo convs, the
+    1x1s and the raw-input projection need no halo."""
 
-    expansion = 4
-
-    def __init__(self, in_ch, mid_ch, stride, ctx, mknorm=nn.BatchNorm2d):
+    def __init__(self, in_ch, mid_ch, out_ch, stride, ctx,
+                 mknorm=nn.BatchNorm2d, preact=True):
         super().__init__()
-        out_ch = mid_ch * self.expansion
-        self.pre = nn.Sequential(_bn_relu(mknorm, in_ch), nn.Identity())
-        self.body = nn.Sequential(
-            nn.Conv2d(in_ch, mid_ch, 1, bias=False),
-            _bn_relu(mknorm, mid_ch),
-            nn.Identity(),
-            sconv(mid_ch, mid_ch, 3, stride, ctx),
-            _bn_relu(mknorm, mid_ch),
-            nn.Identity(),
-            nn.Conv2d(mid_ch, out_ch, 1, bias=False),
+        self.pre1 = (
+            nn.Sequential(_bn_relu(mknorm, in_ch), nn.Identity())
+            if preact
+            else nn.Identity()
         )
+        self.conv1 = sconv(in_ch, mid_ch, 3, stride, ctx, bias=True)
+        self.pre2 = nn.Sequential(_bn_relu(mknorm, mid_ch), nn.Identity())
+        self.conv2 = sconv(mid_ch, mid_ch, 3, 1, ctx, bias=True)
+        self.pre3 = nn.Sequential(_bn_relu(mknorm, mid_ch), nn.Identity())
+        self.conv3 = nn.Conv2d(mid_ch, out_ch, 1)
         self.proj = None
         if stride != 1 or in_ch != out_ch:
-            self.proj = nn.Conv2d(in_ch, out_ch, 1, stride=stride, bias=False)
+            self.proj = nn.Conv2d(in_ch, out_ch, 1, stride=stride)
 
     def forward(self, x):
-        h = self.pre(x)
-        s = x if self.proj is None else self.proj(h)
-        return self.body(h) + s
+        y = self.conv1(self.pre1(x))
+        y = self.conv2(self.pre2(y))
+        y = self.conv3(self.pre3(y))
+        s = x if self.proj is None else self.proj(x)
+        return s + y
 
 
 class StemS(nn.Module):
@@ -101,7 +103,7 @@ class StemS(nn.Module):
             )
         else:
             self.ops = nn.Sequential(
-                sconv(in_ch, filters, 3, 1, ctx),
+                sconv(in_ch, filters, 3, 1, ctx, bias=True),
                 mknorm(filters),
                 nn.ReLU(inplace=True),
             )
@@ -133,11 +135,16 @@ def _build(input_shape, num_classes, n, num_filters, plan, version):
                 cells.append(BasicBlockV1S(ch, out_ch, stride, ctx(), mknorm()))
                 ch = out_ch
         else:
-            mid = num_filters * (2**group)
+            # reference width schedule: stage 0 4x / stride 1, later 2x
+            mid = num_filters if group == 0 else ch
+            out = mid * (4 if group == 0 else 2)
             for block in range(n):
                 stride = 2 if (group > 0 and block == 0) else 1
-                cells.append(BottleneckV2S(ch, mid, stride, ctx(), mknorm()))
-                ch = mid * BottleneckV2.expansion
+                cells.append(BottleneckV2S(
+                    ch, mid, out, stride, ctx(), mknorm(),
+                    preact=not (group == 0 and block == 0),
+                ))
+                ch = out
     cells.append(Head(ch, num_classes, final_bn=(version == 2), mknorm=mknorm()))
     return nn.Sequential(*cells)
 
@@ -177,7 +184,7 @@ def get_resnet101_cells(
         mid = width * (2**group)
         for b in range(blocks):
             stride = 2 if (group > 0 and b == 0) else 1
-            cells.append(BottleneckV2S(ch, mid, stride, ctx(), mknorm()))
-            ch = mid * BottleneckV2.expansion
+            cells.append(BottleneckV2S(ch, mid, mid * 4, stride, ctx(), mknorm()))
+            ch = mid * 4
     cells.append(Head(ch, num_classes, final_bn=True, mknorm=mknorm()))
     return nn.Sequential(*cells)

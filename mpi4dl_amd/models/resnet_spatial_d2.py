@@ -54,17 +54,18 @@ class CropInterior(nn.Module):
 
 
 class BottleneckV2D2(nn.Module):
-    """Pre-activation bottleneck consuming 1 surplus pixel (stride 1).
+    """Pre-activation 3x3 -> 3x3 -> 1x1 block (reference make_cell_v2
+    structure) consuming TWO surplus pixels per block (one per 3x3;
+    stride-1 / no-projection blocks only — others run D1-style).
 
-    ``pre_h`` > 0 prepends the fused halo exchange that grants the next
-    ``pre_h`` blocks their surplus.
+    ``pre_h`` > 0 prepends the fused halo exchange granting ``pre_h``
+    surplus pixels to the blocks that follow.
     """
 
-    expansion = 4
-
-    def __init__(self, in_ch, mid_ch, ctx, mknorm=nn.BatchNorm2d, pre_h: int = 0):
+    def __init__(self, in_ch, mid_ch, out_ch, ctx, mknorm=nn.BatchNorm2d,
+                 pre_h: int = 0):
         super().__init__()
-        out_ch = mid_ch * self.expansion
+        assert in_ch == out_ch, "D2 blocks are the no-projection blocks"
         from ..ops.halo import HaloExchanger, TileLayout
 
         self.layout = TileLayout(ctx["num_spatial_parts"], ctx["slice_method"])
@@ -76,29 +77,28 @@ class BottleneckV2D2(nn.Module):
         )
         self.grad_mode = ctx.get("grad_mode", "exact")
         self.pre_h = pre_h
-        self.pre = nn.Sequential(mknorm(in_ch), nn.ReLU(inplace=True))
-        self.conv1 = nn.Conv2d(in_ch, mid_ch, 1, bias=False)
-        self.bn1 = mknorm(mid_ch)
-        self.act = nn.ReLU(inplace=True)
-        # the 3x3: interior sides unpadded (consume surplus), boundary
-        # sides freshly zero-padded
-        self.conv2 = nn.Conv2d(mid_ch, mid_ch, 3, stride=1, padding=0, bias=False)
-        self.bn2 = mknorm(mid_ch)
-        self.conv3 = nn.Conv2d(mid_ch, out_ch, 1, bias=False)
-        self.proj = None
-        if in_ch != out_ch:
-            self.proj = nn.Conv2d(in_ch, out_ch, 1, bias=False)
-        self.crop = CropInterior(self.exchanger, 1)
+        self.pre1 = nn.Sequential(mknorm(in_ch), nn.ReLU(inplace=True))
+        # 3x3s: interior sides unpadded (consume surplus), image-boundary
+        # sides freshly zero-padded (outer_pad_only)
+        self.conv1 = nn.Conv2d(in_ch, mid_ch, 3, padding=0)
+        self.pre2 = nn.Sequential(mknorm(mid_ch), nn.ReLU(inplace=True))
+        self.conv2 = nn.Conv2d(mid_ch, mid_ch, 3, padding=0)
+        self.pre3 = nn.Sequential(mknorm(mid_ch), nn.ReLU(inplace=True))
+        self.conv3 = nn.Conv2d(mid_ch, out_ch, 1)
+        self.crop = CropInterior(self.exchanger, 2)
+
+    def _outer_pad(self, y):
+        if self.exchanger is not None:
+            return outer_pad_only(y, self.layout, self.tile, 1)
+        return nn.functional.pad(y, (1, 1, 1, 1))
 
     def forward(self, x):
         if self.pre_h > 0:
             x = halo_pad_d2(x, self.pre_h, self.exchanger, self.grad_mode)
-        h = self.pre(x)
-        s = x if self.proj is None else self.proj(h)
-        y = self.act(self.bn1(self.conv1(h)))
-        y = outer_pad_only(y, self.layout, self.tile, 1) if self.exchanger is not None else nn.functional.pad(y, (1, 1, 1, 1))
-        y = self.conv3(self.act(self.bn2(self.conv2(y))))
-        return y + self.crop(s)
+        y = self.conv1(self._outer_pad(self.pre1(x)))
+        y = self.conv2(self._outer_pad(self.pre2(y)))
+        y = self.conv3(self.pre3(y))
+        return y + self.crop(x)
 
 
 def get_resnet_v2(
@@ -126,30 +126,38 @@ def get_resnet_v2(
 
     cells.append(StemS(in_ch, num_filters, min(H, W), ctx(), mknorm()))
     ch = num_filters
+    mid = num_filters
     for group in range(3):
-        mid = num_filters * (2**group)
+        # reference width schedule: stage 0 expands 4x (stride 1), later
+        # stages 2x with stride 2 at block 0 (resnet.py:288-300)
+        out_ch = mid * (4 if group == 0 else 2)
         surplus_left = 0
         for block in range(n):
             stride = 2 if (group > 0 and block == 0) else 1
             c = ctx()
-            out_ch = mid * BottleneckV2D2.expansion
             d1_style = (
                 c is None
                 or stride != 1
                 or block == 0  # channel-change block: keep D1 (has proj)
             )
             if d1_style:
-                cells.append(BottleneckV2S(ch, mid, stride, c, mknorm()))
+                cells.append(BottleneckV2S(
+                    ch, mid, out_ch, stride, c, mknorm(),
+                    preact=not (group == 0 and block == 0),
+                ))
                 surplus_left = 0
             else:
                 pre_h = 0
                 if surplus_left == 0:
-                    pre_h = min(fused_layers, n - block)
-                    surplus_left = pre_h
+                    # 2 surplus pixels per fused block (two 3x3 convs)
+                    nblk = min(fused_layers, n - block)
+                    pre_h = 2 * nblk
+                    surplus_left = nblk
                 cells.append(
-                    BottleneckV2D2(ch, mid, c, mknorm(), pre_h=pre_h)
+                    BottleneckV2D2(ch, mid, out_ch, c, mknorm(), pre_h=pre_h)
                 )
                 surplus_left -= 1
             ch = out_ch
+        mid = out_ch
     cells.append(Head(ch, num_classes, final_bn=True, mknorm=mknorm()))
     return nn.Sequential(*cells)
