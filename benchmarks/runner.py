@@ -66,7 +66,7 @@ def build_model(args, plan, mb):
         return M.get_resnet_v2(
             shape, args.num_classes, n=max(args.num_layers // 9, 1),
             num_filters=args.num_filters if args.num_filters <= 64 else 16,
-            plan=plan,
+            plan=plan, ref_stem=args.ref_stem,
         )
     from mpi4dl_amd.models.amoebanet import amoebanetd
 

@@ -112,11 +112,14 @@ class Head(nn.Module):
         return self.fc(torch.flatten(x, 1))
 
 
-def _stem(in_ch: int, filters: int, image_size: int) -> nn.Module:
+def _stem(in_ch: int, filters: int, image_size: int,
+          ref_stem: bool = False) -> nn.Module:
     """Stem sized to the image: big images get a stride-2 7x7 + maxpool so
     activations stay tractable (the reference keeps a 3x3 stride-1 stem for
-    CIFAR-scale and relies on SP for big images; we keep stride-1 below 128)."""
-    if image_size >= 128:
+    CIFAR-scale and relies on SP for big images; we keep stride-1 below 128).
+    ``ref_stem=True`` forces the reference's stride-1 3x3 stem at ANY size —
+    use it when comparing against the reference's published ResNet numbers."""
+    if image_size >= 128 and not ref_stem:
         return nn.Sequential(
             nn.Conv2d(in_ch, filters, 7, stride=2, padding=3, bias=False),
             nn.BatchNorm2d(filters),
@@ -132,11 +135,12 @@ def get_resnet_v1(
     n: int = 3,
     num_filters: int = 16,
     device: Optional[str] = None,
+    ref_stem: bool = False,
 ) -> nn.Sequential:
     """6n+2-layer v1 ResNet as a flat Sequential of cells
     (reference resnet.py:145-178)."""
     _, in_ch, H, W = input_shape
-    cells = [_stem(in_ch, num_filters, min(H, W))]
+    cells = [_stem(in_ch, num_filters, min(H, W), ref_stem)]
     ch = num_filters
     for group in range(3):
         out_ch = num_filters * (2**group)
@@ -157,11 +161,12 @@ def get_resnet_v2(
     n: int = 12,
     num_filters: int = 16,
     device: Optional[str] = None,
+    ref_stem: bool = False,
 ) -> nn.Sequential:
     """9n+2-layer v2 (pre-activation bottleneck) ResNet
     (reference resnet.py:270-326)."""
     _, in_ch, H, W = input_shape
-    cells = [_stem(in_ch, num_filters, min(H, W))]
+    cells = [_stem(in_ch, num_filters, min(H, W), ref_stem)]
     ch = num_filters
     mid = num_filters
     for group in range(3):
@@ -189,11 +194,12 @@ def get_resnet18_cells(
     num_classes: int = 1000,
     width: int = 64,
     device: Optional[str] = None,
+    ref_stem: bool = False,
 ) -> nn.Sequential:
     """ImageNet-style ResNet-18 as flat cells (BASELINE config 1:
     ResNet-18 layer parallelism at 224²). Basic-block counts 2-2-2-2."""
     _, in_ch, H, W = input_shape
-    cells = [_stem(in_ch, width, min(H, W))]
+    cells = [_stem(in_ch, width, min(H, W), ref_stem)]
     ch = width
     for group, blocks in enumerate([2, 2, 2, 2]):
         out = width * (2**group)
@@ -213,11 +219,12 @@ def get_resnet101_cells(
     num_classes: int = 1000,
     width: int = 64,
     device: Optional[str] = None,
+    ref_stem: bool = False,
 ) -> nn.Sequential:
     """ImageNet-style ResNet-101 as flat cells (for BASELINE config 4:
     ResNet-101 SP+PP at 2048²). Bottleneck counts 3-4-23-3."""
     _, in_ch, H, W = input_shape
-    cells = [_stem(in_ch, width, min(H, W))]
+    cells = [_stem(in_ch, width, min(H, W), ref_stem)]
     ch = width
     for group, blocks in enumerate([3, 4, 23, 3]):
         mid = width * (2**group)

@@ -90,11 +90,13 @@ class BottleneckV2S(nn.Module):
 
 
 class StemS(nn.Module):
-    """Spatial stem: 7x7/2 halo conv + BN/ReLU + 3x3/2 halo max pool."""
+    """Spatial stem: 7x7/2 halo conv + BN/ReLU + 3x3/2 halo max pool
+    (or the reference's stride-1 3x3 stem with ref_stem=True)."""
 
-    def __init__(self, in_ch, filters, image_size, ctx, mknorm=nn.BatchNorm2d):
+    def __init__(self, in_ch, filters, image_size, ctx,
+                 mknorm=nn.BatchNorm2d, ref_stem=False):
         super().__init__()
-        if image_size >= 128:
+        if image_size >= 128 and not ref_stem:
             self.ops = nn.Sequential(
                 sconv(in_ch, filters, 7, 2, ctx),
                 _bn_relu(mknorm, filters),
@@ -112,7 +114,8 @@ class StemS(nn.Module):
         return self.ops(x)
 
 
-def _build(input_shape, num_classes, n, num_filters, plan, version):
+def _build(input_shape, num_classes, n, num_filters, plan, version,
+           ref_stem=False):
     _, in_ch, H, W = input_shape
     cells = []
 
@@ -125,7 +128,8 @@ def _build(input_shape, num_classes, n, num_filters, plan, version):
         i = len(cells)
         return lambda ch: plan.norm(ch, i)
 
-    cells.append(StemS(in_ch, num_filters, min(H, W), ctx(), mknorm()))
+    cells.append(StemS(in_ch, num_filters, min(H, W), ctx(), mknorm(),
+                       ref_stem=ref_stem))
     ch = num_filters
     for group in range(3):
         if version == 1:
@@ -150,15 +154,19 @@ def _build(input_shape, num_classes, n, num_filters, plan, version):
 
 
 def get_resnet_v1(
-    input_shape, num_classes=10, n=3, num_filters=16, plan: Optional[SpatialPlan] = None
+    input_shape, num_classes=10, n=3, num_filters=16,
+    plan: Optional[SpatialPlan] = None, ref_stem=False,
 ):
-    return _build(input_shape, num_classes, n, num_filters, plan, version=1)
+    return _build(input_shape, num_classes, n, num_filters, plan, version=1,
+                  ref_stem=ref_stem)
 
 
 def get_resnet_v2(
-    input_shape, num_classes=10, n=12, num_filters=16, plan: Optional[SpatialPlan] = None
+    input_shape, num_classes=10, n=12, num_filters=16,
+    plan: Optional[SpatialPlan] = None, ref_stem=False,
 ):
-    return _build(input_shape, num_classes, n, num_filters, plan, version=2)
+    return _build(input_shape, num_classes, n, num_filters, plan, version=2,
+                  ref_stem=ref_stem)
 
 
 def get_resnet101_cells(

@@ -67,6 +67,10 @@ def get_parser() -> argparse.ArgumentParser:
     p.add_argument("--grad-mode", default="exact", choices=["exact", "drop"],
                    help="halo backward: exact transposed exchange or the "
                         "reference's drop semantics")
+    p.add_argument("--ref-stem", action="store_true",
+                   help="use the reference's stride-1 3x3 stem at any image "
+                        "size (for apples-to-apples A/B with its published "
+                        "ResNet numbers; default stem downsamples >=128^2)")
     p.add_argument("--act-ckpt", action="store_true",
                    help="activation checkpointing: recompute cell forwards "
                         "in backward (cuts GPipe peak activation memory)")
