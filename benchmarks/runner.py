@@ -60,6 +60,7 @@ def build_model(args, plan, mb):
                 shape, args.num_classes, n=max(args.num_layers // 9, 1),
                 num_filters=args.num_filters if args.num_filters <= 64 else 16,
                 plan=plan, fused_layers=args.fused_layers,
+                ref_stem=args.ref_stem,
             )
         from mpi4dl_amd.models import resnet_spatial as M
 

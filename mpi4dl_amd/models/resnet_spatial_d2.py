@@ -108,6 +108,7 @@ def get_resnet_v2(
     num_filters: int = 16,
     plan: Optional[SpatialPlan] = None,
     fused_layers: int = 4,
+    ref_stem: bool = False,
 ):
     """D2 ResNet v2: cell layout identical to resnet/resnet_spatial v2
     builders (stem + 3n bottlenecks + head); spatial cells inside
@@ -124,7 +125,8 @@ def get_resnet_v2(
         i = len(cells)
         return lambda ch: plan.norm(ch, i)
 
-    cells.append(StemS(in_ch, num_filters, min(H, W), ctx(), mknorm()))
+    cells.append(StemS(in_ch, num_filters, min(H, W), ctx(), mknorm(),
+                       ref_stem=ref_stem))
     ch = num_filters
     mid = num_filters
     for group in range(3):
