@@ -82,6 +82,7 @@ def body(rank, world, steps, batch, parts, lr, cfg):
     eng = train_model_spatial(
         gen, comm.local_rank, batch, parts, comm, slice_method=cfg["slice"],
         optimizer=opt, grad_mode=cfg["grad_mode"], schedule=cfg["schedule"],
+        act_ckpt=cfg.get("act_ckpt", False),
         device=torch.device("cpu"),
     )
     red = GradReducer(comm)
@@ -107,6 +108,7 @@ def sample_cfg(rng):
     return dict(slice=slice_, nsp=nsp, spatial_size=spatial_size,
                 split=split, ldp=ldp,
                 schedule=rng.choice(["gpipe", "1f1b"]),
+                act_ckpt=rng.random() < 0.3,
                 grad_mode=rng.choice(["exact", "exact", "drop"]))
 
 
@@ -133,7 +135,8 @@ def main():
         batch = parts * cfg["ldp"] * 2
         tag = (f"[{i}] {cfg['slice']} nsp={cfg['nsp']} ss={cfg['spatial_size']} "
                f"split={cfg['split']} ldp={cfg['ldp']} parts={parts} "
-               f"{cfg['schedule']}/{cfg['grad_mode']} world={world}")
+               f"{cfg['schedule']}/{cfg['grad_mode']}"
+               f"{'/ckpt' if cfg.get('act_ckpt') else ''} world={world}")
         try:
             expected = serial_losses(steps, batch, parts, lr)
             got = run_distributed(body, world, (steps, batch, parts, lr, cfg),
