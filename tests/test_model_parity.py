@@ -54,3 +54,53 @@ def test_resnet_param_parity(ver, n):
     rp, op = _params(ref), _params(ours)
     # small residual diff: our GAP head vs the reference's flatten head
     assert abs(rp - op) / rp < 0.02, (rp, op)
+
+
+def test_resnet_v2_forward_equivalence():
+    """Copy the reference v2 model's weights into ours (structure-aware:
+    the reference's resnet_layer holds one USED and one dead BatchNorm)
+    and compare trunk outputs — must be bitwise identical."""
+    ref_resnet, _ = _ref_models()
+    import torch.nn as nn
+
+    import mpi4dl_amd.models.resnet as ours_r
+
+    torch.manual_seed(0)
+    ref = ref_resnet.get_resnet_v2((2, 3, 64, 64), 29, num_classes=10)
+    torch.manual_seed(1)  # different init on purpose: the copy must win
+    ours = ours_r.get_resnet_v2((2, 3, 64, 64), 10, n=3)
+
+    def copy_layer(rl, conv_dst, bn_dst):
+        conv_dst.weight.data.copy_(rl.conv1.weight)
+        conv_dst.bias.data.copy_(rl.conv1.bias)
+        if bn_dst is not None:
+            bn_dst.load_state_dict(rl.batch_first.state_dict())
+
+    cells_r, cells_o = list(ref), list(ours)
+    assert len(cells_r) == len(cells_o)
+    st_r, st_o = cells_r[0], cells_o[0]
+    st_o[0].weight.data.copy_(st_r.conv1.weight)
+    st_o[0].bias.data.copy_(st_r.conv1.bias)
+    st_o[1].load_state_dict(st_r.batch_last.state_dict())
+    for cr, co in zip(cells_r[1:-1], cells_o[1:-1]):
+        pre1 = None if isinstance(co.pre1, nn.Identity) else co.pre1[0]
+        copy_layer(cr.r1, co.conv1, pre1)
+        copy_layer(cr.r2, co.conv2, co.pre2[0])
+        copy_layer(cr.r3, co.conv3, co.pre3[0])
+        if hasattr(cr, "r4"):
+            assert co.proj is not None
+            co.proj.weight.data.copy_(cr.r4.conv1.weight)
+            co.proj.bias.data.copy_(cr.r4.conv1.bias)
+        else:
+            assert co.proj is None
+    ref.eval()
+    ours.eval()
+    torch.manual_seed(5)
+    x = torch.randn(2, 3, 64, 64)
+    with torch.no_grad():
+        yr, yo = x, x
+        for c in cells_r[:-1]:
+            yr = c(yr)
+        for c in cells_o[:-1]:
+            yo = c(yo)
+    assert torch.equal(yr, yo), (yr - yo).abs().max()
