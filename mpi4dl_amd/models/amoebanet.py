@@ -13,9 +13,12 @@ amoebanet.py:318-336); everything else here is our own implementation:
 * 1x7/7x1 convs use per-axis halos ((0,3) / (3,0)) — the reference
   routes them through full square halos;
 * avg_pool_3x3 keeps exact count_include_pad=False semantics in tile
-  mode (the reference's spatial path silently changes them, and its
-  spatial max_pool_3x3 even instantiates an AvgPool — bug not copied,
-  SURVEY.md quirks).
+  mode (the reference's spatial path silently changes them);
+* the reference's max_pool_3x3 builder instantiates an AvgPool in BOTH
+  its serial and spatial paths (amoebanet.py:108-125) — its published
+  numbers ran avg pools there. Default here is the faithful genotype
+  (real max pool); ``ref_quirks=True`` reproduces the reference's
+  behaviour bitwise (tests/test_model_parity.py proves equality).
 
 Cells pass ``(x, skip)`` tuples between stages — the pipeline and every
 spatial seam handle multi-tensor activations.
@@ -120,7 +123,8 @@ def _bn_relu(mknorm, ch):
     return nn.Sequential(bn, nn.ReLU(inplace=False))
 
 
-def make_op(name: str, c: int, stride: int, ctx, mknorm) -> nn.Module:
+def make_op(name: str, c: int, stride: int, ctx, mknorm,
+            ref_quirks: bool = False) -> nn.Module:
     if name == "none":
         return (
             nn.Identity() if stride == 1 else FactorizedReduce(c, c, mknorm)
@@ -133,6 +137,15 @@ def make_op(name: str, c: int, stride: int, ctx, mknorm) -> nn.Module:
             **(ctx or {})
         )
     if name == "max_pool_3x3":
+        if ref_quirks:
+            # the reference's max_pool_3x3 builder instantiates an AVG
+            # pool in both its serial and spatial paths
+            # (amoebanet.py:108-125) — its published numbers ran this.
+            # Default (faithful genotype) uses a real max pool.
+            return HaloPool2d(
+                "avg", 3, stride=stride, padding=1,
+                count_include_pad=False, **(ctx or {})
+            )
         return HaloPool2d("max", 3, stride=stride, padding=1, **(ctx or {}))
     if name == "max_pool_2x2":
         # padding 0, stride 2 in reduction cells: tile-local, no halo
@@ -207,6 +220,7 @@ class Cell(nn.Module):
         reduction_prev: bool,
         ctx=None,
         mknorm=nn.BatchNorm2d,
+        ref_quirks: bool = False,
     ):
         super().__init__()
         self.reduce1 = relu_conv_bn(channels_prev, channels, mknorm)
@@ -223,7 +237,10 @@ class Cell(nn.Module):
         self.operations = nn.ModuleList()
         for i, name in ops:
             stride = 2 if (reduction and i < 2) else 1
-            self.operations.append(make_op(name, channels, stride, ctx, mknorm))
+            self.operations.append(
+                make_op(name, channels, stride, ctx, mknorm,
+                        ref_quirks=ref_quirks)
+            )
 
     def forward(self, input_or_states):
         if isinstance(input_or_states, tuple):
@@ -271,6 +288,7 @@ def amoebanetd(
     num_layers: int = 6,
     num_filters: int = 64,
     plan: Optional[SpatialPlan] = None,
+    ref_quirks: bool = False,
 ) -> nn.Sequential:
     """AmoebaNet-D as a flat Sequential of cells (reference :535-616).
     ``plan`` != None builds the spatial variant (reference
@@ -311,6 +329,7 @@ def amoebanetd(
             state["red_prev"],
             ctx=ctx(),
             mknorm=mknorm(),
+            ref_quirks=ref_quirks,
         )
         state["c_pp"] = state["c_p"]
         state["c_p"] = state["c"] * len(cell.concat)

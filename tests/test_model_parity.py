@@ -104,3 +104,33 @@ def test_resnet_v2_forward_equivalence():
         for c in cells_o[:-1]:
             yo = c(yo)
     assert torch.equal(yr, yo), (yr - yo).abs().max()
+
+
+def test_amoebanet_forward_equivalence_ref_quirks():
+    """With ref_quirks=True (reproducing the reference's
+    max_pool_3x3-is-actually-AvgPool builder, amoebanet.py:108-125) and
+    its weights copied in order, our AmoebaNet-D is bitwise identical
+    to the reference model, cell by cell and through the head."""
+    _, ref_amoeba = _ref_models()
+    from mpi4dl_amd.models.amoebanet import amoebanetd
+
+    torch.manual_seed(0)
+    ref = ref_amoeba.amoebanetd(num_classes=100, num_layers=6, num_filters=64)
+    torch.manual_seed(1)
+    ours = amoebanetd(100, 6, 64, ref_quirks=True)
+    with torch.no_grad():
+        for a, b in zip(ref.parameters(), ours.parameters()):
+            assert a.shape == b.shape
+            b.copy_(a)
+    ref.eval()
+    ours.eval()
+    torch.manual_seed(5)
+    x = torch.randn(2, 3, 64, 64)
+    yr = yo = x
+    with torch.no_grad():
+        for cr, co in zip(list(ref), list(ours)):
+            yr, yo = cr(yr), co(yo)
+            ta = yr if isinstance(yr, tuple) else (yr,)
+            tb = yo if isinstance(yo, tuple) else (yo,)
+            for a, b in zip(ta, tb):
+                assert torch.equal(a, b), (a - b).abs().max()
