@@ -102,6 +102,10 @@ def main():
     ap.add_argument("--gems", action="store_true",
                     help="GEMS bidirectional pipelines on top of SP "
                          "(two mirrored engines per GPU, 2x batch/step)")
+    ap.add_argument("--ref-quirks", action="store_true",
+                    help="reproduce the reference model bitwise (its "
+                         "max_pool_3x3 builder creates an AvgPool; see "
+                         "models/amoebanet.py)")
     ap.add_argument("--act-ckpt", action="store_true",
                     help="recompute cell forwards in backward (fits larger "
                          "global batches in HBM at ~1 extra forward cost)")
@@ -153,7 +157,8 @@ def main():
     # model (full, cells) — probe cell count cheaply on meta
     def build(plan=None):
         torch.manual_seed(0)
-        return amoebanetd(args.num_classes, args.num_layers, args.num_filters, plan)
+        return amoebanetd(args.num_classes, args.num_layers, args.num_filters,
+                          plan, ref_quirks=args.ref_quirks)
 
     with torch.device("meta"):
         ncells = len(build())
