@@ -72,7 +72,8 @@ def build_model(args, plan, mb):
     from mpi4dl_amd.models.amoebanet import amoebanetd
 
     layers = args.num_layers - (args.num_layers % 3) or 3
-    return amoebanetd(args.num_classes, layers, args.num_filters, plan=plan)
+    return amoebanetd(args.num_classes, layers, args.num_filters, plan=plan,
+                      ref_quirks=getattr(args, "ref_quirks", False))
 
 
 def make_engines(args, mode):

@@ -67,6 +67,9 @@ def get_parser() -> argparse.ArgumentParser:
     p.add_argument("--grad-mode", default="exact", choices=["exact", "drop"],
                    help="halo backward: exact transposed exchange or the "
                         "reference's drop semantics")
+    p.add_argument("--ref-quirks", action="store_true",
+                   help="reproduce the reference's AmoebaNet bitwise (its "
+                        "max_pool_3x3 builder creates an AvgPool)")
     p.add_argument("--ref-stem", action="store_true",
                    help="use the reference's stride-1 3x3 stem at any image "
                         "size (for apples-to-apples A/B with its published "
