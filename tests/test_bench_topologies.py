@@ -73,3 +73,9 @@ def test_bench_n4_gems():
     out = run_distributed(_bench_body, 4, (("--gems",),), timeout=300)
     _check(out, 4, "sp2+gems+pp3")
     assert out[0]["config"]["global_batch"] == 16  # two replicas per step
+
+
+def test_bench_n4_1f1b():
+    out = run_distributed(_bench_body, 4, (("--schedule", "1f1b"),),
+                          timeout=300)
+    _check(out, 4, "sp2+pp3")
