@@ -119,11 +119,45 @@ def world_of(cfg):
                            cfg["ldp"])
 
 
+def gems_sweep(args, rng):
+    """GEMS LP sweep: split counts (odd/even mp), parts, MASTER-OPT."""
+    sys.path.insert(0, os.path.join(
+        os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "tests"))
+    import test_gems as TG
+
+    steps, lr = 2, 0.01
+    failures = 0
+    for i in range(args.combos):
+        split = rng.choice([2, 3, 4])
+        parts = rng.choice([1, 2])
+        comm_opt = rng.random() < 0.5
+        B = 2 * parts
+        tag = f"[gems {i}] split={split} parts={parts} B={B} comm_opt={comm_opt}"
+        try:
+            expected = TG._serial_losses(steps, B, parts, lr)
+            got = run_distributed(TG._gems_body, split,
+                                  (steps, B, parts, lr, comm_opt), timeout=300)
+            combined = [sum(g[s] for g in got) for s in range(steps)]
+            errs = [abs(e - g) for e, g in zip(expected, combined)]
+            ok = max(errs) < 5e-4
+            print(f"{tag}  ->  {'OK' if ok else 'FAIL'} maxerr={max(errs):.2e}")
+            failures += 0 if ok else 1
+        except Exception as e:  # noqa: BLE001
+            print(f"{tag}  ->  ERROR {type(e).__name__}: {str(e)[:200]}")
+            failures += 1
+    print(f"\n{args.combos - failures}/{args.combos} gems passed")
+    sys.exit(1 if failures else 0)
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--combos", type=int, default=8)
     ap.add_argument("--seed", type=int, default=0)
+    ap.add_argument("--gems", action="store_true",
+                    help="sweep GEMS LP configs instead of SP/LP seams")
     args = ap.parse_args()
+    if args.gems:
+        return gems_sweep(args, random.Random(args.seed))
     rng = random.Random(args.seed)
     steps, lr = 2, 0.01
     failures = 0
