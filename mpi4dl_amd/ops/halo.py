@@ -469,20 +469,20 @@ class _OverlapExactPadFn(torch.autograd.Function):
     directly (autograd sums it with this Function's cropped output)."""
 
     @staticmethod
-    def forward(ctx, x, h, exchanger):
+    def forward(ctx, x, h, exchanger, fill=0.0):
         ctx.h = h
         ctx.exchanger = exchanger
         hh, hw = _hpair(h)
-        return F.pad(x, (hw, hw, hh, hh))
+        return F.pad(x, (hw, hw, hh, hh), value=fill)
 
     @staticmethod
     def backward(ctx, gp):
         if gp.is_meta:
             hh, hw = _hpair(ctx.h)
             H, W = gp.shape[-2] - 2 * hh, gp.shape[-1] - 2 * hw
-            return gp[:, :, hh : hh + H, hw : hw + W], None, None
+            return gp[:, :, hh : hh + H, hw : hw + W], None, None, None
         g = ctx.exchanger.exchange_grad_padded(gp.contiguous(), ctx.h)
-        return g, None, None
+        return g, None, None, None
 
 
 class _HaloPadD2Fn(torch.autograd.Function):

@@ -51,6 +51,40 @@ def outer_pad_only(x: torch.Tensor, layout: TileLayout, tile: int, pad, fill: fl
     return x
 
 
+def _overlap_halo_apply(op, x, hh, hw, exchanger, grad_mode, fill=0.0):
+    """Generic halo/compute overlap for any k=2h+1 / stride-1 / pad-h op
+    (conv or pool): start the ring exchange, run ``op`` on the interior
+    while it flies, then compute the border bands from the landed ring
+    and concatenate. Output == op(halo_padded(x)) exactly; gradient per
+    grad_mode (exact: transposed exchange via _OverlapExactPadFn; drop:
+    ring treated as constant)."""
+    if grad_mode == "exact":
+        from .halo import _OverlapExactPadFn
+
+        xp = _OverlapExactPadFn.apply(x, (hh, hw), exchanger, fill)
+    else:
+        xp = F.pad(x, (hw, hw, hh, hh), value=fill)
+    with torch.no_grad():
+        finish = exchanger.exchange_padded_async(xp, (hh, hw))
+    interior = op(x)
+    with torch.no_grad():
+        finish()
+    H, W = x.shape[-2], x.shape[-1]
+    rows = []
+    if hh > 0:
+        rows.append(op(xp[:, :, 0 : 3 * hh, :]))
+    mid = [interior]
+    if hw > 0:
+        mid.insert(0, op(xp[:, :, hh : H + hh, 0 : 3 * hw].contiguous()))
+        mid.append(op(xp[:, :, hh : H + hh, W - hw : W + 2 * hw].contiguous()))
+    rows_mid = torch.cat(mid, dim=3) if len(mid) > 1 else interior
+    if hh > 0:
+        return torch.cat(
+            [rows[0], rows_mid, op(xp[:, :, H - hh : H + 2 * hh, :])], dim=2
+        )
+    return rows_mid
+
+
 class _SpatialBase(nn.Module):
     """Shared plumbing: layout, exchanger, rank map."""
 
@@ -168,37 +202,9 @@ class HaloConv2d(_SpatialBase):
             if isinstance(self.halo_len, tuple)
             else (self.halo_len, self.halo_len)
         )
-        if self.grad_mode == "exact":
-            from .halo import _OverlapExactPadFn
-
-            xp = _OverlapExactPadFn.apply(x, (hh, hw), self.exchanger)
-        else:
-            xp = F.pad(x, (hw, hw, hh, hh))
-        with torch.no_grad():
-            finish = self.exchanger.exchange_padded_async(xp, (hh, hw))
-        # interior while the halo is in flight
-        interior = self.conv(x)
-        with torch.no_grad():
-            finish()
-        H, W = x.shape[-2], x.shape[-1]
-        rows = []
-        if hh > 0:
-            rows.append(self.conv(xp[:, :, 0 : 3 * hh, :]))
-        mid = [interior]
-        if hw > 0:
-            mid.insert(0, self.conv(xp[:, :, hh : H + hh, 0 : 3 * hw].contiguous()))
-            mid.append(
-                self.conv(xp[:, :, hh : H + hh, W - hw : W + 2 * hw].contiguous())
-            )
-        rows_mid = torch.cat(mid, dim=3) if len(mid) > 1 else interior
-        if hh > 0:
-            out = torch.cat(
-                [rows[0], rows_mid, self.conv(xp[:, :, H - hh : H + 2 * hh, :])],
-                dim=2,
-            )
-        else:
-            out = rows_mid
-        return out
+        return _overlap_halo_apply(
+            self.conv, x, hh, hw, self.exchanger, self.grad_mode
+        )
 
 
 class HaloExchangeLayer(_SpatialBase):
@@ -301,6 +307,43 @@ class HaloPool2d(_SpatialBase):
             )
 
         # ---- tiled: halo-pad then pool with padding=0 -------------------
+        # stride-1 pools (AmoebaNet normal cells) get the same
+        # halo/compute overlap as convs: interior pool while the ring
+        # is on the wire, border bands after
+        if (
+            s == 1
+            and h > 0
+            and not self.d2
+            and self.exchanger is not None
+            and not x.is_meta
+            and H_loc > 2 * h
+            and W_loc > 2 * h
+            and os.environ.get("MPI4DL_NO_OVERLAP", "0") != "1"
+        ):
+            if self.kind == "max":
+                if on_gpu:
+                    from .native import native_maxpool
+
+                    op = lambda t: native_maxpool(t, k, s, 0)  # noqa: E731
+                else:
+                    op = lambda t: F.max_pool2d(t, k, s, padding=0)  # noqa: E731
+                return _overlap_halo_apply(
+                    op, x, h, h, self.exchanger, self.grad_mode, fill=fill
+                )
+            # avg: assemble include_pad sums, then global-geometry divisors
+            op = lambda t: F.avg_pool2d(  # noqa: E731
+                t, k, s, padding=0, count_include_pad=True
+            )
+            out = _overlap_halo_apply(
+                op, x, h, h, self.exchanger, self.grad_mode, fill=0.0
+            )
+            if self.count_include_pad:
+                return out
+            sums = out * float(k * k)
+            div = self._avg_divisors(
+                sums.shape[-2], sums.shape[-1], H_loc, W_loc, sums.device
+            )
+            return sums / div
         if self.d2 and h > 0:
             xp = (
                 outer_pad_only(x, self.layout, self.tile, h, fill=fill)

@@ -240,3 +240,46 @@ def _exchange_layer_body(rank, world, h, d2):
 
 def test_halo_exchange_layer():
     run_distributed(_exchange_layer_body, 2, (3, False))
+
+
+def _pool_overlap_body(rank, world, no_overlap):
+    import os
+
+    os.environ["MPI4DL_NO_OVERLAP"] = "1" if no_overlap else "0"
+    import torch
+
+    from mpi4dl_amd.ops.spatial_conv import HaloPool2d
+
+    torch.manual_seed(0)
+    x = torch.randn(2, 4, 16, 16 // world * world)
+    from mpi4dl_amd.ops.halo import TileLayout
+
+    layout = TileLayout(world, "vertical")
+    tile = layout.slice_input(x, rank).contiguous().requires_grad_(True)
+    outs = {}
+    for kind, cip in (("max", True), ("avg", False), ("avg", True)):
+        pool = HaloPool2d(
+            kind, 3, stride=1, padding=1, num_spatial_parts=world,
+            slice_method="vertical", spatial_local_rank=rank,
+            count_include_pad=cip,
+        )
+        y = pool(tile)
+        g = torch.ones_like(y)
+        (gx,) = torch.autograd.grad(y, tile, g)
+        outs[(kind, cip)] = (y.detach().tolist(), gx.tolist())
+    return outs
+
+
+def test_pool_overlap_equals_blocking():
+    """Stride-1 pool halo overlap == blocking path, forward AND exact
+    backward, for max and both avg semantics."""
+    a = run_distributed(_pool_overlap_body, 2, (False,))
+    b = run_distributed(_pool_overlap_body, 2, (True,))
+    for r in range(2):
+        for key in a[r]:
+            ya, ga = a[r][key]
+            yb, gb = b[r][key]
+            assert torch.allclose(torch.tensor(ya), torch.tensor(yb),
+                                  atol=1e-6), key
+            assert torch.allclose(torch.tensor(ga), torch.tensor(gb),
+                                  atol=1e-6), key
