@@ -331,9 +331,16 @@ class HaloPool2d(_SpatialBase):
                     op, x, h, h, self.exchanger, self.grad_mode, fill=fill
                 )
             # avg: assemble include_pad sums, then global-geometry divisors
-            op = lambda t: F.avg_pool2d(  # noqa: E731
-                t, k, s, padding=0, count_include_pad=True
-            )
+            if on_gpu:
+                from .native import native_avgpool
+
+                op = lambda t: native_avgpool(  # noqa: E731
+                    t, k, s, 0, include_pad=True
+                )
+            else:
+                op = lambda t: F.avg_pool2d(  # noqa: E731
+                    t, k, s, padding=0, count_include_pad=True
+                )
             out = _overlap_halo_apply(
                 op, x, h, h, self.exchanger, self.grad_mode, fill=0.0
             )
