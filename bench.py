@@ -14,9 +14,13 @@ Topology per N (strong scaling, fixed global batch):
   N=4: 2 spatial tiles on the first partition + 2 more LP stages (SP+PP)
   N=8: 4 spatial tiles + 4 more LP stages (SP+PP)
 
-Synthetic data (no network for datasets), random-init weights, bf16
-autocast compute (>= the reference's fp32), loss/optimizer step included
-in the timed region. One JSON line on rank 0.
+Synthetic data (no network for datasets), random-init weights,
+loss/optimizer step included in the timed region. One JSON line on
+rank 0. Default compute dtype is bf16 autocast; note the reference's
+published numbers are fp32 on different (NVIDIA) hardware, so
+vs_baseline is context, not a controlled same-precision A/B — run
+--dtype fp32 for the like-for-like precision datum
+(profiles/PERF_NOTES.md records both).
 """
 
 from __future__ import annotations
@@ -113,6 +117,9 @@ def main():
                     help="halo backward: 'drop' = reference semantics + "
                          "halo/compute overlap (benchmark default); "
                          "'exact' = transposed-gradient exchange")
+    ap.add_argument("--no-overlap-grads", action="store_true",
+                    help="disable bucketed allreduce-during-backward on the "
+                         "primary gradient group (overlap is the default)")
     args = ap.parse_args()
 
     from mpi4dl_amd.comm import Communicator, GradReducer, init_distributed
@@ -240,9 +247,21 @@ def main():
             eng.allreduce_and_update()
             return loss
     else:
+        # bucketed allreduce-during-backward on the primary gradient
+        # group (spatial tiles here), on by default — the reference gets
+        # this from its DDP wrap (mp_pipeline.py:92-124)
+        overlap = (
+            reducer.setup_overlap(eng.models)
+            if (spatial or comm.dp_size > 1) and not args.no_overlap_grads
+            else None
+        )
+
         def step():
             loss, _, _ = eng.run_step(x, y)
-            if spatial or comm.dp_size > 1:
+            if overlap is not None:
+                reducer.finish_overlap(eng.models)
+                reducer.apply_allreduce(eng.models, skip_group=overlap["group"])
+            elif spatial or comm.dp_size > 1:
                 reducer.apply_allreduce(eng.models)
             eng.update()
             return loss

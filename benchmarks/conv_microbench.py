@@ -66,6 +66,103 @@ def winograd_ab(args):
         print(f"{tag:<42} {tw:8.3f}ms {tm:8.3f}ms {tm/tw:5.2f}x  relerr {rel:.3g}")
 
 
+def model_pw_shapes(num_layers=18, num_filters=416, image=2048, mb=2):
+    """Every distinct 1x1 conv shape the flagship model runs (meta pass)."""
+    from mpi4dl_amd.models.amoebanet import amoebanetd
+    from mpi4dl_amd.ops.conv_native import NativeConv2d
+
+    shapes = {}
+    orig = NativeConv2d.forward
+
+    def probe(self, x):
+        kh, kw = (
+            (self.kernel_size, self.kernel_size)
+            if isinstance(self.kernel_size, int)
+            else self.kernel_size
+        )
+        if (kh, kw) == (1, 1):
+            s = self.stride[0] if isinstance(self.stride, tuple) else self.stride
+            key = (x.shape[1], self.out_channels, x.shape[2], x.shape[3], s)
+            shapes[key] = shapes.get(key, 0) + 1
+        return orig(self, x)
+
+    NativeConv2d.forward = probe
+    try:
+        with torch.device("meta"):
+            m = amoebanetd(1000, num_layers, num_filters)
+            m(torch.zeros(mb, 3, image, image))
+    finally:
+        NativeConv2d.forward = orig
+    return shapes
+
+
+def pw_ab(args):
+    """1x1 conv A/B: conv_pw streaming GEMM vs MIOpen, fwd + autograd,
+    on the exact shapes of the flagship model."""
+    from mpi4dl_amd.ops import backend
+    from mpi4dl_amd.ops.conv_native import native_conv2d
+
+    ge = backend.ext()
+    shapes = model_pw_shapes(mb=args.batch)
+    print(f"{'shape (xN uses)':<40} {'pw':>9} {'miopen':>9} {'ratio':>6}")
+    tot_n = tot_m = 0.0
+    for (C, K, H, W, s), uses in sorted(shapes.items()):
+        x = torch.randn(args.batch, C, H, W, device="cuda", dtype=torch.bfloat16)
+        w = torch.randn(K, C, 1, 1, device="cuda", dtype=torch.bfloat16) * 0.05
+        tn = timeit(lambda: ge.pw_fwd(x, w, None, s, s), args.iters)
+        tm = timeit(
+            lambda: torch.nn.functional.conv2d(x, w, stride=s), args.iters
+        )
+        y = ge.pw_fwd(x, w, None, s, s).float()
+        ref = torch.nn.functional.conv2d(x.float(), w.float(), stride=s)
+        rel = (y - ref).abs().max().item() / max(ref.abs().max().item(), 1e-3)
+        tag = f"C{C}->K{K} {H}x{W} s{s} (x{uses})"
+        print(f"{tag:<40} {tn:8.3f}ms {tm:8.3f}ms {tm/tn:5.2f}x  relerr {rel:.3g}",
+              flush=True)
+        tot_n += tn * uses
+        tot_m += tm * uses
+        # autograd legs
+        xg = x.clone().requires_grad_(True)
+        wf = w.float().requires_grad_(True)
+        g = torch.randn_like(ref).to(torch.bfloat16)
+
+        def nat_step():
+            yy = native_conv2d(xg, wf, None, (s, s), (0, 0))
+            yy.backward(g)
+            xg.grad = None
+            wf.grad = None
+
+        x2 = x.clone().requires_grad_(True)
+        w2 = w.float().requires_grad_(True)
+
+        def ref_step():
+            yy = torch.nn.functional.conv2d(
+                x2, w2.to(torch.bfloat16), stride=s
+            )
+            yy.backward(g)
+            x2.grad = None
+            w2.grad = None
+
+        # grad correctness
+        yy = native_conv2d(xg, wf, None, (s, s), (0, 0))
+        yy.backward(g)
+        ry = torch.nn.functional.conv2d(x2.float(), w2, stride=s)
+        ry.backward(g.float())
+        relx = (xg.grad.float() - x2.grad).abs().max().item() / max(
+            x2.grad.abs().max().item(), 1e-3
+        )
+        relw = (wf.grad - w2.grad).abs().max().item() / max(
+            w2.grad.abs().max().item(), 1e-3
+        )
+        xg.grad = wf.grad = x2.grad = w2.grad = None
+        tn2 = timeit(nat_step, max(args.iters // 2, 3))
+        tm2 = timeit(ref_step, max(args.iters // 2, 3))
+        print(f"{'':<40} f+b {tn2:8.3f}ms {tm2:8.3f}ms {tm2/tn2:5.2f}x  "
+              f"relgx {relx:.3g} relgw {relw:.3g}", flush=True)
+    print(f"\nuse-weighted fwd total: pw {tot_n:.2f}ms vs miopen {tot_m:.2f}ms "
+          f"({tot_m/max(tot_n,1e-9):.2f}x)")
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--iters", type=int, default=20)
@@ -73,9 +170,14 @@ def main():
     ap.add_argument("--winograd", action="store_true",
                     help="A/B winograd_bmm_conv2d vs MIOpen on the 3x3 "
                          "stride-1 shapes (round-2 staging)")
+    ap.add_argument("--pw", action="store_true",
+                    help="A/B the conv_pw 1x1 kernel vs MIOpen on the "
+                         "flagship model's 1x1 shapes")
     args = ap.parse_args()
     if args.winograd:
         return winograd_ab(args)
+    if args.pw:
+        return pw_ab(args)
     from mpi4dl_amd.ops import backend
 
     ge = backend.ext()

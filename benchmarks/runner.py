@@ -169,9 +169,21 @@ def make_engines(args, mode):
                 optimizer=opt, **eng_kw,
             )
 
+        # bucketed allreduce-during-backward on the primary gradient group
+        # (default on; reference equivalent: DDP wrap mp_pipeline.py:92-124)
+        overlap = (
+            reducer.setup_overlap(eng.models)
+            if not getattr(args, "no_overlap_grads", False)
+            else None
+        )
+
         def step(x, y):
             loss, corr, seen = eng.run_step(x, y)
-            reducer.apply_allreduce(eng.models)
+            if overlap is not None:
+                reducer.finish_overlap(eng.models)
+                reducer.apply_allreduce(eng.models, skip_group=overlap["group"])
+            else:
+                reducer.apply_allreduce(eng.models)
             eng.update()
             return loss, corr, seen
 
@@ -192,6 +204,7 @@ def make_engines(args, mode):
             gen1, gen2, args.batch_size, args.parts, comm,
             replications=args.times,
             enable_comm_opt=args.enable_master_comm_opt,
+            fp16_allreduce=args.fp16_allreduce,
             optimizer=mkopt(gen1.models), **eng_kw,
         )
         eng.train_model2.optimizer = mkopt(gen2.models)
@@ -206,6 +219,7 @@ def make_engines(args, mode):
             gen1, gen2, args.batch_size, args.parts, comm,
             slice_method=args.slice_method, replications=args.times,
             enable_comm_opt=args.enable_master_comm_opt,
+            fp16_allreduce=args.fp16_allreduce,
             optimizer=mkopt(gen1.models), **eng_kw,
         )
         eng.train_model2.optimizer = mkopt(gen2.models)
@@ -218,6 +232,9 @@ def make_engines(args, mode):
     return step, comm, {
         "engine": eng,
         "gen": gen1,
+        "gen2": gen2,
+        "optimizer": eng.train_model1.optimizer,
+        "optimizer2": eng.train_model2.optimizer,
         "batch_per_step": 2 * args.times * args.batch_size,
     }
 
@@ -262,12 +279,15 @@ def run_training(args, mode):
             x, y = x.cuda(), y.cuda()
 
     if args.resume and args.checkpoint_dir:
-        ckpt.load_checkpoint(args.checkpoint_dir, extras["gen"].models,
-                             extras.get("optimizer"), comm)
+        gen2 = extras.get("gen2")
+        ckpt.load_checkpoint(
+            args.checkpoint_dir, extras["gen"].models, extras.get("optimizer"),
+            comm, module2=gen2.models if gen2 is not None else None,
+            optimizer2=extras.get("optimizer2"),
+        )
         eng = extras["engine"]
-        if hasattr(eng, "sync_models"):
-            # GEMS: the checkpoint restored replica 1 only — mirror it
-            # onto replica 2 (otherwise the replicas diverge from step 0)
+        if gen2 is None and hasattr(eng, "sync_models"):
+            # old single-replica checkpoint: mirror replica 1 onto 2
             eng.sync_models()
 
     times = []
@@ -293,8 +313,12 @@ def run_training(args, mode):
                     epoch, it, loss, corr / max(seen, 1), dt * 1e3,
                 )
     if args.checkpoint_dir and not args.resume:
-        ckpt.save_checkpoint(args.checkpoint_dir, extras["gen"].models,
-                             extras.get("optimizer"), comm)
+        gen2 = extras.get("gen2")
+        ckpt.save_checkpoint(
+            args.checkpoint_dir, extras["gen"].models, extras.get("optimizer"),
+            comm, module2=gen2.models if gen2 is not None else None,
+            optimizer2=extras.get("optimizer2"),
+        )
 
     if args.enable_evaluation:
         # forward-only pass over held-out batches (CIFAR test split when

@@ -516,7 +516,7 @@ class GradReducer:
                 dist.all_reduce(fg.buffer, group=group)
         fg.rescale_(divide_by if divide_by is not None else float(n))
 
-    def apply_allreduce(self, module: torch.nn.Module):
+    def apply_allreduce(self, module: torch.nn.Module, skip_group=None):
         """Standard path: spatial group first (tile replicas), then outer DP
         (reference apply_allreduce, comm.py:506-514).
 
@@ -525,18 +525,45 @@ class GradReducer:
         outer-DP groups each see the full weight grad of a batch shard,
         so those AVERAGE. This reproduces the serial gradient exactly
         (the reference folds the same arithmetic into divide_bs,
-        comm.py:440-458)."""
+        comm.py:440-458).
+
+        ``skip_group``: a group already reduced by the bucketed-overlap
+        hooks (finish_overlap) — only the remaining groups run here."""
         if self.comm.ENABLE_SPATIAL and self.comm.split_rank < self.comm.spatial_size:
-            self.allreduce_grads(
-                module,
-                self.comm.spatial_allreduce_groups.get(self.comm.split_rank),
-                divide_by=1.0,
-            )
+            g = self.comm.spatial_allreduce_groups.get(self.comm.split_rank)
+            if g is not skip_group:
+                self.allreduce_grads(module, g, divide_by=1.0)
         if self.comm.LOCAL_DP_LP > 1 and self.comm.split_rank >= self.comm.spatial_size:
-            self.allreduce_grads(
-                module, self.comm.local_dp_groups.get(self.comm.split_rank)
-            )
-        self.allreduce_grads(module, self.comm.outer_dp_group)
+            g = self.comm.local_dp_groups.get(self.comm.split_rank)
+            if g is not skip_group:
+                self.allreduce_grads(module, g)
+        if self.comm.outer_dp_group is not skip_group:
+            self.allreduce_grads(module, self.comm.outer_dp_group)
+
+    def primary_reduce_group(self):
+        """The innermost (largest-traffic) gradient group for this rank —
+        the one worth overlapping with backward: the spatial tile group
+        on tile ranks, the local-DP group on LP ranks under LOCAL_DP_LP,
+        else the outer-DP group. Returns (group, divide_by)."""
+        comm = self.comm
+        if comm.ENABLE_SPATIAL and comm.split_rank < comm.spatial_size:
+            return comm.spatial_allreduce_groups.get(comm.split_rank), 1.0
+        if comm.LOCAL_DP_LP > 1 and comm.split_rank >= comm.spatial_size:
+            return comm.local_dp_groups.get(comm.split_rank), None
+        return comm.outer_dp_group, None
+
+    def setup_overlap(self, module: torch.nn.Module, bucket_mb: float = 25.0):
+        """Default production wiring (reference DDP wrap,
+        mp_pipeline.py:92-124): register bucketed overlap on the primary
+        group. Returns the overlap state (None when no group exists).
+        The engine's run_step arms it on the last micro-batch; call
+        ``finish_overlap`` then ``apply_allreduce(skip_group=...)``."""
+        group, divide = self.primary_reduce_group()
+        if group is None:
+            return None
+        state = self.prepare_overlap(module, group, divide_by=divide,
+                                     bucket_mb=bucket_mb)
+        return state
 
     # -- bucketed overlap (DDP-equivalent, reference mp_pipeline.py:92-124
     # wraps partitions in torch DDP; here the same flat buffer the

@@ -1,0 +1,363 @@
+// Pointwise (1x1) convolution as a streaming MFMA GEMM for gfx950.
+//
+// OUT[K][P] = W[K][C] @ X[C][P],  P = N*OH*OW (NCHW: pixels contiguous
+// per channel). These shapes are BANDWIDTH-bound on MI355X (arithmetic
+// intensity K*C/(K+C) ~ 50-200 FLOP/B vs the ~400 machine balance), so
+// the design goal is streaming efficiency at high occupancy, not MFMA
+// peak:
+//  * B (pixels) is staged global->LDS with `global_load_lds` dwordx4
+//    (16 B/lane), source addresses pre-swizzled so the lane-linear LDS
+//    image IS the ds_read_b64_tr_b16 fragment layout (guide §5 "swizzle
+//    on the SOURCE address" rule) — no ds_write pass, no repack;
+//  * A (weights) is tiny and L2-resident: each lane loads its MFMA
+//    fragment straight from global (no LDS round-trip — guide §5
+//    "operand streamed once and not shared" row);
+//  * M-adaptive tiles: BM in {32, 64, 128} via MFRAG template so K=26
+//    or K=52 layers (AmoebaNet c/4 bottlenecks) don't waste 60-80% of
+//    the M tile;
+//  * small LDS (2 x 8.3 KB) -> 8 waves/SIMD occupancy does the latency
+//    hiding (guide: "pure HBM streaming at high occupancy" regime).
+//
+// Also used for the 1x1 stride-1 BACKWARD-DATA (gx = pw(go, w^T)) and
+// the stride-2 backward-data via the scatter2 epilogue
+// (gx[:, :, ::2, ::2] = pw(go, w^T), rest exact zeros) — VERDICT r1
+// item 9: the conv triple fully in-house.
+//
+// Reference behaviour replaced: cuDNN/MIOpen 1x1 conv under torchgems
+// (/root/reference/src/torchgems/spatial.py:1027); in the round-1
+// profile these were ~20-30% of the timed window as hipBLASLt Cijk_*
+// GEMMs + batched_transpose glue.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include <cstdint>
+
+namespace conv_pw {
+
+using bf16 = __hip_bfloat16;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(4))) short s16x4;
+typedef __attribute__((ext_vector_type(8))) short s16x8;
+
+#define PW_BN 128       // pixels per block
+#define PW_BK 32        // channels per k-step
+#define LDSB_BLK 520    // shorts per 16-px LDS block (512 + 8 pad)
+
+struct PwGeom {
+  int N, C, K;
+  int HW_in;        // input pixels per image (H*W)
+  int OH, OW;       // output dims (per image)
+  int sh, sw;       // stride (1 or 2); pad is always 0 for 1x1
+  int W_in;         // input row length
+  int osh, osw;     // output scatter strides (1 = dense)
+  int oW;           // output row length in elements (for scatter)
+  int oHW;          // output pixels per image in the OUT TENSOR
+};
+
+// lane -> (k, b) map such that LDS offset l*8 shorts == tr16 layout
+// offset of (k, px0=8b): off/8 = 16*(k>>3) + 8*((k>>2)&1) + 2*(k&3) + b
+__device__ __forceinline__ void lane_kb(int l, int& k, int& b) {
+  k = ((l >> 4) << 3) + (((l >> 3) & 1) << 2) + ((l >> 1) & 3);
+  b = l & 1;
+}
+
+// ---------------------------------------------------------------------------
+// Forward kernel. MFRAG = 16-row fragments per wave (BM = 2*16*MFRAG).
+// 256 threads = 4 waves as 2x2; wave tile (16*MFRAG) x 64.
+// ---------------------------------------------------------------------------
+
+template <int MFRAG, int SW>
+__global__ __launch_bounds__(256) void pw_kernel(
+    const bf16* __restrict__ x, const bf16* __restrict__ w,
+    const float* __restrict__ bias, bf16* __restrict__ out, PwGeom g) {
+  constexpr int BM = 32 * MFRAG;
+  const int m_tiles = (g.K + BM - 1) / BM;
+  // XCD-aware bijective swizzle (guide T1)
+  const int nwg = gridDim.x;
+  const int q8 = nwg >> 3, r8 = nwg & 7;
+  const int xcd = blockIdx.x & 7, sub = blockIdx.x >> 3;
+  int bid = (xcd < r8 ? xcd * (q8 + 1) : r8 * (q8 + 1) + (xcd - r8) * q8) + sub;
+  const int mt = bid % m_tiles;
+  const int pt = bid / m_tiles;          // pixel-tile index
+  const int OHW = g.OH * g.OW;
+  const int ptiles_per_img = OHW / PW_BN;
+  const int n = pt / ptiles_per_img;
+  const int q0 = (pt - n * ptiles_per_img) * PW_BN;  // first out pixel
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = wid >> 1, wn = wid & 1;
+  const int k0out = mt * BM;
+
+  __shared__ __attribute__((aligned(16))) short lds[2 * 8 * LDSB_BLK];
+  auto ldsB = [&](int buf) { return lds + buf * 8 * LDSB_BLK; };
+
+  f32x4 acc[MFRAG][4];
+#pragma unroll
+  for (int i = 0; i < MFRAG; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  const int64_t in_n = (int64_t)n * g.C * g.HW_in;
+  const int ksteps = (g.C + PW_BK - 1) / PW_BK;
+  const int full_ksteps = g.C / PW_BK;
+
+  int lk, lb;
+  lane_kb(lane, lk, lb);
+
+  // per-lane source pixel offset for the glds path (stride 1 only):
+  // wave wid stages blocks pb = wid*2 and wid*2+1
+  // source = x[n][c0+lk][q0 + pb*16 + 8*lb .. +8]
+  const int64_t src_base =
+      in_n + (int64_t)lk * g.HW_in + q0 + lb * 8;  // + c0*HW + pb*16
+
+  auto stage_glds = [&](int buf, int c0) {
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const int pb = wid * 2 + i;
+      const bf16* src = x + src_base + (int64_t)c0 * g.HW_in + pb * 16;
+      auto* dst = (__attribute__((address_space(3))) void*)(ldsB(buf) +
+                                                            pb * LDSB_BLK);
+      __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1)))
+                                        unsigned int*)src,
+                                       (__attribute__((address_space(3)))
+                                        unsigned int*)dst,
+                                       16, 0, 0);
+    }
+  };
+
+  // register-staged path: tail k-step (c0+32 > C) or stride-2 gather.
+  // 512 (k, px8) chunks -> 2 per thread, zero-filled beyond C.
+  auto stage_reg = [&](int buf, int c0) {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      const int idx = it * 256 + tid;
+      const int kk = idx & 31;
+      const int pxc = idx >> 5;
+      const int px0 = pxc * 8;
+      const int c = c0 + kk;
+      short v[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (c < g.C) {
+        if (SW == 1) {
+          const short* src =
+              (const short*)(x + in_n + (int64_t)c * g.HW_in + q0 + px0);
+          *(s16x8*)v = *(const s16x8*)src;
+        } else {
+          // stride 2: out pixel q -> input (2*oh)*W_in + 2*ow.
+          // chunk stays inside one out row (OW % 8 == 0).
+          const int q = q0 + px0;
+          const int oh = q / g.OW;
+          const int ow = q - oh * g.OW;
+          const short* src = (const short*)(x + in_n + (int64_t)c * g.HW_in +
+                                            (int64_t)(oh * g.sh) * g.W_in +
+                                            ow * 2);
+          short raw[16];
+          *(s16x8*)raw = *(const s16x8*)src;
+          *(s16x8*)(raw + 8) = *(const s16x8*)(src + 8);
+#pragma unroll
+          for (int e = 0; e < 8; ++e) v[e] = raw[2 * e];
+        }
+      }
+      const int base = pxc * LDSB_BLK + ((kk >> 3) << 7) +
+                       (((kk >> 2) & 1) << 6) + ((kk & 3) << 4);
+      *(s16x8*)(ldsB(buf) + base) = *(const s16x8*)v;
+    }
+  };
+
+  auto stage = [&](int buf, int step) {
+    const int c0 = step * PW_BK;
+    if (SW == 1 && step < full_ksteps)
+      stage_glds(buf, c0);
+    else
+      stage_reg(buf, c0);
+  };
+
+  // A fragments straight from global (L2-resident weights).
+  // fragment (mf): rows k0out + wm*16*MFRAG + mf*16 + (lane&15),
+  // k-chunk c0 + (lane>>4)*8.
+  const int arow_base = k0out + wm * 16 * MFRAG + (lane & 15);
+  const int acol = (lane >> 4) << 3;
+
+  stage(0, 0);
+
+  for (int step = 0; step < ksteps; ++step) {
+    const int buf = step & 1;
+    const int c0 = step * PW_BK;
+    __syncthreads();
+    if (step + 1 < ksteps) stage(buf ^ 1, step + 1);
+
+    s16x8 afrag[MFRAG];
+    const bool full = (c0 + PW_BK <= g.C);
+#pragma unroll
+    for (int mf = 0; mf < MFRAG; ++mf) {
+      int row = arow_base + mf * 16;
+      if (row >= g.K) row = g.K - 1;  // clamp: masked at epilogue
+      const short* ap = (const short*)w + (int64_t)row * g.C + c0 + acol;
+      if (full) {
+        afrag[mf] = *(const s16x8*)ap;
+      } else {
+        short v[8];
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          v[e] = (c0 + acol + e < g.C) ? ap[e] : (short)0;
+        afrag[mf] = *(const s16x8*)v;
+      }
+    }
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      const int pb = wn * 4 + nf;
+      __attribute__((address_space(3))) short* bbase =
+          (__attribute__((address_space(3))) short*)(ldsB(buf)) +
+          pb * LDSB_BLK + ((lane >> 4) << 7) + ((lane & 15) << 2);
+      s16x4 b0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+          (__attribute__((address_space(3))) s16x4*)bbase);
+      s16x4 b1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+          (__attribute__((address_space(3))) s16x4*)(bbase + 64));
+      s16x8 bfrag;
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        bfrag[e] = b0[e];
+        bfrag[e + 4] = b1[e];
+      }
+#pragma unroll
+      for (int mf = 0; mf < MFRAG; ++mf)
+        acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag[mf], bfrag, acc[mf][nf], 0, 0, 0);
+    }
+  }
+
+  // epilogue: bias + bf16 store (dense or scatter2)
+  const int64_t out_n = (int64_t)n * g.K * g.oHW;
+#pragma unroll
+  for (int mf = 0; mf < MFRAG; ++mf) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int kout = k0out + wm * 16 * MFRAG + mf * 16 +
+                       ((lane >> 4) << 2) + reg;
+      if (kout >= g.K) continue;
+      const float bv = bias ? bias[kout] : 0.f;
+      const int64_t obase = out_n + (int64_t)kout * g.oHW;
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        const int q = q0 + wn * 64 + nf * 16 + (lane & 15);
+        int64_t oq;
+        if (g.osh == 1 && g.osw == 1) {
+          oq = q;
+        } else {
+          const int oh = q / g.OW;
+          const int ow = q - oh * g.OW;
+          oq = (int64_t)(oh * g.osh) * g.oW + ow * g.osw;
+        }
+        out[obase + oq] = (bf16)(acc[mf][nf][reg] + bv);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Host wrappers
+// ---------------------------------------------------------------------------
+
+static void launch_pw(const torch::Tensor& x, const torch::Tensor& w,
+                      const float* bias, torch::Tensor& out, PwGeom g) {
+  const int K = g.K;
+  const int mfrag = (K <= 32) ? 1 : (K <= 64) ? 2 : 4;
+  const int BM = 32 * mfrag;
+  const int m_tiles = (K + BM - 1) / BM;
+  const int64_t ptiles = (int64_t)g.N * (g.OH * g.OW / PW_BN);
+  const int64_t blocks = ptiles * m_tiles;
+  TORCH_CHECK(blocks > 0 && blocks < (1LL << 31), "pw grid size");
+  auto stream = at::cuda::getCurrentCUDAStream();
+  auto launch = [&](auto kern) {
+    hipLaunchKernelGGL(kern, dim3((uint32_t)blocks), dim3(256), 0,
+                       stream.stream(), (const bf16*)x.data_ptr(),
+                       (const bf16*)w.data_ptr(), bias,
+                       (bf16*)out.data_ptr(), g);
+  };
+  if (g.sw == 1) {
+    if (mfrag == 1) launch(pw_kernel<1, 1>);
+    else if (mfrag == 2) launch(pw_kernel<2, 1>);
+    else launch(pw_kernel<4, 1>);
+  } else {
+    if (mfrag == 1) launch(pw_kernel<1, 2>);
+    else if (mfrag == 2) launch(pw_kernel<2, 2>);
+    else launch(pw_kernel<4, 2>);
+  }
+}
+
+// forward: x [N,C,H,W] bf16, w [K,C,1,1] bf16, stride (sh,sw), pad 0
+torch::Tensor pw_fwd(torch::Tensor x, torch::Tensor w,
+                     c10::optional<torch::Tensor> bias, int64_t sh,
+                     int64_t sw) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && w.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 &&
+              w.scalar_type() == torch::kBFloat16, "pw_fwd: bf16 only");
+  PwGeom g;
+  g.N = (int)x.size(0);
+  g.C = (int)x.size(1);
+  const int H = (int)x.size(2), W = (int)x.size(3);
+  g.K = (int)w.size(0);
+  TORCH_CHECK(w.numel() == (int64_t)g.K * g.C, "pw_fwd: w must be [K,C,1,1]");
+  g.HW_in = H * W;
+  g.W_in = W;
+  g.sh = (int)sh;
+  g.sw = (int)sw;
+  g.OH = (H - 1) / g.sh + 1;
+  g.OW = (W - 1) / g.sw + 1;
+  g.osh = 1; g.osw = 1;
+  g.oW = g.OW;
+  g.oHW = g.OH * g.OW;
+  TORCH_CHECK((g.OH * g.OW) % PW_BN == 0, "pw_fwd: OH*OW % 128 != 0");
+  if (g.sw != 1) TORCH_CHECK(g.OW % 8 == 0, "pw_fwd: OW % 8 != 0 at stride 2");
+  auto out = torch::empty({(int64_t)g.N, (int64_t)g.K, (int64_t)g.OH,
+                           (int64_t)g.OW},
+                          x.options());
+  const float* bptr = nullptr;
+  torch::Tensor b32;
+  if (bias.has_value()) {
+    b32 = bias->to(torch::kFloat).contiguous();
+    bptr = b32.data_ptr<float>();
+  }
+  launch_pw(x, w, bptr, out, g);
+  return out;
+}
+
+// stride-2 backward-data: gx[:, :, ::sh, ::sw] = pw(go, w^T), rest 0.
+// go [N,K,OH,OW]; wt [C,K] (w transposed, contiguous); returns [N,C,H,W].
+torch::Tensor pw_bwd_data_strided(torch::Tensor go, torch::Tensor wt,
+                                  int64_t H, int64_t W, int64_t sh,
+                                  int64_t sw) {
+  TORCH_CHECK(go.is_cuda() && go.is_contiguous() && wt.is_contiguous());
+  TORCH_CHECK(go.scalar_type() == torch::kBFloat16 &&
+              wt.scalar_type() == torch::kBFloat16);
+  PwGeom g;
+  g.N = (int)go.size(0);
+  g.C = (int)go.size(1);            // reduction dim = K of fwd
+  g.K = (int)wt.size(0);            // output channels = C of fwd
+  g.OH = (int)go.size(2);
+  g.OW = (int)go.size(3);
+  g.HW_in = g.OH * g.OW;
+  g.W_in = g.OW;
+  g.sh = 1; g.sw = 1;               // reading go densely
+  g.osh = (int)sh; g.osw = (int)sw; // scattering into gx
+  g.oW = (int)W;
+  g.oHW = (int)(H * W);
+  TORCH_CHECK((g.OH * g.OW) % PW_BN == 0,
+              "pw_bwd_data: OH*OW % 128 != 0");
+  auto gx = torch::zeros({(int64_t)g.N, (int64_t)g.K, H, W}, go.options());
+  launch_pw(go, wt, nullptr, gx, g);
+  return gx;
+}
+
+}  // namespace conv_pw
+
+void register_conv_pw(pybind11::module_& m) {
+  m.def("pw_fwd", &conv_pw::pw_fwd,
+        "1x1 conv forward (streaming MFMA GEMM, bf16 NCHW)");
+  m.def("pw_bwd_data_strided", &conv_pw::pw_bwd_data_strided,
+        "1x1 strided conv backward-data (scatter epilogue)");
+}

@@ -1060,9 +1060,11 @@ torch::Tensor bn_bwd_apply(torch::Tensor go, torch::Tensor x, torch::Tensor y,
 }  // namespace
 
 void register_conv_mfma(pybind11::module_& m);
+void register_conv_pw(pybind11::module_& m);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   register_conv_mfma(m);
+  register_conv_pw(m);
   m.def("halo_pack", &halo_copy<0>, "pack halo strips tile->flat buffer");
   m.def("halo_unpack", &halo_copy<1>, "unpack halo strips buffer->tile");
   m.def("halo_unpack_add", &halo_copy<2>, "accumulate grad strips into tile");

@@ -94,9 +94,11 @@ class FactorizedReduce(nn.Module):
 
     def __init__(self, cin, cout, mknorm=nn.BatchNorm2d):
         super().__init__()
+        from ..ops.conv_native import NativeConv2d
+
         self.relu = nn.ReLU(inplace=False)
-        self.conv1 = nn.Conv2d(cin, cout // 2, 1, stride=2, bias=False)
-        self.conv2 = nn.Conv2d(cin, cout // 2, 1, stride=2, bias=False)
+        self.conv1 = NativeConv2d(cin, cout // 2, 1, stride=2, bias=False)
+        self.conv2 = NativeConv2d(cin, cout // 2, 1, stride=2, bias=False)
         self.bn = mknorm(cout)
 
     def forward(self, x):
@@ -106,9 +108,11 @@ class FactorizedReduce(nn.Module):
 
 
 def relu_conv_bn(cin, cout, mknorm=nn.BatchNorm2d):
+    from ..ops.conv_native import NativeConv2d
+
     return nn.Sequential(
         nn.ReLU(inplace=False),
-        nn.Conv2d(cin, cout, 1, stride=1, bias=False),
+        NativeConv2d(cin, cout, 1, stride=1, bias=False),
         mknorm(cout),
     )
 
@@ -155,24 +159,30 @@ def make_op(name: str, c: int, stride: int, ctx, mknorm,
     # none ops get the raw state), removing ~6 redundant full-tensor ReLU
     # kernels per cell. The leading ReLU module is therefore dropped here.
     if name == "conv_1x1":
+        from ..ops.conv_native import NativeConv2d
+
         return nn.Sequential(
-            nn.Conv2d(c, c, 1, stride=stride, bias=False),
+            NativeConv2d(c, c, 1, stride=stride, bias=False),
             mknorm(c),
         )
     if name == "conv_3x3":
+        from ..ops.conv_native import NativeConv2d
+
         return nn.Sequential(
-            nn.Conv2d(c, c // 4, 1, bias=False),
+            NativeConv2d(c, c // 4, 1, bias=False),
             _bn_relu(mknorm, c // 4),
             nn.Identity(),
             _conv(c // 4, c // 4, 3, stride=stride, ctx=ctx),
             _bn_relu(mknorm, c // 4),
             nn.Identity(),
-            nn.Conv2d(c // 4, c, 1, bias=False),
+            NativeConv2d(c // 4, c, 1, bias=False),
             mknorm(c),
         )
     if name == "conv_1x7_7x1":
+        from ..ops.conv_native import NativeConv2d
+
         return nn.Sequential(
-            nn.Conv2d(c, c // 4, 1, stride=1, bias=False),
+            NativeConv2d(c, c // 4, 1, stride=1, bias=False),
             _bn_relu(mknorm, c // 4),
             nn.Identity(),
             _conv(c // 4, c // 4, (1, 7), stride=(1, stride), padding=(0, 3), ctx=ctx),
@@ -181,7 +191,7 @@ def make_op(name: str, c: int, stride: int, ctx, mknorm,
             _conv(c // 4, c // 4, (7, 1), stride=(stride, 1), padding=(3, 0), ctx=ctx),
             _bn_relu(mknorm, c // 4),
             nn.Identity(),
-            nn.Conv2d(c // 4, c, 1, stride=1, bias=False),
+            NativeConv2d(c // 4, c, 1, stride=1, bias=False),
             mknorm(c),
         )
     raise ValueError(f"unknown op {name}")

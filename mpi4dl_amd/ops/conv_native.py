@@ -29,17 +29,41 @@ def _pair(v):
     return (v, v) if isinstance(v, int) else tuple(v)
 
 
+def _pw_geom_ok(shape, sh, sw):
+    """pw kernel constraints: per-image output pixels % 128 == 0; at
+    stride 2 additionally even dims and OW % 8 (vector evens loads)."""
+    H, W = shape[-2], shape[-1]
+    oh = (H - 1) // sh + 1
+    ow = (W - 1) // sw + 1
+    if (oh * ow) % 128 != 0:
+        return False
+    if (sh, sw) != (1, 1):
+        return (
+            sh == 2 and sw == 2 and H % 2 == 0 and W % 2 == 0 and ow % 8 == 0
+        )
+    return True
+
+
 class ConvFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, bias, sh, sw, ph, pw):
         ge = backend.ext()
         xb = x.contiguous().to(torch.bfloat16)
         wb = w.contiguous().to(torch.bfloat16)
-        y = ge.conv_fwd(xb, wb, bias, sh, sw, ph, pw)
+        K, C, R, S = wb.shape
+        is_pw = (
+            R == 1 and S == 1 and ph == 0 and pw == 0
+            and _pw_geom_ok(xb.shape, sh, sw)
+        )
+        if is_pw:
+            y = ge.pw_fwd(xb, wb, bias, sh, sw)
+        else:
+            y = ge.conv_fwd(xb, wb, bias, sh, sw, ph, pw)
         ctx.save_for_backward(xb, wb)
         ctx.geom = (sh, sw, ph, pw)
         ctx.wdtype = w.dtype
         ctx.has_bias = bias is not None
+        ctx.is_pw = is_pw
         return y
 
     @staticmethod
@@ -51,7 +75,15 @@ class ConvFn(torch.autograd.Function):
         K, C, R, S = wb.shape
         gw = ge.conv_bwd_weight(go, xb, R, S, sh, sw, ph, pw).to(ctx.wdtype)
         gb = go.sum(dim=(0, 2, 3)).to(ctx.wdtype) if ctx.has_bias else None
-        if sh == 1 and sw == 1:
+        if ctx.is_pw:
+            wt = wb.view(K, C).t().contiguous()
+            if (sh, sw) == (1, 1):
+                gx = ge.pw_fwd(go, wt, None, 1, 1)
+            else:
+                gx = ge.pw_bwd_data_strided(
+                    go, wt, xb.shape[-2], xb.shape[-1], sh, sw
+                )
+        elif sh == 1 and sw == 1:
             wt = wb.transpose(0, 1).flip(2, 3).contiguous()
             gx = ge.conv_fwd(go, wt, None, 1, 1, R - 1 - ph, S - 1 - pw)
         else:
@@ -84,6 +116,14 @@ def _dispatchable(x, conv: nn.Conv2d) -> bool:
     sh, sw = _pair(conv.stride)
     ph, pw = _pair(conv.padding)
     kh, kw = _pair(conv.kernel_size)
+    # 1x1: the conv_pw streaming GEMM (round 2) — pixel axis spans rows,
+    # so no OW minimum; all three legs native
+    if (
+        kh == 1 and kw == 1 and ph == 0 and pw == 0
+        and os.environ.get("MPI4DL_NATIVE_PW", "1") != "0"
+        and _pw_geom_ok(x.shape, sh, sw)
+    ):
+        return True
     ow = (x.shape[-1] + 2 * pw - kw) // sw + 1
     if ow < _MIN_OW:
         return False

@@ -18,6 +18,8 @@ the tile group.
 
 from __future__ import annotations
 
+import os
+
 import torch
 import torch.distributed as dist
 import torch.nn as nn
@@ -62,7 +64,12 @@ class TileBatchNorm2d(nn.BatchNorm2d):
         )
 
     def forward(self, x):
-        if x.is_cuda and not x.is_meta and (x.shape[-2] * x.shape[-1]) % 8 == 0:
+        if (
+            x.is_cuda
+            and not x.is_meta
+            and (x.shape[-2] * x.shape[-1]) % 8 == 0
+            and os.environ.get("MPI4DL_NATIVE_BN", "1") != "0"
+        ):
             return self._forward_native(x)
         if not self._use_sync(x):
             y = super().forward(x)

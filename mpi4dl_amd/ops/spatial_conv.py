@@ -168,11 +168,21 @@ class HaloConv2d(_SpatialBase):
                 if isinstance(self.halo_len, tuple)
                 else (self.halo_len, self.halo_len)
             )
+            kh, kw = (
+                (self.kernel_size, self.kernel_size)
+                if isinstance(self.kernel_size, int)
+                else self.kernel_size
+            )
             if (
                 self.exchanger is not None
                 and not x.is_meta
                 and self.stride in (1, (1, 1))
                 and (hh or hw)
+                # band decomposition is only valid for k == 2*halo+1 per
+                # axis (i.e. padding == (k-1)//2); other paddings take the
+                # blocking halo_pad path below
+                and kh == 2 * hh + 1
+                and kw == 2 * hw + 1
                 # interior conv must be valid: tile bigger than the kernel
                 and x.shape[-2] > 2 * hh
                 and x.shape[-1] > 2 * hw
@@ -284,7 +294,11 @@ class HaloPool2d(_SpatialBase):
     def forward(self, x):
         h = self.halo_len
         k, s = self.kernel_size, self.stride
-        on_gpu = x.is_cuda and not x.is_meta
+        on_gpu = (
+            x.is_cuda
+            and not x.is_meta
+            and os.environ.get("MPI4DL_NATIVE_POOL", "1") != "0"
+        )
         # max pool pads with -inf so image-boundary windows match the
         # single-GPU op exactly (zero-pad would win over negative inputs)
         fill = float("-inf") if self.kind == "max" else 0.0
@@ -313,6 +327,7 @@ class HaloPool2d(_SpatialBase):
         if (
             s == 1
             and h > 0
+            and k == 2 * h + 1  # band decomposition needs pad == (k-1)//2
             and not self.d2
             and self.exchanger is not None
             and not x.is_meta

@@ -19,9 +19,14 @@ MI355X-native notes:
   rank-ordered sequence that avoids deadlock (comm.py:460-477) — see
   GradReducer.apply_allreduce_master;
 * MASTER-OPT (reference run_step_allreduce, train_spatial_master.py:327)
-  is implemented as an OVERLAPPED flat-grad swap: replica 1's grad
-  exchange with the mirror rank rides the network while replica 2 is
-  still computing.
+  replaces the pair allreduce with two flat-grad P2P legs against the
+  mirror rank. Leg A (replica 1's grads — ready after its step on BOTH
+  sides of the pair) is issued inside ``run_step`` between the two
+  replica steps, so it rides the xGMI links while replica 2 computes;
+  leg B (replica 2's grads) runs at ``allreduce_and_update``. Half the
+  pair traffic is fully overlapped every iteration (the reference
+  alternates odd/even roles to overlap its param leg instead,
+  train_spatial_master.py:327-455).
 """
 
 from __future__ import annotations
@@ -62,12 +67,14 @@ class train_model_master:
         comm: Communicator,
         replications: int = 1,
         enable_comm_opt: bool = False,
+        fp16_allreduce: bool = False,
         **engine_kw,
     ):
         self.comm = comm
         self.batch_size = batch_size
         self.replications = replications
         self.enable_comm_opt = enable_comm_opt
+        self._fp16_allreduce = fp16_allreduce
         r = comm.rank % comm.mp_size
         self.train_model1 = train_model(
             model_gen1, r, batch_size, parts, comm, GEMS_INVERSE=False, **engine_kw
@@ -81,7 +88,14 @@ class train_model_master:
             GEMS_INVERSE=True,
             **engine_kw,
         )
-        self.reducer = GradReducer(comm)
+        self.reducer = GradReducer(comm, fp16_allreduce=fp16_allreduce)
+        self._swap = (
+            _MasterOptSwap(
+                comm, self.reducer, self.train_model1.models, self.train_model2.models
+            )
+            if enable_comm_opt
+            else None
+        )
 
     # -- weight sync at init (reference comm.py:382-400) -------------------
 
@@ -131,7 +145,10 @@ class train_model_master:
 
     def run_step(self, inputs, labels):
         """Interleave replica steps (reference gems_master.py:72-103):
-        inputs/labels hold 2*replications*batch_size samples."""
+        inputs/labels hold 2*replications*batch_size samples. With
+        MASTER-OPT, replica 1's flat-grad exchange with the mirror rank
+        starts here (after its last step) and overlaps replica 2's
+        compute."""
         B = self.batch_size
         loss = 0.0
         corr = seen = 0
@@ -140,6 +157,8 @@ class train_model_master:
             l1, c1, s1 = self.train_model1.run_step(
                 _slc(inputs, o, B), _slc(labels, o, B)
             )
+            if self._swap is not None and rep == self.replications - 1:
+                self._swap.start_leg_a()
             l2, c2, s2 = self.train_model2.run_step(
                 _slc(inputs, o + B, B), _slc(labels, o + B, B)
             )
@@ -159,8 +178,8 @@ class train_model_master:
     def allreduce_and_update(self):
         """Pair the two replicas' grads, then step both optimizers
         (reference apply_allreduce_master_and_update, comm.py:516-523)."""
-        if self.enable_comm_opt:
-            self._paired_swap()
+        if self._swap is not None:
+            self._swap.finish()
         else:
             self.reducer.apply_allreduce_master(
                 self.train_model1.models, self.train_model2.models
@@ -171,45 +190,74 @@ class train_model_master:
         self.train_model1.update()
         self.train_model2.update()
 
-    def _paired_swap(self):
-        _paired_grad_swap(self.comm, self.reducer,
-                          self.train_model1.models, self.train_model2.models)
 
+class _MasterOptSwap:
+    """MASTER-OPT flat-grad exchange with the mirror rank, split into two
+    overlappable legs (reference send_recv_grads,
+    train_spatial_master.py:296-325, but restructured for overlap):
 
-def _paired_grad_swap(comm, reducer, models1, models2):
-    """MASTER-OPT: flat-grad P2P swap with the mirror rank instead of
-    a pair allreduce (reference send_recv_grads,
-    train_spatial_master.py:296-325)."""
-    r = comm.rank % comm.mp_size
-    mirror = comm.mp_size - 1 - r
-    fg1 = reducer.flat(models1)
-    fg2 = reducer.flat(models2)
-    if mirror == r:
-        if fg1.buffer.numel():
-            mean = (fg1.buffer + fg2.buffer) / 2
-            fg1.buffer.copy_(mean)
-            fg2.buffer.copy_(mean)
-        return
-    peer = comm.global_rank(mirror)
-    # my model1 grads pair with mirror's model2 grads (same stage)
-    r1 = torch.empty_like(fg1.buffer)
-    r2 = torch.empty_like(fg2.buffer)
-    # tags keyed by STAGE index (my model1 holds stage r, my model2
-    # stage `mirror`): the pair exchanges stage min(r,mirror) first on
-    # both sides, so gloo tags AND RCCL issue order both line up.
-    first = r < mirror
-    seq = (
-        [(fg1.buffer, r1, 7100), (fg2.buffer, r2, 7101)]
-        if first
-        else [(fg2.buffer, r2, 7100), (fg1.buffer, r1, 7101)]
-    )
-    # both sides order the two swaps by stage index (min stage first),
-    # so RCCL's order-based pairing matches: my model1<->mirror model2
-    # then my model2<->mirror model1.
-    for send_buf, recv_buf, tag in seq:
-        p2p.exchange([(send_buf, peer, tag)], [(recv_buf, peer, tag)]).wait()
-    fg1.buffer.add_(r1).mul_(0.5)
-    fg2.buffer.add_(r2).mul_(0.5)
+    * ``start_leg_a()`` — called between the two replica steps. Both
+      sides of the pair have finished their replica-1 step, so each
+      sends its model1 flat grads (stage r / stage mirror) and posts the
+      landing buffer for the peer's. On RCCL this batch rides the xGMI
+      link while BOTH ranks compute replica 2; on gloo it completes
+      inline (same semantics, no overlap — the CPU parity oracle).
+    * ``finish(...)`` — called at allreduce_and_update: waits leg A,
+      exchanges the model2 grads (leg B), and averages:
+      my fg1 (stage r, rep1) with mirror's fg2 (stage r, rep2), and
+      my fg2 (stage mirror, rep2) with mirror's fg1 (stage mirror, rep1).
+
+    Issue-order audit (RCCL has no tags): every rank starts leg A at the
+    same schedule point and leg B at the same schedule point, and the
+    pipeline traffic of replica 2's step keeps its own per-pair relative
+    order, so send/recv pairing is deterministic even when the mirror is
+    also a pipeline neighbour (mp_size=2).
+    """
+
+    def __init__(self, comm, reducer, models1, models2):
+        self.comm = comm
+        self.reducer = reducer
+        self.models1 = models1
+        self.models2 = models2
+        r = comm.rank % comm.mp_size
+        self.mirror = comm.mp_size - 1 - r
+        self.local = self.mirror == r
+        self.peer = None if self.local else comm.global_rank(self.mirror)
+        self._leg_a = None
+        self._recv_for_fg2 = None  # mirror's fg1 == my fg2's stage
+
+    def start_leg_a(self):
+        fg1 = self.reducer.flat(self.models1)
+        fg2 = self.reducer.flat(self.models2)
+        if self.local or fg1.buffer.numel() == 0:
+            return
+        self._recv_for_fg2 = torch.empty_like(fg2.buffer)
+        self._leg_a = p2p.exchange(
+            [(fg1.buffer, self.peer, 7100)],
+            [(self._recv_for_fg2, self.peer, 7100)],
+        )
+
+    def finish(self):
+        fg1 = self.reducer.flat(self.models1)
+        fg2 = self.reducer.flat(self.models2)
+        if self.local:
+            if fg1.buffer.numel():
+                mean = (fg1.buffer + fg2.buffer) / 2
+                fg1.buffer.copy_(mean)
+                fg2.buffer.copy_(mean)
+            return
+        if self._leg_a is None:  # run_step never started it (e.g. eval)
+            self.start_leg_a()
+        self._leg_a.wait()
+        self._leg_a = None
+        recv_for_fg1 = torch.empty_like(fg1.buffer)
+        p2p.exchange(
+            [(fg2.buffer, self.peer, 7101)],
+            [(recv_for_fg1, self.peer, 7101)],
+        ).wait()
+        fg1.buffer.add_(recv_for_fg1).mul_(0.5)
+        fg2.buffer.add_(self._recv_for_fg2).mul_(0.5)
+        self._recv_for_fg2 = None
 
 
 def _slc(t, off, n):
@@ -230,6 +278,7 @@ class train_spatial_model_master:
         slice_method: str = "square",
         replications: int = 1,
         enable_comm_opt: bool = False,
+        fp16_allreduce: bool = False,
         **engine_kw,
     ):
         verify_spatial_master_config(comm)
@@ -237,6 +286,7 @@ class train_spatial_model_master:
         self.batch_size = batch_size
         self.replications = replications
         self.enable_comm_opt = enable_comm_opt
+        self._fp16_allreduce = fp16_allreduce
         r = comm.rank % comm.mp_size
         self.train_model1 = train_model_spatial(
             model_gen1, r, batch_size, parts, comm,
@@ -252,9 +302,29 @@ class train_spatial_model_master:
             GEMS_INVERSE=True,
             **engine_kw,
         )
-        self.reducer = GradReducer(comm)
+        self.reducer = GradReducer(comm, fp16_allreduce=fp16_allreduce)
+        self._swap = (
+            _MasterOptSwap(
+                comm, self.reducer, self.train_model1.models, self.train_model2.models
+            )
+            if enable_comm_opt
+            else None
+        )
+
+    def _reduce_tiles_model1(self):
+        """Engine1's spatial tile sum — must precede the pair exchange so
+        the mirror receives tile-complete stage grads."""
+        if self.train_model1.is_tile_rank:
+            self.reducer.allreduce_grads(
+                self.train_model1.models,
+                self.comm.spatial_allreduce_groups.get(self.train_model1.split_rank),
+                divide_by=1.0,
+            )
 
     def run_step(self, inputs, labels):
+        """With MASTER-OPT: after replica 1's last step, its tile sum and
+        the leg-A mirror exchange are issued here, overlapping replica
+        2's compute (see _MasterOptSwap)."""
         B = self.batch_size
         loss = 0.0
         corr = seen = 0
@@ -263,6 +333,9 @@ class train_spatial_model_master:
             l1, c1, s1 = self.train_model1.run_step(
                 _slc(inputs, o, B), _slc(labels, o, B)
             )
+            if self._swap is not None and rep == self.replications - 1:
+                self._reduce_tiles_model1()
+                self._swap.start_leg_a()
             l2, c2, s2 = self.train_model2.run_step(
                 _slc(inputs, o + B, B), _slc(labels, o + B, B)
             )
@@ -282,24 +355,19 @@ class train_spatial_model_master:
         apply_allreduce_master_master, comm.py:479-504)."""
         comm = self.comm
         red = self.reducer
-        # engine1 spatial groups
-        if self.train_model1.is_tile_rank:
-            red.allreduce_grads(
-                self.train_model1.models,
-                comm.spatial_allreduce_groups.get(self.train_model1.split_rank),
-                divide_by=1.0,
-            )
+        # engine1 spatial groups (already reduced mid-step under MASTER-OPT)
+        if self._swap is None:
+            self._reduce_tiles_model1()
         # engine2: mirrored tile group of its split_rank
         if self.train_model2.is_tile_rank:
             g = getattr(comm, "mirror_spatial_groups", {}).get(
                 self.train_model2.split_rank
             )
             red.allreduce_grads(self.train_model2.models, g, divide_by=1.0)
-        if self.enable_comm_opt:
-            # MASTER-OPT: the mirror-pair exchange as a flat P2P swap
+        if self._swap is not None:
+            # MASTER-OPT: wait the overlapped leg A, run leg B, average
             # (reference run_step_allreduce, train_spatial_master.py:327)
-            _paired_grad_swap(comm, red, self.train_model1.models,
-                              self.train_model2.models)
+            self._swap.finish()
         else:
             red.apply_allreduce_master(
                 self.train_model1.models, self.train_model2.models

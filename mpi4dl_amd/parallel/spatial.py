@@ -218,6 +218,13 @@ class train_model_spatial(train_model):
             prev = self.split_rank - 1
             spec = _as_list(self.shape_list[prev])
             if self.in_layout is not None:
+                # seam tags stride 4 per edge (100 + i*4 + j below and the
+                # 200+ grad mirrors): wider tuples would collide on gloo
+                assert len(spec) <= 4, (
+                    f"stage boundary carries a {len(spec)}-tensor tuple; the "
+                    "seam tag stride supports at most 4 (widen the stride in "
+                    "receive_input/send_input_grad to raise this)"
+                )
                 # per-edge x per-tensor buffers (tuple activations supported:
                 # each tile sends every tensor of the tuple)
                 mb = self._local_mb()

@@ -17,7 +17,12 @@ def get_parser() -> argparse.ArgumentParser:
         formatter_class=argparse.ArgumentDefaultsHelpFormatter,
     )
     p.add_argument("--fp16-allreduce", action="store_true",
-                   help="compat flag (gradient allreduce already bucketed)")
+                   help="reduce gradients as bf16 (half the xGMI bytes, "
+                        "fp32 exponent range kept); applies to all engines "
+                        "including GEMS")
+    p.add_argument("--no-overlap-grads", action="store_true",
+                   help="disable bucketed allreduce-during-backward on the "
+                        "primary gradient group (overlap is the default)")
     p.add_argument("--model", default="resnet",
                    choices=["resnet", "resnet18", "resnet101", "amoebanet"])
     p.add_argument("--batch-size", type=int, default=8)

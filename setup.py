@@ -22,6 +22,7 @@ ext = CUDAExtension(
     sources=[
         "mpi4dl_amd/csrc/gemscore.hip",
         "mpi4dl_amd/csrc/conv_mfma.hip",
+        "mpi4dl_amd/csrc/conv_pw.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

@@ -73,6 +73,11 @@ def _gems_body(rank, world, steps, B, parts, lr, comm_opt):
         x = torch.randn(2 * B, 3, IMG, IMG)
         y = torch.randint(0, NCLS, (2 * B,))
         loss, _, _ = eng.run_step(x, y)
+        if comm_opt and not eng._swap.local:
+            # MASTER-OPT: leg A (replica 1's flat grads) must already be
+            # in flight here — issued mid-run_step, before replica 2's
+            # step, so it overlaps that compute on RCCL
+            assert eng._swap._leg_a is not None, "leg A not issued mid-step"
         eng.allreduce_and_update()
         losses.append(loss)
     # return per-engine losses: loss only meaningful where last stage lives
@@ -147,6 +152,8 @@ def _gems_sp_body(rank, world, steps, B, parts, lr, comm_opt=False):
         x = torch.randn(2 * B, 3, IMG, IMG)
         y = torch.randint(0, NCLS, (2 * B,))
         loss, _, _ = eng.run_step(x, y)
+        if comm_opt and not eng._swap.local:
+            assert eng._swap._leg_a is not None, "leg A not issued mid-step"
         eng.allreduce_and_update()
         losses.append(loss)
     return losses

@@ -276,7 +276,19 @@ def test_native_conv_autograd_matches_torch():
 @requires_gpu
 def test_act_ckpt_memory_and_parity():
     """--act-ckpt on GPU: same loss/gradients as the stored-activation
-    path, with measurably lower peak activation memory."""
+    path, with measurably lower peak activation memory.
+
+    Grad parity requires DETERMINISTIC kernels: recompute-based
+    checkpointing replays the forward, and any kernel whose output
+    varies between bitwise-identical invocations makes the recomputed
+    activations differ at the ulp level, which chaos-amplifies through
+    deep BN chains into O(1) gradient differences. One MIOpen solver
+    here is exactly such a kernel (bitwise-equal inputs, 7.8e-3 output
+    drift — tools/debug_ckpt_gpu3.py; the gemscore HIP kernels are all
+    deterministic). Pinning torch.backends.cudnn.deterministic makes
+    the whole-model parity EXACT (0/363 params off). Training quality
+    without the flag is unaffected — this is the standard caveat of
+    activation checkpointing over nondeterministic vendor kernels."""
     from mpi4dl_amd.comm import Communicator
     from mpi4dl_amd.models.amoebanet import amoebanetd
     from mpi4dl_amd.parallel.partition import model_generator
@@ -284,6 +296,10 @@ def test_act_ckpt_memory_and_parity():
 
     comm = Communicator(split_size=1)
     dev = torch.device("cuda", 0)
+    prev_det = torch.backends.cudnn.deterministic
+    prev_bench = torch.backends.cudnn.benchmark
+    torch.backends.cudnn.deterministic = True
+    torch.backends.cudnn.benchmark = False
 
     def run(ckpt):
         torch.manual_seed(0)
@@ -308,11 +324,19 @@ def test_act_ckpt_memory_and_parity():
         g = [p.grad.detach().float().clone() for p in gen.models.parameters()]
         return loss, peak, g
 
-    loss_a, peak_a, g_a = run(False)
-    loss_b, peak_b, g_b = run(True)
+    try:
+        loss_a, peak_a, g_a = run(False)
+        loss_b, peak_b, g_b = run(True)
+    finally:
+        torch.backends.cudnn.deterministic = prev_det
+        torch.backends.cudnn.benchmark = prev_bench
     assert abs(loss_a - loss_b) < 1e-3, (loss_a, loss_b)
-    for x, y in zip(g_a, g_b):
-        assert torch.allclose(x, y, rtol=1e-2, atol=1e-3)
+    bad = [
+        i
+        for i, (x, y) in enumerate(zip(g_a, g_b))
+        if not torch.allclose(x, y, rtol=1e-2, atol=1e-3)
+    ]
+    assert not bad, f"{len(bad)} params mismatch, first at {bad[:5]}"
     # recompute must shrink held activations (whole-step peak incl.
     # weights/grads: expect at least ~20% lower)
     assert peak_b < peak_a * 0.8, (peak_a, peak_b)
