@@ -232,16 +232,22 @@ __global__ __launch_bounds__(256) void conv_fwd_kernel(
 }
 
 // ---------------------------------------------------------------------------
-// Forward v2 (stride-1, wide-C shapes): R passes over kernel rows with
-// weights pre-permuted to [K][R][C*S] on the host, so
+// Forward v2 (stride-1 in W): R passes over kernel rows with weights
+// pre-permuted to [K][R][C*S] on the host, so
 //  * the A (weight) tile stays dwordx4-contiguous per pass, and
 //  * the B (pixel) loader loads ONE row segment per (c, px-chunk) and
 //    writes its S shifted copies — S x fewer global loads than v1 and
 //    16-byte vector loads instead of u16 scalars (guide G13).
 // Accumulators persist across the R passes.
+//
+// Round 2: the pixel axis is GLOBAL (n*OH*OW flattened, one 8-px chunk
+// never crosses an output row — OW % 8 == 0), so OW < BN shapes
+// (AmoebaNet's 1x7/7x1 at 64..256 px) fill whole tiles; and the M tile
+// adapts (MFRAG template: BM 32/64/128) so K=26/52/104 layers don't
+// burn 60-80% of the MFMA work on masked rows.
 // ---------------------------------------------------------------------------
 
-template <int SS>
+template <int SS, int MFRAG>
 __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ w2,
     const float* __restrict__ bias, bf16* __restrict__ out, ConvGeom g) {
@@ -251,36 +257,37 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
   // hidden under the previous iteration's MFMAs.
   const int S = SS > 0 ? SS : g.S;
   constexpr int UNITS = (SS == 1) ? 2 : 1;  // S==1: 32 c's -> 512 units
+  constexpr int BMV = 32 * MFRAG;
+  constexpr int ACH = BMV * 4;              // A chunks (8 shorts each)
 
-  const int m_tiles = (g.K + BM - 1) / BM;
+  const int m_tiles = (g.K + BMV - 1) / BMV;
   const int nwg = gridDim.x;
   const int q8 = nwg >> 3, r8 = nwg & 7;
   const int xcd = blockIdx.x & 7, sub = blockIdx.x >> 3;
   int bid = (xcd < r8 ? xcd * (q8 + 1) : r8 * (q8 + 1) + (xcd - r8) * q8) + sub;
   const int mt = bid % m_tiles;
-  bid /= m_tiles;
-  const int owt = bid % g.row_tiles;
-  bid /= g.row_tiles;
-  const int oh = bid % g.OH;
-  const int n = bid / g.OH;
+  const int pt = bid / m_tiles;
+  const int OHW = g.OH * g.OW;
+  const int ptiles_per_img = OHW / BN;
+  const int n = pt / ptiles_per_img;
+  const int q0 = (pt - n * ptiles_per_img) * BN;  // first output pixel
 
-  const int k0out = mt * BM;
-  const int ow0 = owt * BN;
+  const int k0out = mt * BMV;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;
   const int wm = wid >> 1, wn = wid & 1;
 
-  __shared__ __attribute__((aligned(16))) short lds[2 * (BM * (BK + 8)) +
+  __shared__ __attribute__((aligned(16))) short lds[2 * (BMV * (BK + 8)) +
                                                     2 * ((BN / 16) * 520)];
-  auto ldsA = [&](int buf) { return lds + buf * (BM * (BK + 8)); };
+  auto ldsA = [&](int buf) { return lds + buf * (BMV * (BK + 8)); };
   auto ldsB = [&](int buf) {
-    return lds + 2 * (BM * (BK + 8)) + buf * ((BN / 16) * 520);
+    return lds + 2 * (BMV * (BK + 8)) + buf * ((BN / 16) * 520);
   };
 
-  f32x4 acc[4][4];
+  f32x4 acc[MFRAG][4];
 #pragma unroll
-  for (int i = 0; i < 4; ++i)
+  for (int i = 0; i < MFRAG; ++i)
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
@@ -289,19 +296,18 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
   const int CS = g.C * S;
   const int KT = (CS + BK - 1) / BK;
   const int total_it = g.R * KT;
-  const int w_lo = ow0 - g.pw;  // sw == 1
 
   // pipelined register state
   s16x8 aReg[2];
   s16x8 bReg[UNITS][3];
 
-
   auto stage_load = [&](int it) {
     const int r = it / KT;
     const int kk0 = (it % KT) * BK;
 #pragma unroll
-    for (int pass = 0; pass < 2; ++pass) {
+    for (int pass = 0; pass < (ACH + 255) / 256; ++pass) {
       const int idx = pass * 256 + tid;
+      if (ACH < 256 && idx >= ACH) break;
       const int row = idx >> 2;
       const int kc = (idx & 3) * 8;
       const int kout = k0out + row;
@@ -310,8 +316,6 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
                                      ((int64_t)kout * g.R + r) * CS + kk0 + kc);
       }
     }
-    const int ih = oh * g.sh - g.ph + r;
-    const bool row_ok = (ih >= 0 && ih < g.H);
     const int c_lo = kk0 / S;
 #pragma unroll
     for (int u0 = 0; u0 < UNITS; ++u0) {
@@ -319,10 +323,13 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
       const int ci = u >> 4;
       const int pxc = u & 15;
       const int c = c_lo + ci;
-      const int px0 = pxc * 8;
-      const int a0 = w_lo + px0;
+      const int q = q0 + pxc * 8;
+      const int oh = q / g.OW;
+      const int ow00 = q - oh * g.OW;
+      const int a0 = ow00 - g.pw;
       const int a0a = a0 & ~7;  // aligned floor; shift resolved at write
-      if (row_ok && c < g.C && a0a >= 0 &&
+      const int ih = oh * g.sh - g.ph + r;
+      if (ih >= 0 && ih < g.H && c < g.C && a0a >= 0 &&
           a0a + 16 + ((SS > 1) ? 8 : 0) <= g.W) {
         const short* src =
             (const short*)(x + in_n + (int64_t)c * HW + (int64_t)ih * g.W);
@@ -338,8 +345,9 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
     const int r = it / KT;
     const int kk0 = (it % KT) * BK;
 #pragma unroll
-    for (int pass = 0; pass < 2; ++pass) {
+    for (int pass = 0; pass < (ACH + 255) / 256; ++pass) {
       const int idx = pass * 256 + tid;
+      if (ACH < 256 && idx >= ACH) break;
       const int row = idx >> 2;
       const int kc = (idx & 3) * 8;
       const int kout = k0out + row;
@@ -358,8 +366,6 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
       }
       *(s16x8*)(ldsA(buf) + row * (BK + 8) + kc) = *(const s16x8*)v;
     }
-    const int ih = oh * g.sh - g.ph + r;
-    const bool row_ok = (ih >= 0 && ih < g.H);
     const int c_lo = kk0 / S;
 #pragma unroll
     for (int u0 = 0; u0 < UNITS; ++u0) {
@@ -368,13 +374,17 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
       const int pxc = u & 15;
       const int c = c_lo + ci;
       const int px0 = pxc * 8;
-      const int a0 = w_lo + px0;
+      const int q = q0 + px0;
+      const int oh = q / g.OW;
+      const int ow00 = q - oh * g.OW;
+      const int a0 = ow00 - g.pw;
       const int a0a = a0 & ~7;
       const int d = a0 - a0a;
-      const int ow_px0 = ow0 + px0;
+      const int ih = oh * g.sh - g.ph + r;
+      const bool row_ok = (ih >= 0 && ih < g.H);
       const bool fast = row_ok && c < g.C && a0a >= 0 &&
                         a0a + 16 + ((SS > 1) ? 8 : 0) <= g.W;
-      if (fast && ow_px0 + 8 <= g.OW) {
+      if (fast && ow00 + 8 <= g.OW) {
         short raw24[24];
         *(s16x8*)raw24 = bReg[u0][0];
         *(s16x8*)(raw24 + 8) = bReg[u0][1];
@@ -412,7 +422,7 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
           *(s16x8*)(ldsB(buf) + base) = *(const s16x8*)v;
         }
       } else {
-        // edge path: synchronous scalar load+write (rare)
+        // edge path: synchronous scalar load+write (row borders)
         const short* src =
             (const short*)(x + in_n + (int64_t)c * HW + (int64_t)ih * g.W);
         const bool ok = row_ok && c < g.C;
@@ -423,9 +433,7 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
 #pragma unroll
           for (int e = 0; e < 8; ++e) {
             const int col = a0 + ss + e;
-            v[e] = (ok && ow_px0 + e < g.OW && col >= 0 && col < g.W)
-                       ? src[col]
-                       : (short)0;
+            v[e] = (ok && col >= 0 && col < g.W) ? src[col] : (short)0;
           }
           const int base = (px0 >> 4) * 520 + ((kk >> 3) << 7) +
                            (((kk >> 2) & 1) << 6) + ((kk & 3) << 4) +
@@ -436,7 +444,7 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
     }
   };
 
-  const int a_row0 = wm * 64;
+  const int a_row0 = wm * 16 * MFRAG;
   const int b_px0 = wn * 64;
 
   stage_load(0);
@@ -449,7 +457,7 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
     if (it + 1 < total_it) stage_write(it + 1);
     if (it + 2 < total_it) stage_load(it + 2);
 #pragma unroll
-    for (int mf = 0; mf < 4; ++mf) {
+    for (int mf = 0; mf < MFRAG; ++mf) {
       const short* arow =
           ldsA(buf) + (a_row0 + mf * 16 + (lane & 15)) * (BK + 8) +
           ((lane >> 4) << 3);
@@ -476,20 +484,19 @@ __global__ __launch_bounds__(256) void conv_fwd_v2_kernel(
     }
   }
 
-  const int64_t out_n = ((int64_t)n * g.K) * g.OH * g.OW;
+  const int64_t out_n = (int64_t)n * g.K * OHW;
 #pragma unroll
-  for (int mf = 0; mf < 4; ++mf) {
+  for (int mf = 0; mf < MFRAG; ++mf) {
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg) {
       const int kout = k0out + a_row0 + mf * 16 + ((lane >> 4) << 2) + reg;
       if (kout >= g.K) continue;
       const float b = bias ? bias[kout] : 0.f;
-      const int64_t orow =
-          out_n + (int64_t)kout * g.OH * g.OW + (int64_t)oh * g.OW;
+      const int64_t obase = out_n + (int64_t)kout * OHW;
 #pragma unroll
       for (int nf = 0; nf < 4; ++nf) {
-        const int ow = ow0 + b_px0 + nf * 16 + (lane & 15);
-        if (ow < g.OW) out[orow + ow] = (bf16)(acc[mf][nf][reg] + b);
+        const int q = q0 + b_px0 + nf * 16 + (lane & 15);
+        out[obase + q] = (bf16)(acc[mf][nf][reg] + b);
       }
     }
   }
@@ -674,32 +681,46 @@ torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor w,
     b32 = bias->to(torch::kFloat).contiguous();
     bptr = b32.data_ptr<float>();
   }
-  const int m_tiles = (g.K + BM - 1) / BM;
-  const int64_t blocks = (int64_t)m_tiles * g.row_tiles * g.OH * g.N;
-  TORCH_CHECK(blocks < (1LL << 31), "grid too large");
   auto stream = at::cuda::getCurrentCUDAStream();
-  if (g.sh == 1 && g.sw == 1 && g.C * g.S >= BK) {
-    // v2: row-pass kernel with [K][R][C*S]-permuted weights
+  const int64_t OHW = (int64_t)g.OH * g.OW;
+  if (g.sh == 1 && g.sw == 1 && g.C * g.S >= BK && OHW % BN == 0 &&
+      g.OW % 8 == 0) {
+    // v2: row-pass kernel with [K][R][C*S]-permuted weights; global
+    // pixel axis + M-adaptive BM (32/64/128 by K)
     auto w2 = w.view({g.K, g.C, g.R, g.S})
                   .permute({0, 2, 1, 3})
                   .reshape({g.K, g.R, (int64_t)g.C * g.S})
                   .contiguous();
+    const int mfrag = (g.K <= 32) ? 1 : (g.K <= 64) ? 2 : 4;
+    const int m_tiles2 = (g.K + 32 * mfrag - 1) / (32 * mfrag);
+    const int64_t blocks2 = (int64_t)m_tiles2 * (OHW / BN) * g.N;
+    TORCH_CHECK(blocks2 > 0 && blocks2 < (1LL << 31), "grid too large");
     auto launch_v2 = [&](auto* kern) {
-      hipLaunchKernelGGL(kern, dim3((uint32_t)blocks), dim3(256), 0,
+      hipLaunchKernelGGL(kern, dim3((uint32_t)blocks2), dim3(256), 0,
                          stream.stream(), (const bf16*)x.data_ptr(),
                          (const bf16*)w2.data_ptr(), bptr,
                          (bf16*)out.data_ptr(), g);
     };
-    if (g.S == 3)
-      launch_v2(conv_fwd_v2_kernel<3>);
-    else if (g.S == 7)
-      launch_v2(conv_fwd_v2_kernel<7>);
-    else if (g.S == 1)
-      launch_v2(conv_fwd_v2_kernel<1>);
-    else
-      launch_v2(conv_fwd_v2_kernel<0>);
+    switch (g.S * 10 + mfrag) {
+      case 31: launch_v2(conv_fwd_v2_kernel<3, 1>); break;
+      case 32: launch_v2(conv_fwd_v2_kernel<3, 2>); break;
+      case 34: launch_v2(conv_fwd_v2_kernel<3, 4>); break;
+      case 71: launch_v2(conv_fwd_v2_kernel<7, 1>); break;
+      case 72: launch_v2(conv_fwd_v2_kernel<7, 2>); break;
+      case 74: launch_v2(conv_fwd_v2_kernel<7, 4>); break;
+      case 11: launch_v2(conv_fwd_v2_kernel<1, 1>); break;
+      case 12: launch_v2(conv_fwd_v2_kernel<1, 2>); break;
+      case 14: launch_v2(conv_fwd_v2_kernel<1, 4>); break;
+      default:
+        if (mfrag == 1) launch_v2(conv_fwd_v2_kernel<0, 1>);
+        else if (mfrag == 2) launch_v2(conv_fwd_v2_kernel<0, 2>);
+        else launch_v2(conv_fwd_v2_kernel<0, 4>);
+    }
     return out;
   }
+  const int m_tiles = (g.K + BM - 1) / BM;
+  const int64_t blocks = (int64_t)m_tiles * g.row_tiles * g.OH * g.N;
+  TORCH_CHECK(blocks < (1LL << 31), "grid too large");
   hipLaunchKernelGGL(conv_fwd_kernel, dim3((uint32_t)blocks), dim3(256), 0,
                      stream.stream(), (const bf16*)x.data_ptr(),
                      (const bf16*)w.data_ptr(), bptr, (bf16*)out.data_ptr(),

@@ -162,8 +162,8 @@ __global__ __launch_bounds__(256) void pw_kernel(
           for (int e = 0; e < 8; ++e) v[e] = raw[2 * e];
         }
       }
-      const int base = pxc * LDSB_BLK + ((kk >> 3) << 7) +
-                       (((kk >> 2) & 1) << 6) + ((kk & 3) << 4);
+      const int base = (px0 >> 4) * LDSB_BLK + ((kk >> 3) << 7) +
+                       (((kk >> 2) & 1) << 6) + ((kk & 3) << 4) + (px0 & 15);
       *(s16x8*)(ldsB(buf) + base) = *(const s16x8*)v;
     }
   };
@@ -259,19 +259,398 @@ __global__ __launch_bounds__(256) void pw_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// Fat-shape kernel: C >= 256 and K >= 128 make the GEMM compute-bound,
+// so this variant uses the guide's "step 3" GEMM structure (§5 ladder,
+// ~874 TF class at 4096^3): 128x128 tile, BK=64, BOTH operands staged
+// global->LDS with `global_load_lds` dwordx4:
+//  * A (weights) lands lane-linear as [128][64] with the st_16x32 XOR
+//    swizzle applied on the SOURCE address (k-chunk ^= 2 for lanes in
+//    the upper 512B of each 1 KB group) so ds_read_b128 fragment reads
+//    are bank-conflict-free;
+//  * B (pixels) lands in the same tr16 block layout as the skinny
+//    kernel (16 blocks of 512 shorts + 8 pad per buffer).
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void pw_fat_kernel(
+    const bf16* __restrict__ x, const bf16* __restrict__ w,
+    const float* __restrict__ bias, bf16* __restrict__ out, PwGeom g) {
+  constexpr int BM = 128;
+  constexpr int FBK = 64;  // channels per K-step (2 x 32 sub-steps)
+  const int m_tiles = (g.K + BM - 1) / BM;
+  const int nwg = gridDim.x;
+  const int q8 = nwg >> 3, r8 = nwg & 7;
+  const int xcd = blockIdx.x & 7, sub = blockIdx.x >> 3;
+  int bid = (xcd < r8 ? xcd * (q8 + 1) : r8 * (q8 + 1) + (xcd - r8) * q8) + sub;
+  const int mt = bid % m_tiles;
+  const int pt = bid / m_tiles;
+  const int OHW = g.OH * g.OW;
+  const int ptiles_per_img = OHW / PW_BN;
+  const int n = pt / ptiles_per_img;
+  const int q0 = (pt - n * ptiles_per_img) * PW_BN;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = wid >> 1, wn = wid & 1;
+  const int k0out = mt * BM;
+
+  // LDS: A 2 x [128][64] linear (16 KB each), B 2 x 16 blocks x 520
+  __shared__ __attribute__((aligned(16))) short lds[2 * (BM * FBK) +
+                                                    2 * 16 * LDSB_BLK];
+  auto ldsA = [&](int buf) { return lds + buf * (BM * FBK); };
+  auto ldsB = [&](int buf) {
+    return lds + 2 * (BM * FBK) + buf * 16 * LDSB_BLK;
+  };
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  const int64_t in_n = (int64_t)n * g.C * g.HW_in;
+  const int ksteps = (g.C + FBK - 1) / FBK;
+  const int full_ksteps = g.C / FBK;
+
+  int lk, lb;
+  lane_kb(lane, lk, lb);
+  const int64_t bsrc_base = in_n + (int64_t)lk * g.HW_in + q0 + lb * 8;
+
+  // A glds: 16 instructions (4 per wave); instruction i covers rows
+  // [i*8, i*8+8); lane l -> row i*8 + (l>>3), k-chunk (l&7) with the
+  // XOR source swizzle (chunk ^= 2 on the upper half-group).
+  const int a_row_l = (lane >> 3);
+  const int a_chunk = (lane & 7) ^ (((lane >> 5) & 1) << 1);
+
+  auto stage_glds = [&](int buf, int c0) {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int inst = wid * 4 + i;  // 0..15
+      int row = k0out + inst * 8 + a_row_l;
+      if (row >= g.K) row = g.K - 1;  // clamp; masked at epilogue
+      const bf16* srcA = w + (int64_t)row * g.C + c0 + a_chunk * 8;
+      auto* dstA = (__attribute__((address_space(3))) void*)(ldsA(buf) +
+                                                             inst * 512);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)srcA,
+          (__attribute__((address_space(3))) unsigned int*)dstA, 16, 0, 0);
+    }
+#pragma unroll
+    for (int ksub = 0; ksub < 2; ++ksub) {
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        const int pb = wid * 2 + i;
+        const bf16* srcB =
+            x + bsrc_base + (int64_t)(c0 + ksub * 32) * g.HW_in + pb * 16;
+        auto* dstB = (__attribute__((address_space(3))) void*)(
+            ldsB(buf) + (pb * 2 + ksub) * LDSB_BLK);
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)srcB,
+            (__attribute__((address_space(3))) unsigned int*)dstB, 16, 0, 0);
+      }
+    }
+  };
+
+  // register tail path (C % 64): zero-filled, plain stores
+  auto stage_tail = [&](int buf, int c0) {
+    // A: 512 chunks of 8 shorts per 32-sub-step x2 -> 4 per thread
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      const int idx = it * 256 + tid;  // 0..1023 over [128 rows][8 chunks]
+      const int row = idx >> 3;
+      const int ch = idx & 7;
+      int krow = k0out + row;
+      if (krow >= g.K) krow = g.K - 1;
+      short v[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+      const int cbase = c0 + ch * 8;
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        if (cbase + e < g.C)
+          v[e] = ((const short*)w)[(int64_t)krow * g.C + cbase + e];
+      // match the glds image incl. swizzle: within each 8-row group,
+      // rows 4..7 store chunk ch at position ch^2
+      const int grp = row >> 3;
+      const int rl = row & 7;
+      const int sch = (rl >= 4) ? (ch ^ 2) : ch;
+      *(s16x8*)(ldsA(buf) + grp * 512 + rl * 64 + sch * 8) = *(const s16x8*)v;
+    }
+    // B: two 32-sub-steps of 512 chunks -> 4 per thread
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      const int idx = it * 256 + tid;
+      const int ksub = idx >> 9;
+      const int kk = idx & 31;
+      const int pxc = (idx >> 5) & 15;
+      const int px0 = pxc * 8;
+      const int c = c0 + ksub * 32 + kk;
+      short v[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (c < g.C) {
+        const short* src =
+            (const short*)(x + in_n + (int64_t)c * g.HW_in + q0 + px0);
+        *(s16x8*)v = *(const s16x8*)src;
+      }
+      const int base = ((px0 >> 4) * 2 + ksub) * LDSB_BLK + ((kk >> 3) << 7) +
+                       (((kk >> 2) & 1) << 6) + ((kk & 3) << 4) + (px0 & 15);
+      *(s16x8*)(ldsB(buf) + base) = *(const s16x8*)v;
+    }
+  };
+
+  auto stage = [&](int buf, int step) {
+    if (step < full_ksteps)
+      stage_glds(buf, step * FBK);
+    else
+      stage_tail(buf, step * FBK);
+  };
+
+  stage(0, 0);
+
+  const int a_row0 = wm * 64 + (lane & 15);
+  const int a_ch_rd = (lane >> 4);  // 16-B chunk within 32-k sub-step
+
+  for (int step = 0; step < ksteps; ++step) {
+    const int buf = step & 1;
+    __syncthreads();
+    if (step + 1 < ksteps) stage(buf ^ 1, step + 1);
+#pragma unroll
+    for (int ksub = 0; ksub < 2; ++ksub) {
+      s16x8 afrag[4];
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf) {
+        const int row = a_row0 + mf * 16;
+        int ch = ksub * 4 + a_ch_rd;
+        ch ^= ((row >> 2) & 1) << 1;  // st_16x32 read-side swizzle
+        afrag[mf] = *(const s16x8*)(ldsA(buf) + ((row >> 3) * 512) +
+                                    (row & 7) * 64 + ch * 8);
+      }
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        const int pb = wn * 4 + nf;
+        __attribute__((address_space(3))) short* bbase =
+            (__attribute__((address_space(3))) short*)(ldsB(buf)) +
+            (pb * 2 + ksub) * LDSB_BLK + ((lane >> 4) << 7) +
+            ((lane & 15) << 2);
+        s16x4 b0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (__attribute__((address_space(3))) s16x4*)bbase);
+        s16x4 b1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (__attribute__((address_space(3))) s16x4*)(bbase + 64));
+        s16x8 bfrag;
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          bfrag[e] = b0[e];
+          bfrag[e + 4] = b1[e];
+        }
+#pragma unroll
+        for (int mf = 0; mf < 4; ++mf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[mf], bfrag, acc[mf][nf], 0, 0, 0);
+      }
+    }
+  }
+
+  const int64_t out_n = (int64_t)n * g.K * g.oHW;
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int kout = k0out + wm * 64 + mf * 16 + ((lane >> 4) << 2) + reg;
+      if (kout >= g.K) continue;
+      const float bv = bias ? bias[kout] : 0.f;
+      const int64_t obase = out_n + (int64_t)kout * g.oHW;
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        const int q = q0 + wn * 64 + nf * 16 + (lane & 15);
+        out[obase + q] = (bf16)(acc[mf][nf][reg] + bv);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// 1x1 backward-weight: GW[K][C] = sum_p GO[K][p] * X[C][p].
+// Both operands are PIXEL-contiguous rows, so both stage exactly like
+// the fat kernel's A operand (linear [128][64] glds + XOR swizzle) and
+// both fragments are plain ds_read_b128 — MFMA with B read as A-style
+// k-major. Pixel-slab split-K: each block reduces one slab of one
+// image and atomically adds its fp32 128x128 tile once.
+// Replaces conv_bwdw for 1x1 (whose per-output-row grid does
+// m*n*N*OH blocks of 64-px reductions with per-element atomics).
+// ---------------------------------------------------------------------------
+
+struct BwGeom {
+  int Nimg, K, C;
+  int OHW;     // pixels per image
+  int slab;    // pixels per block (divides OHW)
+};
+
+__global__ __launch_bounds__(256) void pw_bwdw_kernel(
+    const bf16* __restrict__ go, const bf16* __restrict__ x,
+    float* __restrict__ gw, BwGeom g) {
+  constexpr int BM = 128, BNW = 128, FBK = 64;
+  const int m_tiles = (g.K + BM - 1) / BM;
+  const int n_tiles = (g.C + BNW - 1) / BNW;
+  int bid = blockIdx.x;
+  const int mt = bid % m_tiles;
+  bid /= m_tiles;
+  const int nt = bid % n_tiles;
+  bid /= n_tiles;
+  const int slabs_per_img = g.OHW / g.slab;
+  const int img = bid / slabs_per_img;
+  const int p0 = (bid - img * slabs_per_img) * g.slab;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = wid >> 1, wn = wid & 1;
+  const int k0out = mt * BM;
+  const int c0out = nt * BNW;
+
+  __shared__ __attribute__((aligned(16))) short lds[4 * (BM * FBK)];
+  auto ldsA = [&](int buf) { return lds + buf * (BM * FBK); };
+  auto ldsB = [&](int buf) { return lds + (2 + buf) * (BM * FBK); };
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  const int64_t go_img = (int64_t)img * g.K * g.OHW;
+  const int64_t x_img = (int64_t)img * g.C * g.OHW;
+  const int a_row_l = (lane >> 3);
+  const int a_chunk = (lane & 7) ^ (((lane >> 5) & 1) << 1);
+
+  auto stage = [&](int buf, int p) {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int inst = wid * 4 + i;
+      int row = k0out + inst * 8 + a_row_l;
+      if (row >= g.K) row = g.K - 1;
+      const bf16* srcA = go + go_img + (int64_t)row * g.OHW + p + a_chunk * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)srcA,
+          (__attribute__((address_space(3))) unsigned int*)(
+              (__attribute__((address_space(3))) void*)(ldsA(buf) +
+                                                        inst * 512)),
+          16, 0, 0);
+      int rowB = c0out + inst * 8 + a_row_l;
+      if (rowB >= g.C) rowB = g.C - 1;
+      const bf16* srcB = x + x_img + (int64_t)rowB * g.OHW + p + a_chunk * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)srcB,
+          (__attribute__((address_space(3))) unsigned int*)(
+              (__attribute__((address_space(3))) void*)(ldsB(buf) +
+                                                        inst * 512)),
+          16, 0, 0);
+    }
+  };
+
+  stage(0, p0);
+  const int ksteps = g.slab / FBK;
+  const int a_row0 = wm * 64 + (lane & 15);
+  const int b_row0 = wn * 64 + (lane & 15);
+  const int ch_rd = (lane >> 4);
+
+  for (int step = 0; step < ksteps; ++step) {
+    const int buf = step & 1;
+    __syncthreads();
+    if (step + 1 < ksteps) stage(buf ^ 1, p0 + (step + 1) * FBK);
+#pragma unroll
+    for (int ksub = 0; ksub < 2; ++ksub) {
+      s16x8 afrag[4], bfrag[4];
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        const int ra = a_row0 + f * 16;
+        int cha = (ksub * 4 + ch_rd) ^ (((ra >> 2) & 1) << 1);
+        afrag[f] = *(const s16x8*)(ldsA(buf) + ((ra >> 3) * 512) +
+                                   (ra & 7) * 64 + cha * 8);
+        const int rb = b_row0 + f * 16;
+        int chb = (ksub * 4 + ch_rd) ^ (((rb >> 2) & 1) << 1);
+        bfrag[f] = *(const s16x8*)(ldsB(buf) + ((rb >> 3) * 512) +
+                                   (rb & 7) * 64 + chb * 8);
+      }
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[mf], bfrag[nf], acc[mf][nf], 0, 0, 0);
+    }
+  }
+
+  // epilogue: fp32 atomic add (slabs/images race on the same tile).
+  // MFMA D layout: row = m-frag row block + (lane>>4)*4 + reg, col = lane&15
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int kout = k0out + wm * 64 + mf * 16 + ((lane >> 4) << 2) + reg;
+      if (kout >= g.K) continue;
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        const int c = c0out + wn * 64 + nf * 16 + (lane & 15);
+        if (c < g.C) atomicAdd(&gw[(int64_t)kout * g.C + c], acc[mf][nf][reg]);
+      }
+    }
+  }
+}
+
+torch::Tensor pw_bwdw(torch::Tensor go, torch::Tensor x) {
+  TORCH_CHECK(go.is_cuda() && go.is_contiguous() && x.is_contiguous());
+  TORCH_CHECK(go.scalar_type() == torch::kBFloat16 &&
+              x.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(go.size(0) == x.size(0) && go.size(2) == x.size(2) &&
+              go.size(3) == x.size(3), "pw_bwdw: stride-1 1x1 only");
+  BwGeom g;
+  g.Nimg = (int)go.size(0);
+  g.K = (int)go.size(1);
+  g.C = (int)x.size(1);
+  g.OHW = (int)(go.size(2) * go.size(3));
+  TORCH_CHECK(g.OHW % 64 == 0, "pw_bwdw: OHW % 64");
+  // slab: fill the chip (>= ~512 blocks) without excess atomic traffic
+  const int mn = ((g.K + 127) / 128) * ((g.C + 127) / 128);
+  int slab = g.OHW;
+  while (slab > 64 && (int64_t)mn * g.Nimg * (g.OHW / slab) < 512 &&
+         slab % 2 == 0 && (slab / 2) % 64 == 0)
+    slab /= 2;
+  auto gw = torch::zeros({(int64_t)g.K, (int64_t)g.C},
+                         x.options().dtype(torch::kFloat));
+  g.slab = slab;
+  const int64_t blocks =
+      (int64_t)mn * g.Nimg * (g.OHW / slab);
+  TORCH_CHECK(blocks > 0 && blocks < (1LL << 31), "pw_bwdw grid");
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(pw_bwdw_kernel, dim3((uint32_t)blocks), dim3(256), 0,
+                     stream.stream(), (const bf16*)go.data_ptr(),
+                     (const bf16*)x.data_ptr(), gw.data_ptr<float>(), g);
+  return gw;
+}
+
+// ---------------------------------------------------------------------------
 // Host wrappers
 // ---------------------------------------------------------------------------
 
 static void launch_pw(const torch::Tensor& x, const torch::Tensor& w,
                       const float* bias, torch::Tensor& out, PwGeom g) {
   const int K = g.K;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  // fat shapes (compute-bound): 128x128xBK64 dual-glds kernel
+  if (g.sw == 1 && g.osw == 1 && g.C >= 256 && K >= 128) {
+    const int m_tiles = (K + 127) / 128;
+    const int64_t blocks =
+        (int64_t)g.N * (g.OH * g.OW / PW_BN) * m_tiles;
+    TORCH_CHECK(blocks > 0 && blocks < (1LL << 31), "pw grid size");
+    hipLaunchKernelGGL(pw_fat_kernel, dim3((uint32_t)blocks), dim3(256), 0,
+                       stream.stream(), (const bf16*)x.data_ptr(),
+                       (const bf16*)w.data_ptr(), bias,
+                       (bf16*)out.data_ptr(), g);
+    return;
+  }
   const int mfrag = (K <= 32) ? 1 : (K <= 64) ? 2 : 4;
   const int BM = 32 * mfrag;
   const int m_tiles = (K + BM - 1) / BM;
   const int64_t ptiles = (int64_t)g.N * (g.OH * g.OW / PW_BN);
   const int64_t blocks = ptiles * m_tiles;
   TORCH_CHECK(blocks > 0 && blocks < (1LL << 31), "pw grid size");
-  auto stream = at::cuda::getCurrentCUDAStream();
   auto launch = [&](auto kern) {
     hipLaunchKernelGGL(kern, dim3((uint32_t)blocks), dim3(256), 0,
                        stream.stream(), (const bf16*)x.data_ptr(),
@@ -360,4 +739,6 @@ void register_conv_pw(pybind11::module_& m) {
         "1x1 conv forward (streaming MFMA GEMM, bf16 NCHW)");
   m.def("pw_bwd_data_strided", &conv_pw::pw_bwd_data_strided,
         "1x1 strided conv backward-data (scatter epilogue)");
+  m.def("pw_bwdw", &conv_pw::pw_bwdw,
+        "1x1 stride-1 conv backward-weight (NT MFMA GEMM, fp32 out)");
 }

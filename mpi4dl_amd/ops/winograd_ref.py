@@ -60,7 +60,8 @@ def filter_transform(w: torch.Tensor) -> torch.Tensor:
     """[K, C, 3, 3] -> U [K, C, 4, 4] = G w G^T (host-side, once per step
     in training; the HIP kernel reads U like it reads permuted weights)."""
     assert w.shape[-2:] == (3, 3), w.shape
-    g = G.to(w.dtype if w.dtype.is_floating_point else torch.float32)
+    g = G.to(device=w.device,
+             dtype=w.dtype if w.dtype.is_floating_point else torch.float32)
     return torch.einsum("ir,kcrs,js->kcij", g, w.float(), g)
 
 

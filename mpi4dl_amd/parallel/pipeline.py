@@ -22,6 +22,7 @@ MI355X-native design:
 from __future__ import annotations
 
 import logging
+import os
 from typing import List, Optional
 
 import torch
@@ -101,6 +102,18 @@ class train_model:
 
         self._init_peers()
         self._init_buffers()
+
+        # side HIP stream for stage sends: the send's RCCL kernel waits
+        # only on an event recorded when the activation is ready, not on
+        # everything queued behind it on the compute stream — so a stage
+        # send overlaps the NEXT micro-batch's compute.
+        self._comm_stream = None
+        if (
+            self.device.type == "cuda"
+            and torch.cuda.is_available()
+            and os.environ.get("MPI4DL_P2P_SIDE_STREAM", "1") != "0"
+        ):
+            self._comm_stream = torch.cuda.Stream(device=self.device)
 
         # per-step state
         self.inputs: List = [None] * parts   # leaf inputs per part
@@ -195,9 +208,23 @@ class train_model:
         leaves = [self._leaf(b) for b in bufs]
         return leaves[0] if len(leaves) == 1 else tuple(leaves)
 
+    def _isend_side(self, ts, peer, tag_base):
+        """Issue an isend from the comm stream, gated on an event that
+        fires when the tensors are ready on the compute stream."""
+        if self._comm_stream is None:
+            return p2p.isend_tensors(ts, peer, tag_base=tag_base)
+        ev = torch.cuda.Event()
+        ev.record()
+        with torch.cuda.stream(self._comm_stream):
+            ev.wait()
+            tr = p2p.isend_tensors(ts, peer, tag_base=tag_base)
+            for t in ts:
+                t.record_stream(self._comm_stream)
+        return tr
+
     def send_output(self, y, part: int):
         ts = [t.to(self.act_dtype) for t in (y if isinstance(y, tuple) else (y,))]
-        tr = p2p.isend_tensors(ts, self.next_rank, tag_base=1000 + part * 16)
+        tr = self._isend_side(ts, self.next_rank, 1000 + part * 16)
         self._pending.append(tr)
 
     def forward_pass(self, data_x, data_y, part: int):
@@ -238,7 +265,7 @@ class train_model:
             (t.grad if t.grad is not None else torch.zeros_like(t)).to(self.act_dtype)
             for t in xs
         ]
-        tr = p2p.isend_tensors(gsend, self.prev_rank, tag_base=3000 + part * 16)
+        tr = self._isend_side(gsend, self.prev_rank, 3000 + part * 16)
         self._pending.append(tr)
 
     def backward_pass(self, part: int):

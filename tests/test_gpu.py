@@ -274,6 +274,48 @@ def test_native_conv_autograd_matches_torch():
 
 @gpu
 @requires_gpu
+@pytest.mark.parametrize(
+    "C,K,HW,k,s,p",
+    [
+        (104, 208, 64, 1, 1, 0),   # pw skinny (K via MFRAG=4)
+        (104, 48, 64, 1, 1, 0),    # pw MFRAG=2
+        (104, 24, 64, 1, 1, 0),    # pw MFRAG=1
+        (328, 208, 64, 1, 1, 0),   # pw C tail (328 % 32 = 8)
+        (104, 104, 64, 1, 2, 0),   # pw stride-2 + scatter bwd-data
+        (288, 160, 64, 1, 1, 0),   # pw_fat (C>=256, K>=128)
+        (292, 133, 64, 1, 1, 0),   # pw_fat with C and K tails
+        (16, 32, 64, 3, 2, 1),     # zero-stuff native s2 bwd-data
+        (16, 32, 62, 3, 2, 1),     # odd output-padding case
+    ],
+)
+def test_native_conv_round2_paths(C, K, HW, k, s, p):
+    """Round-2 kernels: conv_pw (skinny/fat/stride-2) and the
+    zero-stuffed stride-2 backward-data — full autograd vs fp32 torch."""
+    from mpi4dl_amd.ops.conv_native import native_conv2d
+
+    torch.manual_seed(0)
+    x = torch.randn(2, C, HW, HW, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    w = (torch.randn(K, C, k, k, device="cuda") * 0.1).requires_grad_(True)
+    y = native_conv2d(x, w, None, (s, s), (p, p))
+    x2 = x.detach().float().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    ref = torch.nn.functional.conv2d(x2, w2, None, stride=s, padding=p)
+    rel = (y.float() - ref).abs().max() / ref.abs().max()
+    assert rel < 0.05, float(rel)
+    g = torch.randn_like(ref)
+    y.backward(g.to(torch.bfloat16))
+    ref.backward(g)
+    for a, bb, name in [
+        (x.grad.float(), x2.grad, "gx"),
+        (w.grad, w2.grad, "gw"),
+    ]:
+        rel = (a - bb).abs().max() / max(bb.abs().max().item(), 1e-3)
+        assert rel < 0.08, (name, float(rel))
+
+
+@gpu
+@requires_gpu
 def test_act_ckpt_memory_and_parity():
     """--act-ckpt on GPU: same loss/gradients as the stored-activation
     path, with measurably lower peak activation memory.
