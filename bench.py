@@ -122,6 +122,11 @@ def main():
                          "primary gradient group (overlap is the default)")
     args = ap.parse_args()
 
+    # multi-rank: arm the P2P watchdog so an ordering bug produces a
+    # diagnosis (pending peer set) instead of a silent driver timeout
+    if int(os.environ.get("WORLD_SIZE", "1")) > 1:
+        os.environ.setdefault("MPI4DL_WAIT_TIMEOUT", "300")
+
     from mpi4dl_amd.comm import Communicator, GradReducer, init_distributed
     from mpi4dl_amd.models.amoebanet import amoebanetd
     from mpi4dl_amd.ops.plan import SpatialPlan

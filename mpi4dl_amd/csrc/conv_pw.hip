@@ -271,6 +271,7 @@ __global__ __launch_bounds__(256) void pw_kernel(
 //    kernel (16 blocks of 512 shorts + 8 pad per buffer).
 // ---------------------------------------------------------------------------
 
+template <int BUFS>
 __global__ __launch_bounds__(256) void pw_fat_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ w,
     const float* __restrict__ bias, bf16* __restrict__ out, PwGeom g) {
@@ -294,12 +295,12 @@ __global__ __launch_bounds__(256) void pw_fat_kernel(
   const int wm = wid >> 1, wn = wid & 1;
   const int k0out = mt * BM;
 
-  // LDS: A 2 x [128][64] linear (16 KB each), B 2 x 16 blocks x 520
-  __shared__ __attribute__((aligned(16))) short lds[2 * (BM * FBK) +
-                                                    2 * 16 * LDSB_BLK];
+  // LDS: A BUFS x [128][64] linear (16 KB each), B BUFS x 16 blk x 520
+  __shared__ __attribute__((aligned(16))) short lds[BUFS * (BM * FBK) +
+                                                    BUFS * 16 * LDSB_BLK];
   auto ldsA = [&](int buf) { return lds + buf * (BM * FBK); };
   auto ldsB = [&](int buf) {
-    return lds + 2 * (BM * FBK) + buf * 16 * LDSB_BLK;
+    return lds + BUFS * (BM * FBK) + buf * 16 * LDSB_BLK;
   };
 
   f32x4 acc[4][4];
@@ -402,15 +403,42 @@ __global__ __launch_bounds__(256) void pw_fat_kernel(
       stage_tail(buf, step * FBK);
   };
 
-  stage(0, 0);
-
   const int a_row0 = wm * 64 + (lane & 15);
   const int a_ch_rd = (lane >> 4);  // 16-B chunk within 32-k sub-step
 
+  // BUFS == 2: plain double-buffer (glds flies during compute, drained
+  // by the vmcnt(0) inside __syncthreads).
+  // BUFS == 3: the guide's span pipeline — one whole K-step stays in
+  // flight ACROSS the raw barrier. Each wave issues 8 glds per step.
+  // Ordering per iteration: wait MY glds for this step (counted vmcnt,
+  // leaving the next step's in flight) -> raw barrier (now EVERY
+  // wave's loads for this buffer have landed) -> issue step+2's glds
+  // into the buffer all waves just stopped reading -> compute.
+  if (BUFS == 3) {
+    stage(0, 0);
+    if (1 < full_ksteps) stage(1, 1);
+  } else {
+    stage(0, 0);
+  }
+
   for (int step = 0; step < ksteps; ++step) {
-    const int buf = step & 1;
-    __syncthreads();
-    if (step + 1 < ksteps) stage(buf ^ 1, step + 1);
+    const int buf = (BUFS == 3) ? (step % 3) : (step & 1);
+    if (BUFS == 3 && step < full_ksteps) {
+      if (step + 1 < full_ksteps)
+        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      if (step + 2 < full_ksteps) stage((step + 2) % 3, step + 2);
+    } else if (BUFS == 3) {
+      // tail step (C % 64): plain fences around the register staging
+      __syncthreads();
+      stage(buf, step);
+      __syncthreads();
+    } else {
+      __syncthreads();
+      if (step + 1 < ksteps) stage(buf ^ 1, step + 1);
+    }
 #pragma unroll
     for (int ksub = 0; ksub < 2; ++ksub) {
       s16x8 afrag[4];
@@ -482,6 +510,7 @@ struct BwGeom {
   int slab;    // pixels per block (divides OHW)
 };
 
+template <int BUFS>
 __global__ __launch_bounds__(256) void pw_bwdw_kernel(
     const bf16* __restrict__ go, const bf16* __restrict__ x,
     float* __restrict__ gw, BwGeom g) {
@@ -504,9 +533,9 @@ __global__ __launch_bounds__(256) void pw_bwdw_kernel(
   const int k0out = mt * BM;
   const int c0out = nt * BNW;
 
-  __shared__ __attribute__((aligned(16))) short lds[4 * (BM * FBK)];
+  __shared__ __attribute__((aligned(16))) short lds[2 * BUFS * (BM * FBK)];
   auto ldsA = [&](int buf) { return lds + buf * (BM * FBK); };
-  auto ldsB = [&](int buf) { return lds + (2 + buf) * (BM * FBK); };
+  auto ldsB = [&](int buf) { return lds + (BUFS + buf) * (BM * FBK); };
 
   f32x4 acc[4][4];
 #pragma unroll
@@ -544,16 +573,26 @@ __global__ __launch_bounds__(256) void pw_bwdw_kernel(
     }
   };
 
-  stage(0, p0);
   const int ksteps = g.slab / FBK;
+  stage(0, p0);
+  if (BUFS == 3 && 1 < ksteps) stage(1, p0 + FBK);
   const int a_row0 = wm * 64 + (lane & 15);
   const int b_row0 = wn * 64 + (lane & 15);
   const int ch_rd = (lane >> 4);
 
   for (int step = 0; step < ksteps; ++step) {
-    const int buf = step & 1;
-    __syncthreads();
-    if (step + 1 < ksteps) stage(buf ^ 1, p0 + (step + 1) * FBK);
+    const int buf = (BUFS == 3) ? (step % 3) : (step & 1);
+    if (BUFS == 3) {
+      if (step + 1 < ksteps)
+        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      if (step + 2 < ksteps) stage((step + 2) % 3, p0 + (step + 2) * FBK);
+    } else {
+      __syncthreads();
+      if (step + 1 < ksteps) stage(buf ^ 1, p0 + (step + 1) * FBK);
+    }
 #pragma unroll
     for (int ksub = 0; ksub < 2; ++ksub) {
       s16x8 afrag[4], bfrag[4];
@@ -619,9 +658,18 @@ torch::Tensor pw_bwdw(torch::Tensor go, torch::Tensor x) {
       (int64_t)mn * g.Nimg * (g.OHW / slab);
   TORCH_CHECK(blocks > 0 && blocks < (1LL << 31), "pw_bwdw grid");
   auto stream = at::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL(pw_bwdw_kernel, dim3((uint32_t)blocks), dim3(256), 0,
-                     stream.stream(), (const bf16*)go.data_ptr(),
-                     (const bf16*)x.data_ptr(), gw.data_ptr<float>(), g);
+  static const int bufs = []() {
+    const char* e = getenv("MPI4DL_PW_PIPE");
+    return (e && e[0] == '3') ? 3 : 2;
+  }();
+  if (bufs == 3)
+    hipLaunchKernelGGL(pw_bwdw_kernel<3>, dim3((uint32_t)blocks), dim3(256),
+                       0, stream.stream(), (const bf16*)go.data_ptr(),
+                       (const bf16*)x.data_ptr(), gw.data_ptr<float>(), g);
+  else
+    hipLaunchKernelGGL(pw_bwdw_kernel<2>, dim3((uint32_t)blocks), dim3(256),
+                       0, stream.stream(), (const bf16*)go.data_ptr(),
+                       (const bf16*)x.data_ptr(), gw.data_ptr<float>(), g);
   return gw;
 }
 
@@ -639,10 +687,23 @@ static void launch_pw(const torch::Tensor& x, const torch::Tensor& w,
     const int64_t blocks =
         (int64_t)g.N * (g.OH * g.OW / PW_BN) * m_tiles;
     TORCH_CHECK(blocks > 0 && blocks < (1LL << 31), "pw grid size");
-    hipLaunchKernelGGL(pw_fat_kernel, dim3((uint32_t)blocks), dim3(256), 0,
-                       stream.stream(), (const bf16*)x.data_ptr(),
-                       (const bf16*)w.data_ptr(), bias,
-                       (bf16*)out.data_ptr(), g);
+    // measured (gpurun_out/pipe_ab.log): 2-buffer + 2 blocks/CU beats
+    // the 3-buffer span pipeline ~2x here (the span lever only pays in
+    // the 1-block/CU VGPR~250 regime — guide §5 glds table)
+    static const int bufs = []() {
+      const char* e = getenv("MPI4DL_PW_PIPE");
+      return (e && e[0] == '3') ? 3 : 2;
+    }();
+    if (bufs == 3)
+      hipLaunchKernelGGL(pw_fat_kernel<3>, dim3((uint32_t)blocks), dim3(256),
+                         0, stream.stream(), (const bf16*)x.data_ptr(),
+                         (const bf16*)w.data_ptr(), bias,
+                         (bf16*)out.data_ptr(), g);
+    else
+      hipLaunchKernelGGL(pw_fat_kernel<2>, dim3((uint32_t)blocks), dim3(256),
+                         0, stream.stream(), (const bf16*)x.data_ptr(),
+                         (const bf16*)w.data_ptr(), bias,
+                         (bf16*)out.data_ptr(), g);
     return;
   }
   const int mfrag = (K <= 32) ? 1 : (K <= 64) ? 2 : 4;

@@ -220,7 +220,7 @@ torch::Tensor bn_stats(torch::Tensor x) {
 // into a register sliding window (guide G13 — vectorize ALWAYS).
 template <typename T, int KK, int SS>
 __global__ void maxpool_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
-                                   int32_t* __restrict__ idx, int64_t NC,
+                                   uint8_t* __restrict__ idx, int64_t NC,
                                    int H, int W, int OH, int OW, int k_, int s_,
                                    int p) {
   const int k = KK > 0 ? KK : k_;
@@ -237,7 +237,7 @@ __global__ void maxpool_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
     const int ow0 = owc * 8;
     const int h0 = oh * s - p;
     float best[8];
-    int32_t bidx[8];
+    uint8_t bidx[8];
 #pragma unroll
     for (int e = 0; e < 8; ++e) { best[e] = -INFINITY; bidx[e] = 0; }
     const int w_lo = ow0 * s - p;
@@ -261,7 +261,7 @@ __global__ void maxpool_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
             const float v = seg[e * SS + kw];
             if (v > best[e]) {
               best[e] = v;
-              bidx[e] = h * W + w_lo + e * SS + kw;
+              bidx[e] = (uint8_t)(kh * KK + kw);
             }
           }
         }
@@ -273,7 +273,7 @@ __global__ void maxpool_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
             const int w = wbase + kw;
             if (w < 0 || w >= W) continue;
             const float v = (float)row[w];
-            if (v > best[e]) { best[e] = v; bidx[e] = h * W + w; }
+            if (v > best[e]) { best[e] = v; bidx[e] = (uint8_t)(kh * k + kw); }
           }
         }
       }
@@ -284,24 +284,6 @@ __global__ void maxpool_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
       const int ow = ow0 + e;
       if (ow < OW) { y[obase + ow] = (T)best[e]; idx[obase + ow] = bidx[e]; }
     }
-  }
-}
-
-// scatter formulation: each output atomically adds its grad to the
-// argmax input (fp32 scratch; one cast pass). Traffic: ~10B/element vs
-// the gather's (k/s+1)^2 * 6B — ~3x less at stride 1.
-template <typename T>
-__global__ void maxpool_bwd_scatter_kernel(const T* __restrict__ go,
-                                           const int32_t* __restrict__ idx,
-                                           float* __restrict__ gi32,
-                                           int64_t NC, int64_t HW,
-                                           int64_t OHW) {
-  const int64_t total = NC * OHW;
-  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += stride) {
-    const int64_t plane = i / OHW;
-    atomicAdd(&gi32[plane * HW + idx[i]], (float)go[i]);
   }
 }
 
@@ -321,7 +303,7 @@ __global__ void cast_f32_kernel(const float* __restrict__ src,
 // of the 8 inputs overlap, so go/idx rows are read once per (oh, thread).
 template <typename T>
 __global__ void maxpool_bwd_kernel(const T* __restrict__ go,
-                                   const int32_t* __restrict__ idx,
+                                   const uint8_t* __restrict__ idx,
                                    T* __restrict__ gi, int64_t NC, int H,
                                    int W, int OH, int OW, int k, int s,
                                    int p) {
@@ -342,12 +324,14 @@ __global__ void maxpool_bwd_kernel(const T* __restrict__ go,
     int ow_lo = (w0 + p - k + 1 + s - 1) / s; if (ow_lo < 0) ow_lo = 0;
     int ow_hi = (w0 + 7 + p) / s; if (ow_hi > OW - 1) ow_hi = OW - 1;
     const T* gop = go + plane * OH * OW;
-    const int32_t* ip = idx + plane * OH * OW;
+    const uint8_t* ip = idx + plane * OH * OW;
     for (int oh = oh_lo; oh <= oh_hi; ++oh) {
       for (int ow = ow_lo; ow <= ow_hi; ++ow) {
         const int o = oh * OW + ow;
-        const int32_t winner = ip[o];
-        const int dw = winner - h * W - w0;  // which of my 8 inputs (if any)
+        const int off = ip[o];
+        const int r = oh * s - p + off / k;   // winner's input row/col
+        if (r != h) continue;
+        const int dw = ow * s - p + off % k - w0;
         if (dw >= 0 && dw < 8) acc[dw] += (float)gop[o];
       }
     }
@@ -816,7 +800,7 @@ std::vector<torch::Tensor> maxpool_fwd(torch::Tensor x, int64_t k, int64_t s,
   const int OH = (H + 2 * (int)p - (int)k) / (int)s + 1;
   const int OW = (W + 2 * (int)p - (int)k) / (int)s + 1;
   auto y = torch::empty({N, C, OH, OW}, x.options());
-  auto idx = torch::empty({N, C, OH, OW}, x.options().dtype(torch::kInt));
+  auto idx = torch::empty({N, C, OH, OW}, x.options().dtype(torch::kByte));
   const int64_t total = N * C * (int64_t)OH * OW;
   auto stream = at::cuda::getCurrentCUDAStream();
   AT_DISPATCH_FLOATING_TYPES_AND2(
@@ -825,7 +809,7 @@ std::vector<torch::Tensor> maxpool_fwd(torch::Tensor x, int64_t k, int64_t s,
         auto launch = [&](auto kernel) {
           hipLaunchKernelGGL(kernel, dim3(grid_for(total, 256)), dim3(256), 0,
                              stream.stream(), x.data_ptr<scalar_t>(),
-                             y.data_ptr<scalar_t>(), idx.data_ptr<int32_t>(),
+                             y.data_ptr<scalar_t>(), idx.data_ptr<uint8_t>(),
                              N * C, H, W, OH, OW, (int)k, (int)s, (int)p);
         };
         if (k == 3 && s == 2)
@@ -848,35 +832,13 @@ torch::Tensor maxpool_bwd(torch::Tensor go, torch::Tensor idx, int64_t H,
   auto gi = torch::empty({N, C, H, W}, go.options());
   const int64_t total = N * C * H * W;
   auto stream = at::cuda::getCurrentCUDAStream();
-  // scatter (atomicAdd) formulation measured 3.4x SLOWER than the gather
-  // at stride 1 (neighbouring outputs share an argmax input -> serialized
-  // atomics; profiles/PERF_NOTES.md). Keep the gather; scatter retained
-  // for reference behind this disabled guard.
-  if (false && (H * W) % 8 == 0) {
-    auto gi32 = torch::zeros({N, C, H, W}, go.options().dtype(torch::kFloat));
-    const int64_t ototal = N * C * (int64_t)OH * OW;
-    AT_DISPATCH_FLOATING_TYPES_AND2(
-        at::ScalarType::Half, at::ScalarType::BFloat16, go.scalar_type(),
-        "maxpool_bwd_scatter", [&] {
-          hipLaunchKernelGGL((maxpool_bwd_scatter_kernel<scalar_t>),
-                             dim3(grid_for(ototal, 256)), dim3(256), 0,
-                             stream.stream(), go.data_ptr<scalar_t>(),
-                             idx.data_ptr<int32_t>(), gi32.data_ptr<float>(),
-                             N * C, H * W, (int64_t)OH * OW);
-          hipLaunchKernelGGL((cast_f32_kernel<scalar_t>),
-                             dim3(grid_for(total / 8, 256)), dim3(256), 0,
-                             stream.stream(), gi32.data_ptr<float>(),
-                             gi.data_ptr<scalar_t>(), total / 8);
-        });
-    return gi;
-  }
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::Half, at::ScalarType::BFloat16, go.scalar_type(),
       "maxpool_bwd", [&] {
         hipLaunchKernelGGL(
             (maxpool_bwd_kernel<scalar_t>), dim3(grid_for(total, 256)),
             dim3(256), 0, stream.stream(), go.data_ptr<scalar_t>(),
-            idx.data_ptr<int32_t>(), gi.data_ptr<scalar_t>(), N * C, (int)H,
+            idx.data_ptr<uint8_t>(), gi.data_ptr<scalar_t>(), N * C, (int)H,
             (int)W, OH, OW, (int)k, (int)s, (int)p);
       });
   return gi;

@@ -244,6 +244,11 @@ class Cell(nn.Module):
         self.concat = REDUCTION_CONCAT if reduction else NORMAL_CONCAT
         self.indices = [i for i, _ in ops]
         self.wants_relu = [name.startswith("conv") for _, name in ops]
+        # sum states consumed ONLY by the final concat never materialise:
+        # AddCat writes h1+h2 straight into their channel slice (ops/fuse.py)
+        self._cat_only = frozenset(
+            i for i in self.concat if i >= 2 and i not in set(self.indices)
+        )
         self.operations = nn.ModuleList()
         for i, name in ops:
             stride = 2 if (reduction and i < 2) else 1
@@ -268,13 +273,25 @@ class Cell(nn.Module):
                 relu_cache[idx] = torch.relu(states[idx])
             return relu_cache[idx]
 
+        pending = {}
         for i in range(0, len(self.operations), 2):
             h1 = self.operations[i](get(self.indices[i], self.wants_relu[i]))
             h2 = self.operations[i + 1](
                 get(self.indices[i + 1], self.wants_relu[i + 1])
             )
-            states.append(h1 + h2)
-        return torch.cat([states[i] for i in self.concat], dim=1), skip
+            sidx = 2 + i // 2
+            if sidx in self._cat_only:
+                pending[sidx] = (h1, h2)
+                states.append(None)
+            else:
+                states.append(h1 + h2)
+        from ..ops.fuse import add_cat
+
+        entries = [
+            pending[i] if i in pending else (states[i], None)
+            for i in self.concat
+        ]
+        return add_cat(entries), skip
 
 
 class Classify(nn.Module):
