@@ -493,6 +493,200 @@ __global__ __launch_bounds__(256) void pw_fat_kernel(
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// 256x256-tile variant for the BIGGEST fat shapes (C >= 512, K >= 256,
+// e.g. the concat-input reduce convs, C up to 6656): doubling the tile
+// doubles MFMA work per staged byte (128 vs 64 FLOP/B) and halves the
+// A re-reads. 512 threads = 8 waves (2M x 4N), wave tile 128x64,
+// BK=64, 2-buffer dual glds (A linear + XOR swizzle, B tr16 blocks).
+// LDS 130.6 KB -> 1 block/CU; these shapes' x operand is L2/L3
+// resident, so the next-step prefetch hides the shorter latencies.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(512) void pw_fat256_kernel(
+    const bf16* __restrict__ x, const bf16* __restrict__ w,
+    const float* __restrict__ bias, bf16* __restrict__ out, PwGeom g) {
+  constexpr int BM = 256;
+  constexpr int BN2 = 256;
+  constexpr int FBK = 64;
+  const int m_tiles = (g.K + BM - 1) / BM;
+  const int nwg = gridDim.x;
+  const int q8 = nwg >> 3, r8 = nwg & 7;
+  const int xcd = blockIdx.x & 7, sub = blockIdx.x >> 3;
+  int bid = (xcd < r8 ? xcd * (q8 + 1) : r8 * (q8 + 1) + (xcd - r8) * q8) + sub;
+  const int mt = bid % m_tiles;
+  const int pt = bid / m_tiles;
+  const int OHW = g.OH * g.OW;
+  const int ptiles_per_img = OHW / BN2;
+  const int n = pt / ptiles_per_img;
+  const int q0 = (pt - n * ptiles_per_img) * BN2;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;        // 0..7
+  const int wm = wid >> 2, wn = wid & 3;
+  const int k0out = mt * BM;
+
+  __shared__ __attribute__((aligned(16))) short lds[2 * (BM * FBK) +
+                                                    2 * 32 * LDSB_BLK];
+  auto ldsA = [&](int buf) { return lds + buf * (BM * FBK); };
+  auto ldsB = [&](int buf) {
+    return lds + 2 * (BM * FBK) + buf * 32 * LDSB_BLK;
+  };
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  const int64_t in_n = (int64_t)n * g.C * g.HW_in;
+  const int ksteps = (g.C + FBK - 1) / FBK;
+  const int full_ksteps = g.C / FBK;
+
+  int lk, lb;
+  lane_kb(lane, lk, lb);
+  const int64_t bsrc_base = in_n + (int64_t)lk * g.HW_in + q0 + lb * 8;
+  const int a_row_l = (lane >> 3);
+  const int a_chunk = (lane & 7) ^ (((lane >> 5) & 1) << 1);
+
+  auto stage_glds = [&](int buf, int c0) {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int inst = wid * 4 + i;  // 0..31: A rows inst*8..+8
+      int row = k0out + inst * 8 + a_row_l;
+      if (row >= g.K) row = g.K - 1;
+      const bf16* srcA = w + (int64_t)row * g.C + c0 + a_chunk * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)srcA,
+          (__attribute__((address_space(3))) unsigned int*)(
+              (__attribute__((address_space(3))) void*)(ldsA(buf) +
+                                                        inst * 512)),
+          16, 0, 0);
+      // B: inst -> (pxblk, ksub)
+      const int pxblk = inst >> 1;
+      const int ksub = inst & 1;
+      const bf16* srcB =
+          x + bsrc_base + (int64_t)(c0 + ksub * 32) * g.HW_in + pxblk * 16;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)srcB,
+          (__attribute__((address_space(3))) unsigned int*)(
+              (__attribute__((address_space(3))) void*)(
+                  ldsB(buf) + (pxblk * 2 + ksub) * LDSB_BLK)),
+          16, 0, 0);
+    }
+  };
+
+  auto stage_tail = [&](int buf, int c0) {
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      const int idx = it * 512 + tid;  // 2048 A chunks
+      const int row = idx >> 3;
+      const int ch = idx & 7;
+      int krow = k0out + row;
+      if (krow >= g.K) krow = g.K - 1;
+      short v[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+      const int cbase = c0 + ch * 8;
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        if (cbase + e < g.C)
+          v[e] = ((const short*)w)[(int64_t)krow * g.C + cbase + e];
+      const int grp = row >> 3;
+      const int rl = row & 7;
+      const int sch = (rl >= 4) ? (ch ^ 2) : ch;
+      *(s16x8*)(ldsA(buf) + grp * 512 + rl * 64 + sch * 8) = *(const s16x8*)v;
+    }
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      const int idx = it * 512 + tid;  // 2048 B chunks
+      const int ksub = idx >> 10;
+      const int kk = idx & 31;
+      const int pxc = (idx >> 5) & 31;
+      const int px0 = pxc * 8;
+      const int c = c0 + ksub * 32 + kk;
+      short v[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (c < g.C) {
+        const short* src =
+            (const short*)(x + in_n + (int64_t)c * g.HW_in + q0 + px0);
+        *(s16x8*)v = *(const s16x8*)src;
+      }
+      const int base = ((px0 >> 4) * 2 + ksub) * LDSB_BLK + ((kk >> 3) << 7) +
+                       (((kk >> 2) & 1) << 6) + ((kk & 3) << 4) + (px0 & 15);
+      *(s16x8*)(ldsB(buf) + base) = *(const s16x8*)v;
+    }
+  };
+
+  auto stage = [&](int buf, int step) {
+    if (step < full_ksteps)
+      stage_glds(buf, step * FBK);
+    else
+      stage_tail(buf, step * FBK);
+  };
+
+  stage(0, 0);
+
+  const int a_row0 = wm * 128 + (lane & 15);
+  const int a_ch_rd = (lane >> 4);
+
+  for (int step = 0; step < ksteps; ++step) {
+    const int buf = step & 1;
+    __syncthreads();
+    if (step + 1 < ksteps) stage(buf ^ 1, step + 1);
+#pragma unroll
+    for (int ksub = 0; ksub < 2; ++ksub) {
+      s16x8 afrag[8];
+#pragma unroll
+      for (int mf = 0; mf < 8; ++mf) {
+        const int row = a_row0 + mf * 16;
+        int ch = ksub * 4 + a_ch_rd;
+        ch ^= ((row >> 2) & 1) << 1;
+        afrag[mf] = *(const s16x8*)(ldsA(buf) + ((row >> 3) * 512) +
+                                    (row & 7) * 64 + ch * 8);
+      }
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        const int pb = wn * 4 + nf;
+        __attribute__((address_space(3))) short* bbase =
+            (__attribute__((address_space(3))) short*)(ldsB(buf)) +
+            (pb * 2 + ksub) * LDSB_BLK + ((lane >> 4) << 7) +
+            ((lane & 15) << 2);
+        s16x4 b0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (__attribute__((address_space(3))) s16x4*)bbase);
+        s16x4 b1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+            (__attribute__((address_space(3))) s16x4*)(bbase + 64));
+        s16x8 bfrag;
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          bfrag[e] = b0[e];
+          bfrag[e + 4] = b1[e];
+        }
+#pragma unroll
+        for (int mf = 0; mf < 8; ++mf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[mf], bfrag, acc[mf][nf], 0, 0, 0);
+      }
+    }
+  }
+
+  const int64_t out_n = (int64_t)n * g.K * g.oHW;
+#pragma unroll
+  for (int mf = 0; mf < 8; ++mf) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int kout = k0out + wm * 128 + mf * 16 + ((lane >> 4) << 2) + reg;
+      if (kout >= g.K) continue;
+      const float bv = bias ? bias[kout] : 0.f;
+      const int64_t obase = out_n + (int64_t)kout * g.oHW;
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        const int q = q0 + wn * 64 + nf * 16 + (lane & 15);
+        out[obase + q] = (bf16)(acc[mf][nf][reg] + bv);
+      }
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 // 1x1 backward-weight: GW[K][C] = sum_p GO[K][p] * X[C][p].
 // Both operands are PIXEL-contiguous rows, so both stage exactly like
@@ -681,6 +875,21 @@ static void launch_pw(const torch::Tensor& x, const torch::Tensor& w,
                       const float* bias, torch::Tensor& out, PwGeom g) {
   const int K = g.K;
   auto stream = at::cuda::getCurrentCUDAStream();
+  // biggest fat shapes: 256x256 tile (2x MFMA per staged byte)
+  if (g.sw == 1 && g.osw == 1 && g.C >= 512 && K >= 256 &&
+      (g.OH * g.OW) % 256 == 0 &&
+      [] { const char* e = getenv("MPI4DL_PW_FAT256");
+           return !(e && e[0] == '0'); }()) {
+    const int m_tiles = (K + 255) / 256;
+    const int64_t blocks =
+        (int64_t)g.N * (g.OH * g.OW / 256) * m_tiles;
+    TORCH_CHECK(blocks > 0 && blocks < (1LL << 31), "pw grid size");
+    hipLaunchKernelGGL(pw_fat256_kernel, dim3((uint32_t)blocks), dim3(512),
+                       0, stream.stream(), (const bf16*)x.data_ptr(),
+                       (const bf16*)w.data_ptr(), bias,
+                       (bf16*)out.data_ptr(), g);
+    return;
+  }
   // fat shapes (compute-bound): 128x128xBK64 dual-glds kernel
   if (g.sw == 1 && g.osw == 1 && g.C >= 256 && K >= 128) {
     const int m_tiles = (K + 127) / 128;
