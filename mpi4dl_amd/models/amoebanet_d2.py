@@ -184,11 +184,23 @@ class CellD2(nn.Module):
                 t = et
             return t
 
+        used = set(self.indices)
+        pending = {}
         for i in range(0, len(self.operations), 2):
             h1 = self.operations[i](get(i))
             h2 = self.operations[i + 1](get(i + 1))
-            states.append(h1 + h2)
-        return torch.cat([states[i] for i in self.concat], dim=1), skip
+            sidx = 2 + i // 2
+            if sidx in self.concat and sidx not in used:
+                pending[sidx] = (h1, h2)
+                states.append(None)
+            else:
+                states.append(h1 + h2)
+        from ..ops.fuse import add_cat
+
+        return add_cat([
+            pending[i] if i in pending else (states[i], None)
+            for i in self.concat
+        ]), skip
 
 
 def amoebanetd_d2(

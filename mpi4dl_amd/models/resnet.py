@@ -22,6 +22,8 @@ from typing import Optional
 import torch
 import torch.nn as nn
 
+from ..ops.conv_native import NativeConv2d
+
 
 def conv_bn_act(
     in_ch: int,
@@ -34,7 +36,7 @@ def conv_bn_act(
     # bias=True even under BN: the reference's resnet_layer uses the
     # nn.Conv2d default everywhere (resnet.py:40-46)
     pad = kernel // 2
-    layers = [nn.Conv2d(in_ch, out_ch, kernel, stride=stride, padding=pad)]
+    layers = [NativeConv2d(in_ch, out_ch, kernel, stride=stride, padding=pad)]
     if bn:
         layers.append(nn.BatchNorm2d(out_ch))
     if act:
@@ -76,15 +78,15 @@ class BottleneckV2(nn.Module):
             if preact
             else nn.Identity()
         )
-        self.conv1 = nn.Conv2d(in_ch, mid_ch, 3, stride=stride, padding=1)
+        self.conv1 = NativeConv2d(in_ch, mid_ch, 3, stride=stride, padding=1)
         self.pre2 = nn.Sequential(nn.BatchNorm2d(mid_ch), nn.ReLU(inplace=True))
-        self.conv2 = nn.Conv2d(mid_ch, mid_ch, 3, padding=1)
+        self.conv2 = NativeConv2d(mid_ch, mid_ch, 3, padding=1)
         self.pre3 = nn.Sequential(nn.BatchNorm2d(mid_ch), nn.ReLU(inplace=True))
-        self.conv3 = nn.Conv2d(mid_ch, out_ch, 1)
+        self.conv3 = NativeConv2d(mid_ch, out_ch, 1)
         self.proj = None
         if stride != 1 or in_ch != out_ch:
             # reference r4: plain conv on the RAW input (resnet.py:212-219)
-            self.proj = nn.Conv2d(in_ch, out_ch, 1, stride=stride)
+            self.proj = NativeConv2d(in_ch, out_ch, 1, stride=stride)
 
     def forward(self, x):
         y = self.conv1(self.pre1(x))
@@ -121,7 +123,7 @@ def _stem(in_ch: int, filters: int, image_size: int,
     use it when comparing against the reference's published ResNet numbers."""
     if image_size >= 128 and not ref_stem:
         return nn.Sequential(
-            nn.Conv2d(in_ch, filters, 7, stride=2, padding=3, bias=False),
+            NativeConv2d(in_ch, filters, 7, stride=2, padding=3, bias=False),
             nn.BatchNorm2d(filters),
             nn.ReLU(inplace=True),
             nn.MaxPool2d(3, stride=2, padding=1),

@@ -16,6 +16,7 @@ from typing import Optional
 import torch.nn as nn
 
 from ..ops.plan import SpatialPlan
+from ..ops.conv_native import NativeConv2d
 from ..ops.spatial_conv import HaloConv2d, HaloPool2d
 from .resnet import Head  # noqa: F401
 
@@ -51,7 +52,7 @@ class BasicBlockV1S(nn.Module):
         self.proj = None
         if stride != 1 or in_ch != out_ch:
             # 1x1 stride-s conv needs no halo
-            self.proj = nn.Conv2d(in_ch, out_ch, 1, stride=stride, bias=True)
+            self.proj = NativeConv2d(in_ch, out_ch, 1, stride=stride, bias=True)
         self.act = nn.ReLU(inplace=True)
 
     def forward(self, x):
