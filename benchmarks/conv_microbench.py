@@ -163,6 +163,40 @@ def pw_ab(args):
           f"({tot_m/max(tot_n,1e-9):.2f}x)")
 
 
+def bwdw_ab(args):
+    """pw_bwdw vs torch conv2d_weight on the flagship 1x1 shapes."""
+    from mpi4dl_amd.ops import backend
+
+    ge = backend.ext()
+    shapes = model_pw_shapes(mb=args.batch)
+    print(f"{'shape (xN uses)':<40} {'pw_bwdw':>9} {'torch':>9} {'ratio':>6}")
+    tot_n = tot_m = 0.0
+    for (C, K, H, W, s), uses in sorted(shapes.items()):
+        if s != 1:
+            continue
+        x = torch.randn(args.batch, C, H, W, device="cuda", dtype=torch.bfloat16)
+        go = torch.randn(args.batch, K, H, W, device="cuda", dtype=torch.bfloat16)
+        tn = timeit(lambda: ge.pw_bwdw(go, x), args.iters)
+        tm = timeit(
+            lambda: torch.nn.grad.conv2d_weight(
+                x, (K, C, 1, 1), go, stride=1, padding=0
+            ),
+            args.iters,
+        )
+        gw = ge.pw_bwdw(go, x)
+        ref = torch.nn.grad.conv2d_weight(
+            x.float(), (K, C, 1, 1), go.float(), stride=1, padding=0
+        ).view(K, C)
+        rel = (gw - ref).abs().max().item() / max(ref.abs().max().item(), 1e-3)
+        tag = f"C{C}->K{K} {H}x{W} (x{uses})"
+        print(f"{tag:<40} {tn:8.3f}ms {tm:8.3f}ms {tm/tn:5.2f}x  relerr {rel:.3g}",
+              flush=True)
+        tot_n += tn * uses
+        tot_m += tm * uses
+    print(f"\nuse-weighted bwdw total: pw {tot_n:.2f}ms vs torch {tot_m:.2f}ms "
+          f"({tot_m/max(tot_n,1e-9):.2f}x)")
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--iters", type=int, default=20)
@@ -173,11 +207,15 @@ def main():
     ap.add_argument("--pw", action="store_true",
                     help="A/B the conv_pw 1x1 kernel vs MIOpen on the "
                          "flagship model's 1x1 shapes")
+    ap.add_argument("--bwdw", action="store_true",
+                    help="A/B pw_bwdw vs torch conv2d_weight")
     args = ap.parse_args()
     if args.winograd:
         return winograd_ab(args)
     if args.pw:
         return pw_ab(args)
+    if args.bwdw:
+        return bwdw_ab(args)
     from mpi4dl_amd.ops import backend
 
     ge = backend.ext()

@@ -827,6 +827,123 @@ __global__ __launch_bounds__(256) void pw_bwdw_kernel(
   }
 }
 
+
+// 256x256-tile bwdw for big (K, C): 512 threads, 8 waves (2M x 4N),
+// wave tile 128x64; same dual linear-glds staging. 2x MFMA per staged
+// byte vs the 128 tile and 1/4 the (mt, nt) re-reads.
+__global__ __launch_bounds__(512) void pw_bwdw256_kernel(
+    const bf16* __restrict__ go, const bf16* __restrict__ x,
+    float* __restrict__ gw, BwGeom g) {
+  constexpr int BM = 256, BNW = 256, FBK = 64;
+  const int m_tiles = (g.K + BM - 1) / BM;
+  const int n_tiles = (g.C + BNW - 1) / BNW;
+  int bid = blockIdx.x;
+  const int mt = bid % m_tiles;
+  bid /= m_tiles;
+  const int nt = bid % n_tiles;
+  bid /= n_tiles;
+  const int slabs_per_img = g.OHW / g.slab;
+  const int img = bid / slabs_per_img;
+  const int p0 = (bid - img * slabs_per_img) * g.slab;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = wid >> 2, wn = wid & 3;
+  const int k0out = mt * BM;
+  const int c0out = nt * BNW;
+
+  __shared__ __attribute__((aligned(16))) short lds[4 * (BM * FBK)];
+  auto ldsA = [&](int buf) { return lds + buf * (BM * FBK); };
+  auto ldsB = [&](int buf) { return lds + (2 + buf) * (BM * FBK); };
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  const int64_t go_img = (int64_t)img * g.K * g.OHW;
+  const int64_t x_img = (int64_t)img * g.C * g.OHW;
+  const int a_row_l = (lane >> 3);
+  const int a_chunk = (lane & 7) ^ (((lane >> 5) & 1) << 1);
+
+  auto stage = [&](int buf, int p) {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int inst = wid * 4 + i;  // 0..31
+      int row = k0out + inst * 8 + a_row_l;
+      if (row >= g.K) row = g.K - 1;
+      const bf16* srcA = go + go_img + (int64_t)row * g.OHW + p + a_chunk * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)srcA,
+          (__attribute__((address_space(3))) unsigned int*)(
+              (__attribute__((address_space(3))) void*)(ldsA(buf) +
+                                                        inst * 512)),
+          16, 0, 0);
+      int rowB = c0out + inst * 8 + a_row_l;
+      if (rowB >= g.C) rowB = g.C - 1;
+      const bf16* srcB = x + x_img + (int64_t)rowB * g.OHW + p + a_chunk * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)srcB,
+          (__attribute__((address_space(3))) unsigned int*)(
+              (__attribute__((address_space(3))) void*)(ldsB(buf) +
+                                                        inst * 512)),
+          16, 0, 0);
+    }
+  };
+
+  const int ksteps = g.slab / FBK;
+  stage(0, p0);
+  const int a_row0 = wm * 128 + (lane & 15);
+  const int b_row0 = wn * 64 + (lane & 15);
+  const int ch_rd = (lane >> 4);
+
+  for (int step = 0; step < ksteps; ++step) {
+    const int buf = step & 1;
+    __syncthreads();
+    if (step + 1 < ksteps) stage(buf ^ 1, p0 + (step + 1) * FBK);
+#pragma unroll
+    for (int ksub = 0; ksub < 2; ++ksub) {
+      s16x8 afrag[8], bfrag[4];
+#pragma unroll
+      for (int f = 0; f < 8; ++f) {
+        const int ra = a_row0 + f * 16;
+        int cha = (ksub * 4 + ch_rd) ^ (((ra >> 2) & 1) << 1);
+        afrag[f] = *(const s16x8*)(ldsA(buf) + ((ra >> 3) * 512) +
+                                   (ra & 7) * 64 + cha * 8);
+      }
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        const int rb = b_row0 + f * 16;
+        int chb = (ksub * 4 + ch_rd) ^ (((rb >> 2) & 1) << 1);
+        bfrag[f] = *(const s16x8*)(ldsB(buf) + ((rb >> 3) * 512) +
+                                   (rb & 7) * 64 + chb * 8);
+      }
+#pragma unroll
+      for (int mf = 0; mf < 8; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[mf], bfrag[nf], acc[mf][nf], 0, 0, 0);
+    }
+  }
+
+#pragma unroll
+  for (int mf = 0; mf < 8; ++mf) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int kout = k0out + wm * 128 + mf * 16 + ((lane >> 4) << 2) + reg;
+      if (kout >= g.K) continue;
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        const int c = c0out + wn * 64 + nf * 16 + (lane & 15);
+        if (c < g.C) atomicAdd(&gw[(int64_t)kout * g.C + c], acc[mf][nf][reg]);
+      }
+    }
+  }
+}
+
 torch::Tensor pw_bwdw(torch::Tensor go, torch::Tensor x) {
   TORCH_CHECK(go.is_cuda() && go.is_contiguous() && x.is_contiguous());
   TORCH_CHECK(go.scalar_type() == torch::kBFloat16 &&
@@ -839,10 +956,15 @@ torch::Tensor pw_bwdw(torch::Tensor go, torch::Tensor x) {
   g.C = (int)x.size(1);
   g.OHW = (int)(go.size(2) * go.size(3));
   TORCH_CHECK(g.OHW % 64 == 0, "pw_bwdw: OHW % 64");
-  // slab: fill the chip (>= ~512 blocks) without excess atomic traffic
-  const int mn = ((g.K + 127) / 128) * ((g.C + 127) / 128);
+  // 256 tile only when the pixel dim is deep enough that k-steps stay
+  // long after slab splitting (small-P shapes measured 0.5x with deep
+  // splits: 2-kstep blocks are staging-latency bound and the 256x256
+  // fp32 atomic tiles dominate — gpurun_out/bwdw_ab.log)
+  const bool big = (g.K >= 256 && g.C >= 256 && g.OHW >= 16384);
+  const int tile = big ? 256 : 128;
+  const int mn = ((g.K + tile - 1) / tile) * ((g.C + tile - 1) / tile);
   int slab = g.OHW;
-  while (slab > 64 && (int64_t)mn * g.Nimg * (g.OHW / slab) < 512 &&
+  while (slab > 512 && (int64_t)mn * g.Nimg * (g.OHW / slab) < 384 &&
          slab % 2 == 0 && (slab / 2) % 64 == 0)
     slab /= 2;
   auto gw = torch::zeros({(int64_t)g.K, (int64_t)g.C},
@@ -852,6 +974,12 @@ torch::Tensor pw_bwdw(torch::Tensor go, torch::Tensor x) {
       (int64_t)mn * g.Nimg * (g.OHW / slab);
   TORCH_CHECK(blocks > 0 && blocks < (1LL << 31), "pw_bwdw grid");
   auto stream = at::cuda::getCurrentCUDAStream();
+  if (big) {
+    hipLaunchKernelGGL(pw_bwdw256_kernel, dim3((uint32_t)blocks), dim3(512),
+                       0, stream.stream(), (const bf16*)go.data_ptr(),
+                       (const bf16*)x.data_ptr(), gw.data_ptr<float>(), g);
+    return gw;
+  }
   static const int bufs = []() {
     const char* e = getenv("MPI4DL_PW_PIPE");
     return (e && e[0] == '3') ? 3 : 2;

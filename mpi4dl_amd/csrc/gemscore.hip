@@ -301,12 +301,16 @@ __global__ void cast_f32_kernel(const float* __restrict__ src,
 
 // 8 consecutive input columns per thread: the candidate output windows
 // of the 8 inputs overlap, so go/idx rows are read once per (oh, thread).
-template <typename T>
+// (KK, SS) compile-time specialisations fully unroll the window loops
+// and fold the offset div/mod (same trick as the forward).
+template <typename T, int KK, int SS>
 __global__ void maxpool_bwd_kernel(const T* __restrict__ go,
                                    const uint8_t* __restrict__ idx,
                                    T* __restrict__ gi, int64_t NC, int H,
-                                   int W, int OH, int OW, int k, int s,
+                                   int W, int OH, int OW, int k_, int s_,
                                    int p) {
+  const int k = KK > 0 ? KK : k_;
+  const int s = KK > 0 ? SS : s_;
   const int W8 = (W + 7) / 8;
   const int64_t total = NC * H * W8;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -325,14 +329,30 @@ __global__ void maxpool_bwd_kernel(const T* __restrict__ go,
     int ow_hi = (w0 + 7 + p) / s; if (ow_hi > OW - 1) ow_hi = OW - 1;
     const T* gop = go + plane * OH * OW;
     const uint8_t* ip = idx + plane * OH * OW;
+    // at most ceil((8 + k - 1)/s) + 1 candidate columns
+    constexpr int NOW = KK > 0 ? (8 + KK - 1) / SS + 1 : 0;
     for (int oh = oh_lo; oh <= oh_hi; ++oh) {
-      for (int ow = ow_lo; ow <= ow_hi; ++ow) {
-        const int o = oh * OW + ow;
-        const int off = ip[o];
-        const int r = oh * s - p + off / k;   // winner's input row/col
-        if (r != h) continue;
-        const int dw = ow * s - p + off % k - w0;
-        if (dw >= 0 && dw < 8) acc[dw] += (float)gop[o];
+      if (KK > 0) {
+#pragma unroll
+        for (int j = 0; j < NOW; ++j) {
+          const int ow = ow_lo + j;
+          if (ow > ow_hi) break;
+          const int o = oh * OW + ow;
+          const int off = ip[o];
+          const int r = oh * SS - p + off / KK;  // winner's input row/col
+          if (r != h) continue;
+          const int dw = ow * SS - p + off % KK - w0;
+          if (dw >= 0 && dw < 8) acc[dw] += (float)gop[o];
+        }
+      } else {
+        for (int ow = ow_lo; ow <= ow_hi; ++ow) {
+          const int o = oh * OW + ow;
+          const int off = ip[o];
+          const int r = oh * s - p + off / k;
+          if (r != h) continue;
+          const int dw = ow * s - p + off % k - w0;
+          if (dw >= 0 && dw < 8) acc[dw] += (float)gop[o];
+        }
       }
     }
     const int64_t ibase = plane * (int64_t)H * W + (int64_t)h * W;
@@ -835,11 +855,21 @@ torch::Tensor maxpool_bwd(torch::Tensor go, torch::Tensor idx, int64_t H,
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::Half, at::ScalarType::BFloat16, go.scalar_type(),
       "maxpool_bwd", [&] {
-        hipLaunchKernelGGL(
-            (maxpool_bwd_kernel<scalar_t>), dim3(grid_for(total, 256)),
-            dim3(256), 0, stream.stream(), go.data_ptr<scalar_t>(),
-            idx.data_ptr<uint8_t>(), gi.data_ptr<scalar_t>(), N * C, (int)H,
-            (int)W, OH, OW, (int)k, (int)s, (int)p);
+        auto launch = [&](auto kernel) {
+          hipLaunchKernelGGL(kernel, dim3(grid_for(total, 256)), dim3(256), 0,
+                             stream.stream(), go.data_ptr<scalar_t>(),
+                             idx.data_ptr<uint8_t>(), gi.data_ptr<scalar_t>(),
+                             N * C, (int)H, (int)W, OH, OW, (int)k, (int)s,
+                             (int)p);
+        };
+        if (k == 3 && s == 1)
+          launch(maxpool_bwd_kernel<scalar_t, 3, 1>);
+        else if (k == 3 && s == 2)
+          launch(maxpool_bwd_kernel<scalar_t, 3, 2>);
+        else if (k == 2 && s == 2)
+          launch(maxpool_bwd_kernel<scalar_t, 2, 2>);
+        else
+          launch(maxpool_bwd_kernel<scalar_t, 0, 0>);
       });
   return gi;
 }

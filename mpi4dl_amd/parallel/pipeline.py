@@ -122,6 +122,9 @@ class train_model:
         self.correct_sum = 0
         self.seen = 0
         self._pending: List[p2p.Transfer] = []
+        # False -> skip host-side loss/accuracy scalarisation (the
+        # .item() syncs), making run_step hipGraph-capturable
+        self.metrics_enabled = True
 
     # ------------------------------------------------------------------
     # topology
@@ -241,10 +244,11 @@ class train_model:
             yl = data_y.to(self.device, non_blocking=True)
             logits = y[0] if isinstance(y, tuple) else y
             loss = self.criterion(logits.float(), yl)
-            self.loss_sum += float(loss.detach())
-            with torch.no_grad():
-                self.correct_sum += int((logits.argmax(dim=1) == yl).sum())
-                self.seen += yl.numel()
+            if self.metrics_enabled:
+                self.loss_sum += float(loss.detach())
+                with torch.no_grad():
+                    self.correct_sum += int((logits.argmax(dim=1) == yl).sum())
+                    self.seen += yl.numel()
             # keep scaled loss for backward
             self.outputs[part] = loss * (1.0 / self.parts)
         else:
