@@ -1043,6 +1043,9 @@ static void launch_pw(const torch::Tensor& x, const torch::Tensor& w,
                          (bf16*)out.data_ptr(), g);
     return;
   }
+  // measured: MFRAG=8 (BM=256, acc 128 f32/lane) LOSES ~1.4x on the
+  // skinny shapes — register pressure halves occupancy and these
+  // few-kstep kernels live on cross-block latency hiding. Keep BM<=128.
   const int mfrag = (K <= 32) ? 1 : (K <= 64) ? 2 : 4;
   const int BM = 32 * mfrag;
   const int m_tiles = (K + BM - 1) / BM;
@@ -1058,11 +1061,13 @@ static void launch_pw(const torch::Tensor& x, const torch::Tensor& w,
   if (g.sw == 1) {
     if (mfrag == 1) launch(pw_kernel<1, 1>);
     else if (mfrag == 2) launch(pw_kernel<2, 1>);
-    else launch(pw_kernel<4, 1>);
+    else if (mfrag == 4) launch(pw_kernel<4, 1>);
+    else launch(pw_kernel<8, 1>);
   } else {
     if (mfrag == 1) launch(pw_kernel<1, 2>);
     else if (mfrag == 2) launch(pw_kernel<2, 2>);
-    else launch(pw_kernel<4, 2>);
+    else if (mfrag == 4) launch(pw_kernel<4, 2>);
+    else launch(pw_kernel<8, 2>);
   }
 }
 

@@ -467,18 +467,54 @@ __global__ void avgpool_bwd_kernel(const T* __restrict__ go, T* __restrict__ gi,
         ((gr0 + (long)oh_lo * s) >= 0 && (gr0 + (long)oh_hi * s + k) <= Hg &&
          (gc0 + (long)ow_lo * s) >= 0 && (gc0 + (long)ow_hi * s + k) <= Wg);
     const float inv_kk = 1.f / (float)(k * k);
-    for (int oh = oh_lo; oh <= oh_hi; ++oh) {
-      for (int ow = ow_lo; ow <= ow_hi; ++ow) {
-        const float g = (float)gop[oh * OW + ow];
-        const float gd =
-            interior ? g * inv_kk
-                     : g / avg_div(oh, ow, s, k, gr0, gc0, Hg, Wg, include_pad);
-        // which of my 8 inputs does window (oh, ow) cover?
-        const int wlo = ow * s - p;
+    if (KK == 3 && SS == 1) {
+      // stride-1 k3: input w is covered by ow in [w-2+p, w+p]; compute
+      // the <=10 per-row contributions once, then a 3-tap sliding sum
+      // (5x fewer VALU than the predicated 8x scan)
+      for (int oh = oh_lo; oh <= oh_hi; ++oh) {
+        float gd[12];
+#pragma unroll
+        for (int j = 0; j < 10; ++j) {
+          const int ow = ow_lo + j;
+          float v = 0.f;
+          if (ow <= ow_hi) {
+            const float g = (float)gop[oh * OW + ow];
+            v = interior ? g * inv_kk
+                         : g / avg_div(oh, ow, 1, 3, gr0, gc0, Hg, Wg,
+                                       include_pad);
+          }
+          gd[j] = v;
+        }
+        gd[10] = 0.f;
+        gd[11] = 0.f;
 #pragma unroll
         for (int e = 0; e < 8; ++e) {
-          const int w = w0 + e;
-          if (w >= wlo && w < wlo + k && w < W) acc[e] += gd;
+          // ow range covering input w0+e: [w0+e+p-2, w0+e+p]
+          const int j0 = w0 + e + p - 2 - ow_lo;
+          float sum = 0.f;
+#pragma unroll
+          for (int d = 0; d < 3; ++d) {
+            const int j = j0 + d;
+            if (j >= 0 && j < 10) sum += gd[j];
+          }
+          acc[e] += sum;
+        }
+      }
+    } else {
+      for (int oh = oh_lo; oh <= oh_hi; ++oh) {
+        for (int ow = ow_lo; ow <= ow_hi; ++ow) {
+          const float g = (float)gop[oh * OW + ow];
+          const float gd =
+              interior ? g * inv_kk
+                       : g / avg_div(oh, ow, s, k, gr0, gc0, Hg, Wg,
+                                     include_pad);
+          // which of my 8 inputs does window (oh, ow) cover?
+          const int wlo = ow * s - p;
+#pragma unroll
+          for (int e = 0; e < 8; ++e) {
+            const int w = w0 + e;
+            if (w >= wlo && w < wlo + k && w < W) acc[e] += gd;
+          }
         }
       }
     }

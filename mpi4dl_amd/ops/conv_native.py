@@ -163,7 +163,8 @@ def _dispatchable(x, conv: nn.Conv2d) -> bool:
     if x.shape[1] <= 8:
         return True
     one_d = (kh == 1 and kw >= 7) or (kw == 1 and kh >= 7)
-    return one_d and ow >= 384 and x.shape[1] <= 256
+    min_ow = int(os.environ.get("MPI4DL_ONE_D_MIN_OW", "384"))
+    return one_d and ow >= min_ow and x.shape[1] <= 256
 
 
 class NativeConv2d(nn.Conv2d):
