@@ -551,31 +551,38 @@ __global__ __launch_bounds__(512) void pw_fat256_kernel(
   const int a_row_l = (lane >> 3);
   const int a_chunk = (lane & 7) ^ (((lane >> 5) & 1) << 1);
 
+  // one paired (A, B) glds issue — slot i of this wave's 4. Issues are
+  // SPREAD one per MFMA cluster inside the compute loop: 8 back-to-back
+  // LDS-DMA issues per wave cost ~as much as the whole MFMA phase
+  // (guide price list), and the PMC probe showed 50% parked waves with
+  // the batched form.
+  auto stage_one = [&](int buf, int c0, int i) {
+    const int inst = wid * 4 + i;  // 0..31: A rows inst*8..+8
+    int row = k0out + inst * 8 + a_row_l;
+    if (row >= g.K) row = g.K - 1;
+    const bf16* srcA = w + (int64_t)row * g.C + c0 + a_chunk * 8;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)srcA,
+        (__attribute__((address_space(3))) unsigned int*)(
+            (__attribute__((address_space(3))) void*)(ldsA(buf) +
+                                                      inst * 512)),
+        16, 0, 0);
+    // B: inst -> (pxblk, ksub)
+    const int pxblk = inst >> 1;
+    const int ksub = inst & 1;
+    const bf16* srcB =
+        x + bsrc_base + (int64_t)(c0 + ksub * 32) * g.HW_in + pxblk * 16;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)srcB,
+        (__attribute__((address_space(3))) unsigned int*)(
+            (__attribute__((address_space(3))) void*)(
+                ldsB(buf) + (pxblk * 2 + ksub) * LDSB_BLK)),
+        16, 0, 0);
+  };
+
   auto stage_glds = [&](int buf, int c0) {
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      const int inst = wid * 4 + i;  // 0..31: A rows inst*8..+8
-      int row = k0out + inst * 8 + a_row_l;
-      if (row >= g.K) row = g.K - 1;
-      const bf16* srcA = w + (int64_t)row * g.C + c0 + a_chunk * 8;
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) unsigned int*)srcA,
-          (__attribute__((address_space(3))) unsigned int*)(
-              (__attribute__((address_space(3))) void*)(ldsA(buf) +
-                                                        inst * 512)),
-          16, 0, 0);
-      // B: inst -> (pxblk, ksub)
-      const int pxblk = inst >> 1;
-      const int ksub = inst & 1;
-      const bf16* srcB =
-          x + bsrc_base + (int64_t)(c0 + ksub * 32) * g.HW_in + pxblk * 16;
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) unsigned int*)srcB,
-          (__attribute__((address_space(3))) unsigned int*)(
-              (__attribute__((address_space(3))) void*)(
-                  ldsB(buf) + (pxblk * 2 + ksub) * LDSB_BLK)),
-          16, 0, 0);
-    }
+    for (int i = 0; i < 4; ++i) stage_one(buf, c0, i);
   };
 
   auto stage_tail = [&](int buf, int c0) {
@@ -632,6 +639,9 @@ __global__ __launch_bounds__(512) void pw_fat256_kernel(
   for (int step = 0; step < ksteps; ++step) {
     const int buf = step & 1;
     __syncthreads();
+    // measured: spreading the 8 glds issues one-per-MFMA-cluster was
+    // 0-9% SLOWER (the runtime branch inside the unrolled loop defeats
+    // hipcc's scheduling — guide §5.4 trap 4c); batched staging stays
     if (step + 1 < ksteps) stage(buf ^ 1, step + 1);
 #pragma unroll
     for (int ksub = 0; ksub < 2; ++ksub) {
