@@ -1,7 +1,10 @@
 """Fused elementwise glue for the AmoebaNet cell epilogue.
 
-The reference (and round-1) cell does ``states.append(h1 + h2)`` per
-genotype pair and then ``torch.cat([states[i] for i in concat], 1)`` —
+Reference behaviour replaced: the cell's per-pair sums and final
+concat (/root/reference/src/models/amoebanet.py:449-533, Cell.forward's
+``torch.cat(states[...])``). The reference (and round-1) cell does
+``states.append(h1 + h2)`` per genotype pair and then
+``torch.cat([states[i] for i in concat], 1)`` —
 at 2048^2 that cat alone re-reads and re-writes the whole cell output
 (~9% of the step was such eager glue in profiles/r01_*). AddCat writes
 each concat slice ONCE: sum slices compute ``h1 + h2`` directly into
