@@ -561,6 +561,9 @@ class GradReducer:
         group, divide = self.primary_reduce_group()
         if group is None:
             return None
+        prev = getattr(module, "_mpi4dl_overlap", None)
+        if prev is not None and prev["group"] is group:
+            return prev  # hooks already registered for this group
         state = self.prepare_overlap(module, group, divide_by=divide,
                                      bucket_mb=bucket_mb)
         return state

@@ -248,6 +248,8 @@ class _MasterOptSwap:
             return
         if self._leg_a is None:  # run_step never started it (e.g. eval)
             self.start_leg_a()
+        if self._leg_a is None:  # empty partition: nothing to exchange
+            return
         self._leg_a.wait()
         self._leg_a = None
         recv_for_fg1 = torch.empty_like(fg1.buffer)
