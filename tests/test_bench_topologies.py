@@ -79,3 +79,21 @@ def test_bench_n4_1f1b():
     out = run_distributed(_bench_body, 4, (("--schedule", "1f1b"),),
                           timeout=300)
     _check(out, 4, "sp2+pp3")
+
+
+def test_bench_n8_gems():
+    """Config-5 composition (SP+GEMS+PP on 8 ranks): the mp_size ==
+    2*total_tiles boundary case of verify_spatial_master_config, two
+    mirrored spatial engines, MASTER-OPT-capable pairing — the shape
+    the driver would run as `bench.py --gpus 8 --gems`."""
+    out = run_distributed(
+        _bench_body, 8, (("--gems", "--image-size", "128"),), timeout=600
+    )
+    _check(out, 8, "sp4+gems+pp5")
+    assert out[0]["config"]["global_batch"] == 16
+
+
+def test_bench_n4_act_ckpt():
+    out = run_distributed(_bench_body, 4, (("--act-ckpt",),), timeout=400)
+    _check(out, 4, "sp2+pp3")
+    assert out[0]["config"]["act_ckpt"] is True
