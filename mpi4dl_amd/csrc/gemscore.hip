@@ -251,25 +251,15 @@ __global__ void maxpool_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
         // compile-time (k,s): span stays in registers, loops fully unroll
         constexpr int SPAN = KK > 0 ? 8 * SS + KK - SS : 1;
         float seg[SPAN];
-        const int a0a = w_lo & ~7;  // aligned floor of the span
-        if (a0a >= 0 && a0a + ((SPAN + 15) & ~7) <= W) {
-          // interior: vector loads of the whole span (scalar u16 loads
-          // measured ~6x off roofline on the 2x2 pools)
-          constexpr int NV = (SPAN + 14) / 8;  // 8-wide vectors covering span+shift
-          short raw[NV * 8];
+        // NOTE: a vectorized span-load variant (bf16 reinterpret, then a
+        // type-generic memcpy form) was measured and REVERTED: the
+        // reinterpret broke fp32 instantiations, and the memcpy form
+        // regressed the bench ~5% (the local span array drops to
+        // scratch). Scalar loads + L1 hitting the overlapped spans stay.
 #pragma unroll
-          for (int v = 0; v < NV; ++v)
-            *(s16x8_*)(raw + v * 8) = *(const s16x8_*)((const short*)row + a0a + v * 8);
-          const int d = w_lo - a0a;
-#pragma unroll
-          for (int j = 0; j < SPAN; ++j)
-            seg[j] = (float)(*(const __hip_bfloat16*)&raw[d + j]);
-        } else {
-#pragma unroll
-          for (int j = 0; j < SPAN; ++j) {
-            const int w = w_lo + j;
-            seg[j] = (w >= 0 && w < W) ? (float)row[w] : -INFINITY;
-          }
+        for (int j = 0; j < SPAN; ++j) {
+          const int w = w_lo + j;
+          seg[j] = (w >= 0 && w < W) ? (float)row[w] : -INFINITY;
         }
 #pragma unroll
         for (int e = 0; e < 8; ++e) {
