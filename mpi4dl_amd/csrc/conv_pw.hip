@@ -1071,13 +1071,11 @@ static void launch_pw(const torch::Tensor& x, const torch::Tensor& w,
   if (g.sw == 1) {
     if (mfrag == 1) launch(pw_kernel<1, 1>);
     else if (mfrag == 2) launch(pw_kernel<2, 1>);
-    else if (mfrag == 4) launch(pw_kernel<4, 1>);
-    else launch(pw_kernel<8, 1>);
+    else launch(pw_kernel<4, 1>);
   } else {
     if (mfrag == 1) launch(pw_kernel<1, 2>);
     else if (mfrag == 2) launch(pw_kernel<2, 2>);
-    else if (mfrag == 4) launch(pw_kernel<4, 2>);
-    else launch(pw_kernel<8, 2>);
+    else launch(pw_kernel<4, 2>);
   }
 }
 

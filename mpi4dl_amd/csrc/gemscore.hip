@@ -294,18 +294,6 @@ __global__ void maxpool_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
   }
 }
 
-template <typename T>
-__global__ void cast_f32_kernel(const float* __restrict__ src,
-                                T* __restrict__ dst, int64_t n8) {
-  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i8 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i8 < n8;
-       i8 += stride) {
-    const int64_t i = i8 * 8;
-#pragma unroll
-    for (int e = 0; e < 8; ++e) dst[i + e] = (T)src[i + e];
-  }
-}
-
 // 8 consecutive input columns per thread: the candidate output windows
 // of the 8 inputs overlap, so go/idx rows are read once per (oh, thread).
 // (KK, SS) compile-time specialisations fully unroll the window loops
