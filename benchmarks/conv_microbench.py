@@ -197,6 +197,31 @@ def bwdw_ab(args):
           f"({tot_m/max(tot_n,1e-9):.2f}x)")
 
 
+def matmul_ab(args):
+    """Fat 1x1 shapes: pw kernels vs a plain hipBLASLt matmul view
+    (w[K,C] @ x[N,C,HW]) — the 'plain library GEMM' alternative."""
+    from mpi4dl_amd.ops import backend
+
+    ge = backend.ext()
+    shapes = model_pw_shapes(mb=args.batch)
+    print(f"{'shape (xN uses)':<40} {'pw':>9} {'matmul':>9} {'ratio':>6}")
+    for (C, K, H, W, s), uses in sorted(shapes.items()):
+        if s != 1 or C < 256 or K < 128:
+            continue
+        x = torch.randn(args.batch, C, H, W, device="cuda", dtype=torch.bfloat16)
+        w = torch.randn(K, C, device="cuda", dtype=torch.bfloat16) * 0.05
+        w4 = w.view(K, C, 1, 1)
+        xv = x.view(args.batch, C, H * W)
+        tn = timeit(lambda: ge.pw_fwd(x, w4, None, 1, 1), args.iters)
+        tm = timeit(lambda: torch.matmul(w, xv), args.iters)
+        y = ge.pw_fwd(x, w4, None, 1, 1).float()
+        ref = torch.matmul(w.float(), xv.float()).view(args.batch, K, H, W)
+        rel = (y - ref).abs().max().item() / max(ref.abs().max().item(), 1e-3)
+        tag = f"C{C}->K{K} {H}x{W} (x{uses})"
+        print(f"{tag:<40} {tn:8.3f}ms {tm:8.3f}ms {tm/tn:5.2f}x  relerr {rel:.3g}",
+              flush=True)
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--iters", type=int, default=20)
@@ -209,7 +234,11 @@ def main():
                          "flagship model's 1x1 shapes")
     ap.add_argument("--bwdw", action="store_true",
                     help="A/B pw_bwdw vs torch conv2d_weight")
+    ap.add_argument("--matmul", action="store_true",
+                    help="A/B pw fat kernels vs a plain hipBLASLt matmul")
     args = ap.parse_args()
+    if args.matmul:
+        return matmul_ab(args)
     if args.winograd:
         return winograd_ab(args)
     if args.pw:

@@ -55,7 +55,22 @@ class ConvFn(torch.autograd.Function):
             R == 1 and S == 1 and ph == 0 and pw == 0
             and _pw_geom_ok(xb.shape, sh, sw)
         )
-        if is_pw:
+        # fat stride-1 1x1s ARE plain GEMMs; hipBLASLt measured 1.3-1.5x
+        # our fat kernels there (gpurun_out/matmul_ab.log) — the brief's
+        # "libraries for plain library GEMMs" case. Our kernels keep
+        # every shape where THEY win: stride-2 (2-7x), skinny K, and
+        # the bwd-weight NT GEMM (1.5-1.8x over the library path).
+        blas_pw = (
+            is_pw and (sh, sw) == (1, 1) and C >= 256 and K >= 128
+            and os.environ.get("MPI4DL_PW_BLASLT", "1") != "0"
+        )
+        if blas_pw:
+            n, _, h_, w_ = xb.shape
+            y = torch.matmul(wb.view(K, C), xb.view(n, C, h_ * w_))
+            if bias is not None:
+                y += bias.to(y.dtype).view(1, K, 1)
+            y = y.view(n, K, h_, w_)
+        elif is_pw:
             y = ge.pw_fwd(xb, wb, bias, sh, sw)
         else:
             y = ge.conv_fwd(xb, wb, bias, sh, sw, ph, pw)
@@ -90,10 +105,22 @@ class ConvFn(torch.autograd.Function):
             # r2 timed window before this gate)
             gx = None
         elif ctx.is_pw:
-            wt = wb.view(K, C).t().contiguous()
             if (sh, sw) == (1, 1):
-                gx = ge.pw_fwd(go, wt, None, 1, 1)
+                # gx is the mirror GEMM (C<->K): same library-vs-native
+                # crossover as forward
+                if (
+                    K >= 256 and C >= 128
+                    and os.environ.get("MPI4DL_PW_BLASLT", "1") != "0"
+                ):
+                    n, _, h_, w_ = go.shape
+                    gx = torch.matmul(
+                        wb.view(K, C).t(), go.view(n, K, h_ * w_)
+                    ).view(n, C, h_, w_).contiguous()
+                else:
+                    wt = wb.view(K, C).t().contiguous()
+                    gx = ge.pw_fwd(go, wt, None, 1, 1)
             else:
+                wt = wb.view(K, C).t().contiguous()
                 gx = ge.pw_bwd_data_strided(
                     go, wt, xb.shape[-2], xb.shape[-1], sh, sw
                 )

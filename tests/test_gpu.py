@@ -290,11 +290,15 @@ def test_native_conv_autograd_matches_torch():
         (16, 32, 62, 3, 2, 1),     # odd output-padding case
     ],
 )
-def test_native_conv_round2_paths(C, K, HW, k, s, p):
+@pytest.mark.parametrize("blaslt", ["0", "1"])
+def test_native_conv_round2_paths(C, K, HW, k, s, p, blaslt, monkeypatch):
     """Round-2 kernels: conv_pw (skinny/fat/stride-2) and the
-    zero-stuffed stride-2 backward-data — full autograd vs fp32 torch."""
+    zero-stuffed stride-2 backward-data — full autograd vs fp32 torch.
+    blaslt=0 forces the hand-written fat kernels; blaslt=1 covers the
+    measured default (hipBLASLt matmul for fat stride-1 1x1s)."""
     from mpi4dl_amd.ops.conv_native import native_conv2d
 
+    monkeypatch.setenv("MPI4DL_PW_BLASLT", blaslt)
     torch.manual_seed(0)
     x = torch.randn(2, C, HW, HW, device="cuda", dtype=torch.bfloat16,
                     requires_grad=True)
