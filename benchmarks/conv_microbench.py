@@ -204,9 +204,10 @@ def matmul_ab(args):
 
     ge = backend.ext()
     shapes = model_pw_shapes(mb=args.batch)
-    print(f"{'shape (xN uses)':<40} {'pw':>9} {'matmul':>9} {'ratio':>6}")
+    print(f"{'shape (xN uses)':<40} {'pw':>9} {'matmul':>9} {'ratio':>6}"
+          f" | {'bwdw':>9} {'einsum':>9} {'ratio':>6}")
     for (C, K, H, W, s), uses in sorted(shapes.items()):
-        if s != 1 or C < 256 or K < 128:
+        if s != 1:
             continue
         x = torch.randn(args.batch, C, H, W, device="cuda", dtype=torch.bfloat16)
         w = torch.randn(K, C, device="cuda", dtype=torch.bfloat16) * 0.05
@@ -217,8 +218,16 @@ def matmul_ab(args):
         y = ge.pw_fwd(x, w4, None, 1, 1).float()
         ref = torch.matmul(w.float(), xv.float()).view(args.batch, K, H, W)
         rel = (y - ref).abs().max().item() / max(ref.abs().max().item(), 1e-3)
+        # bwd-weight: native NT GEMM vs bmm+sum through hipBLASLt
+        go = torch.randn(args.batch, K, H, W, device="cuda", dtype=torch.bfloat16)
+        gv = go.view(args.batch, K, H * W)
+        tb = timeit(lambda: ge.pw_bwdw(go, x), args.iters)
+        te = timeit(
+            lambda: torch.bmm(gv, xv.transpose(1, 2)).sum(dim=0), args.iters
+        )
         tag = f"C{C}->K{K} {H}x{W} (x{uses})"
-        print(f"{tag:<40} {tn:8.3f}ms {tm:8.3f}ms {tm/tn:5.2f}x  relerr {rel:.3g}",
+        print(f"{tag:<40} {tn:8.3f}ms {tm:8.3f}ms {tm/tn:5.2f}x"
+              f" | {tb:8.3f}ms {te:8.3f}ms {te/tb:5.2f}x  relerr {rel:.3g}",
               flush=True)
 
 

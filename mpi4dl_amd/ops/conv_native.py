@@ -55,13 +55,14 @@ class ConvFn(torch.autograd.Function):
             R == 1 and S == 1 and ph == 0 and pw == 0
             and _pw_geom_ok(xb.shape, sh, sw)
         )
-        # fat stride-1 1x1s ARE plain GEMMs; hipBLASLt measured 1.3-1.5x
-        # our fat kernels there (gpurun_out/matmul_ab.log) — the brief's
+        # Stride-1 1x1s ARE plain GEMMs; hipBLASLt measured 1.05-1.6x
+        # our kernels on nearly every s1 shape EXCEPT K<=64 (where ours
+        # wins 1.1-1.3x) — gpurun_out/matmul_ab2.log. The brief's
         # "libraries for plain library GEMMs" case. Our kernels keep
-        # every shape where THEY win: stride-2 (2-7x), skinny K, and
-        # the bwd-weight NT GEMM (1.5-1.8x over the library path).
+        # every shape where THEY win: stride-2 (2-7x), K<=64, most
+        # bwd-weight shapes, stem/1x7/7x1, pools, BN, halo, SGD.
         blas_pw = (
-            is_pw and (sh, sw) == (1, 1) and C >= 256 and K >= 128
+            is_pw and (sh, sw) == (1, 1) and K > 64
             and os.environ.get("MPI4DL_PW_BLASLT", "1") != "0"
         )
         if blas_pw:
@@ -89,7 +90,22 @@ class ConvFn(torch.autograd.Function):
         go = go.contiguous().to(torch.bfloat16)
         K, C, R, S = wb.shape
         if ctx.is_pw and (sh, sw) == (1, 1):
-            gw = ge.pw_bwdw(go, xb).view(K, C, 1, 1).to(ctx.wdtype)
+            n, _, h_, w_ = xb.shape
+            ohw = h_ * w_
+            # big C*K output tiles over small P: the bmm-and-sum through
+            # hipBLASLt wins (1.3-1.7x); everywhere else our NT GEMM
+            # wins up to 3.5x (gpurun_out/matmul_ab2.log)
+            if (
+                C * K >= 350_000 and ohw <= 32_768
+                and os.environ.get("MPI4DL_PW_BLASLT", "1") != "0"
+            ):
+                gw = (
+                    torch.bmm(go.view(n, K, ohw),
+                              xb.view(n, C, ohw).transpose(1, 2))
+                    .float().sum(dim=0).view(K, C, 1, 1).to(ctx.wdtype)
+                )
+            else:
+                gw = ge.pw_bwdw(go, xb).view(K, C, 1, 1).to(ctx.wdtype)
         elif ctx.is_pw:
             # strided 1x1 (FactorizedReduce): only the strided input
             # pixels contribute — subsample once, then the fast NT GEMM
@@ -107,9 +123,9 @@ class ConvFn(torch.autograd.Function):
         elif ctx.is_pw:
             if (sh, sw) == (1, 1):
                 # gx is the mirror GEMM (C<->K): same library-vs-native
-                # crossover as forward
+                # crossover as forward (output channels = C here)
                 if (
-                    K >= 256 and C >= 128
+                    C > 64
                     and os.environ.get("MPI4DL_PW_BLASLT", "1") != "0"
                 ):
                     n, _, h_, w_ = go.shape
