@@ -337,7 +337,12 @@ __global__ void maxpool_bwd_kernel(const T* __restrict__ go,
           const int r = oh * SS - p + off / KK;  // winner's input row/col
           if (r != h) continue;
           const int dw = ow * SS - p + off % KK - w0;
-          if (dw >= 0 && dw < 8) acc[dw] += (float)gop[o];
+          const float gval = (float)gop[o];
+          // acc[dw] with a RUNTIME index would force acc[] to scratch;
+          // the unrolled compare keeps it in registers
+#pragma unroll
+          for (int e = 0; e < 8; ++e)
+            if (dw == e) acc[e] += gval;
         }
       } else {
         for (int ow = ow_lo; ow <= ow_hi; ++ow) {
@@ -346,7 +351,10 @@ __global__ void maxpool_bwd_kernel(const T* __restrict__ go,
           const int r = oh * s - p + off / k;
           if (r != h) continue;
           const int dw = ow * s - p + off % k - w0;
-          if (dw >= 0 && dw < 8) acc[dw] += (float)gop[o];
+          const float gval = (float)gop[o];
+#pragma unroll
+          for (int e = 0; e < 8; ++e)
+            if (dw == e) acc[e] += gval;
         }
       }
     }
