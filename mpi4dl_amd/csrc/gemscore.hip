@@ -805,17 +805,22 @@ void sgd_momentum(torch::Tensor p, torch::Tensor g, torch::Tensor v,
 // bn_finalize: fold the fp64 [sum,sumsq] -> [mean, invstd] conversion,
 // the running-stat EMA update and num_batches_tracked into ONE kernel
 // (the python glue cost ~8 small launches per BN call).
-__global__ void bn_finalize_kernel(const double* __restrict__ stats,
+__global__ void bn_finalize_kernel(double* __restrict__ stats,
                                    float* __restrict__ out,
                                    float* __restrict__ rmean,
                                    float* __restrict__ rvar,
                                    int64_t* __restrict__ tracked, float mom,
-                                   double n, float eps, int64_t C) {
+                                   double n, float eps, int64_t C,
+                                   int rezero) {
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= C) return;
   const double mean = stats[i] / n;
   double var = stats[C + i] / n - mean * mean;
   if (var < 0.0) var = 0.0;
+  if (rezero) {
+    stats[i] = 0.0;       // restore the zero invariant for the next
+    stats[C + i] = 0.0;   // bn_stats64_acc accumulation (no fill kernel)
+  }
   out[i] = (float)mean;
   out[C + i] = rsqrtf((float)var + eps);
   if (rmean) {
@@ -830,7 +835,7 @@ torch::Tensor bn_finalize(torch::Tensor stats,
                           c10::optional<torch::Tensor> rmean,
                           c10::optional<torch::Tensor> rvar,
                           c10::optional<torch::Tensor> tracked, double mom,
-                          double n, double eps) {
+                          double n, double eps, bool rezero = false) {
   const int64_t C = stats.numel() / 2;
   auto out = torch::empty({2 * C}, stats.options().dtype(torch::kFloat));
   auto stream = at::cuda::getCurrentCUDAStream();
@@ -840,9 +845,10 @@ torch::Tensor bn_finalize(torch::Tensor stats,
       rmean.has_value() ? rmean->data_ptr<float>() : nullptr,
       rvar.has_value() ? rvar->data_ptr<float>() : nullptr,
       tracked.has_value() ? tracked->data_ptr<int64_t>() : nullptr,
-      (float)mom, n, (float)eps, C);
+      (float)mom, n, (float)eps, C, rezero ? 1 : 0);
   return out;
 }
+
 
 // ---------------- torch-facing wrappers ----------------
 
@@ -995,6 +1001,30 @@ torch::Tensor bn_stats64(torch::Tensor x) {
   return out;
 }
 
+// accumulate into a caller-owned ZEROED fp64 buffer (persistent across
+// steps; bn_finalize(..., rezero=true) restores the invariant) — kills
+// the per-BN-call torch.zeros fill kernel (~3k launches/step).
+torch::Tensor bn_stats64_acc(torch::Tensor x, torch::Tensor out) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.is_contiguous());
+  const int64_t N = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
+  TORCH_CHECK(out.is_cuda() && out.is_contiguous() &&
+              out.scalar_type() == torch::kDouble && out.numel() == 2 * C);
+  const int64_t splits = stats_splits(C, HW);
+  int64_t chunk = (HW + splits - 1) / splits;
+  chunk = ((chunk + 2047) / 2048) * 2048;
+  const int64_t gy = (HW + chunk - 1) / chunk;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::Half, at::ScalarType::BFloat16, x.scalar_type(),
+      "bn_stats64_acc", [&] {
+        hipLaunchKernelGGL((bn_stats64_kernel<scalar_t>),
+                           dim3((uint32_t)C, (uint32_t)gy), dim3(256), 0,
+                           stream.stream(), x.data_ptr<scalar_t>(),
+                           out.data_ptr<double>(), N, C, HW, chunk);
+      });
+  return out;
+}
+
 torch::Tensor bn_apply(torch::Tensor x, torch::Tensor mean,
                        torch::Tensor invstd, torch::Tensor w, torch::Tensor b,
                        bool relu) {
@@ -1109,6 +1139,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_bwd_stats", &bn_bwd_stats);
   m.def("bn_bwd_apply", &bn_bwd_apply);
   m.def("sgd_momentum", &sgd_momentum);
-  m.def("bn_finalize", &bn_finalize);
+  m.def("bn_finalize", &bn_finalize, pybind11::arg("stats"),
+        pybind11::arg("rmean"), pybind11::arg("rvar"),
+        pybind11::arg("tracked"), pybind11::arg("mom"), pybind11::arg("n"),
+        pybind11::arg("eps"), pybind11::arg("rezero") = false);
+  m.def("bn_stats64_acc", &bn_stats64_acc);
   m.attr("gfx_arch") = "gfx950";
 }

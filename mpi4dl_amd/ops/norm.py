@@ -108,7 +108,15 @@ class TileBatchNorm2d(nn.BatchNorm2d):
         n = x.numel() // C
         group = self.group if (self.group is not None and dist.is_initialized()) else None
         if self.training:
-            stats = ge.bn_stats64(x)  # fp64 [sum, sumsq]
+            # persistent zeroed fp64 buffer: stats64 accumulates into it
+            # and bn_finalize(rezero=True) restores the zero invariant —
+            # removes the per-call torch.zeros fill (~3k tiny launches
+            # per bench step)
+            buf = getattr(self, "_stats64", None)
+            if buf is None or buf.device != x.device or buf.numel() != 2 * C:
+                buf = torch.zeros(2 * C, device=x.device, dtype=torch.float64)
+                self._stats64 = buf
+            stats = ge.bn_stats64_acc(x, buf)
             if group is not None:
                 dist.all_reduce(stats, group=group)
                 n *= dist.get_world_size(group=group)
@@ -118,7 +126,7 @@ class TileBatchNorm2d(nn.BatchNorm2d):
                 self.running_mean if self.track_running_stats else None,
                 self.running_var if self.track_running_stats else None,
                 self.num_batches_tracked if self.track_running_stats else None,
-                m, float(n), self.eps,
+                m, float(n), self.eps, rezero=True,
             )
             mean, invstd = mv[:C], mv[C:]
         else:
