@@ -441,3 +441,82 @@ def test_fused_sgd_matches_torch():
         assert torch.allclose(p1, p2, rtol=1e-5, atol=1e-6), (
             (p1 - p2).abs().max()
         )
+
+
+@gpu
+@requires_gpu
+def test_bn_persistent_stats_rezero():
+    """bn_stats64_acc accumulates into a persistent buffer whose zero
+    invariant bn_finalize(rezero=True) restores: two identical forward
+    calls must produce identical outputs (a stale buffer would double
+    the sums) and leave the buffer zeroed."""
+    from mpi4dl_amd.ops.norm import TileBatchNorm2d
+
+    torch.manual_seed(0)
+    bn = TileBatchNorm2d(32).cuda().train()
+    x = torch.randn(2, 32, 64, 64, device="cuda", dtype=torch.bfloat16)
+    y1 = bn(x)
+    y2 = bn(x)
+    assert torch.equal(y1, y2)
+    buf = bn._stats64
+    torch.cuda.synchronize()
+    assert float(buf.abs().max()) == 0.0, "zero invariant not restored"
+    # eval path must not touch the buffer
+    bn.eval()
+    bn(x)
+    torch.cuda.synchronize()
+    assert float(buf.abs().max()) == 0.0
+
+
+@gpu
+@requires_gpu
+def test_hipgraph_step_capture():
+    """The engine step is hipGraph-capturable with metrics disabled
+    (the bench's N=1 fast path): capture one step, replay twice, and
+    the weights keep moving by the same update rule as eager."""
+    from mpi4dl_amd.comm import Communicator
+    from mpi4dl_amd.models.amoebanet import amoebanetd
+    from mpi4dl_amd.parallel.partition import model_generator
+    from mpi4dl_amd.parallel.pipeline import train_model
+
+    comm = Communicator(split_size=1)
+    dev = torch.device("cuda", 0)
+    torch.manual_seed(0)
+    model = amoebanetd(10, 3, 64)
+    gen = model_generator(model, 1, input_size=(1, 3, 128, 128))
+    gen.get_output_shapes()
+    gen.ready_model(0, device=dev)
+    opt = torch.optim.SGD(gen.models.parameters(), lr=0.01)
+    eng = train_model(
+        gen, 0, 2, 2, comm, optimizer=opt, device=dev,
+        autocast_dtype=torch.bfloat16, act_dtype=torch.bfloat16,
+    )
+    eng.metrics_enabled = False
+    torch.manual_seed(1)
+    x = torch.randn(2, 3, 128, 128, device=dev)
+    y = torch.randint(0, 10, (2,), device=dev)
+
+    def step():
+        eng.run_step(x, y)
+        eng.update()
+
+    # warmup (find/allocator), then capture
+    side = torch.cuda.Stream()
+    side.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(side):
+        step()
+        step()
+    torch.cuda.current_stream().wait_stream(side)
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        step()
+    p = next(gen.models.parameters())
+    before = p.detach().float().clone()
+    g.replay()
+    torch.cuda.synchronize()
+    mid = p.detach().float().clone()
+    g.replay()
+    torch.cuda.synchronize()
+    after = p.detach().float().clone()
+    assert not torch.equal(before, mid), "replay made no update"
+    assert not torch.equal(mid, after), "second replay made no update"
