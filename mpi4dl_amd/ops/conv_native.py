@@ -1,16 +1,23 @@
-"""NativeConv2d: implicit-GEMM MFMA conv (csrc/conv_mfma.hip) behind an
-nn.Conv2d-compatible module.
+"""NativeConv2d: hand-written gfx950 conv kernels behind an
+nn.Conv2d-compatible module, with a fully MEASURED dispatch policy.
 
-Forward and weight-gradient run on the hand-written gfx950 kernels;
-data-gradient for stride-1 convs is the SAME forward kernel applied to
-the (C<->K transposed, 180-rotated) weights — the classic transposed
-convolution identity — so one kernel covers both. Stride-2 data-grad
-falls back to torch (MIOpen) for now.
+Kernel families (csrc/):
+* conv_mfma.hip — implicit-GEMM MFMA conv (general R x S): forward,
+  weight-grad; stride-1 data-grad = the forward kernel on C<->K
+  transposed / 180-rotated weights; STRIDED data-grad = zero-stuffed
+  transposed conv through the same stride-1 kernel.
+* conv_pw.hip — the 1x1 family: skinny/fat/fat256 streaming GEMMs,
+  stride-2 gather + scatter backward-data, NT-GEMM bwd-weight with
+  pixel-slab split-K.
 
-Dispatch: CUDA + bf16 (or under autocast) + dilation/groups == 1 + OW
-wide enough for the row-segment tiling to be efficient. Everything else
-uses F.conv2d. MPI4DL_NATIVE_CONV=0 disables, =1 forces (for A/B
-benchmarking).
+Per-shape, per-LEG policy (all crossovers measured on MI355X —
+profiles/PERF_NOTES.md): stride-2 1x1s, K<=64 1x1s, stems, wide
+1x7/7x1 and most weight-grads run on the hand-written kernels; the
+fat stride-1 1x1 forward/data-grad legs that are plain library GEMMs
+go to hipBLASLt (the brief's "libraries only for plain GEMMs" case;
+MPI4DL_PW_BLASLT=0 restores all-native); everything else falls to
+F.conv2d/MIOpen. MPI4DL_NATIVE_CONV=0 disables all native paths,
+=1 forces them (A/B benchmarking).
 """
 
 from __future__ import annotations
