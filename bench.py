@@ -56,6 +56,9 @@ sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
 # reference best published number on this metric/config:
 # AmoebaNet-D 2048^2 SP-vert-D2 B=2 ~= 5.0 img/s
+# NOTE: "metric" in the JSON is BASELINE.json's label for the benchmark
+# FAMILY; the ACTUAL topology of a given run is config.parallelism
+# ("single" at N=1 — no SP/PP runs on one GPU).
 BASELINE_IMGS = 5.0
 
 
