@@ -954,6 +954,132 @@ __global__ __launch_bounds__(512) void pw_bwdw256_kernel(
   }
 }
 
+
+// Pipelined bwdw256: BK=32 so THREE (A,B) buffer pairs fit in 96 KB —
+// at this kernel's 1-block/CU occupancy the in-kernel span is the only
+// latency hiding available (the 2-buffer __syncthreads form parks
+// waves on the glds drain; PMC probe in profiles/PERF_NOTES.md).
+// Protocol per step: wait MY 4 glds for this step (counted vmcnt(4) —
+// the 4 newer ones stay in flight) -> raw barrier (now EVERY wave's
+// shares landed) -> issue step+2 into the buffer all waves finished
+// reading two steps ago -> 32 MFMA. [256][32]-short images need no
+// swizzle: 64-byte rows decorrelate the b128 lane-group banks.
+__global__ __launch_bounds__(512) void pw_bwdw256p_kernel(
+    const bf16* __restrict__ go, const bf16* __restrict__ x,
+    float* __restrict__ gw, BwGeom g) {
+  constexpr int BM = 256, BNW = 256, FBK = 32;
+  const int m_tiles = (g.K + BM - 1) / BM;
+  const int n_tiles = (g.C + BNW - 1) / BNW;
+  int bid = blockIdx.x;
+  const int mt = bid % m_tiles;
+  bid /= m_tiles;
+  const int nt = bid % n_tiles;
+  bid /= n_tiles;
+  const int slabs_per_img = g.OHW / g.slab;
+  const int img = bid / slabs_per_img;
+  const int p0 = (bid - img * slabs_per_img) * g.slab;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = wid >> 2, wn = wid & 3;
+  const int k0out = mt * BM;
+  const int c0out = nt * BNW;
+
+  __shared__ __attribute__((aligned(16))) short lds[6 * (BM * FBK)];
+  auto ldsA = [&](int buf) { return lds + buf * (BM * FBK); };
+  auto ldsB = [&](int buf) { return lds + (3 + buf) * (BM * FBK); };
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  const int64_t go_img = (int64_t)img * g.K * g.OHW;
+  const int64_t x_img = (int64_t)img * g.C * g.OHW;
+  // glds lane map for [256][32]-short images: 1 KB instr = 16 rows;
+  // lane l -> row (l>>2), 8-short chunk (l&3)
+  const int g_row = lane >> 2;
+  const int g_ch = lane & 3;
+
+  auto stage = [&](int buf, int p) {
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const int inst = wid * 2 + i;  // 0..15: rows inst*16..+16
+      int row = k0out + inst * 16 + g_row;
+      if (row >= g.K) row = g.K - 1;
+      const bf16* srcA = go + go_img + (int64_t)row * g.OHW + p + g_ch * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)srcA,
+          (__attribute__((address_space(3))) unsigned int*)(
+              (__attribute__((address_space(3))) void*)(ldsA(buf) +
+                                                        inst * 512)),
+          16, 0, 0);
+      int rowB = c0out + inst * 16 + g_row;
+      if (rowB >= g.C) rowB = g.C - 1;
+      const bf16* srcB = x + x_img + (int64_t)rowB * g.OHW + p + g_ch * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)srcB,
+          (__attribute__((address_space(3))) unsigned int*)(
+              (__attribute__((address_space(3))) void*)(ldsB(buf) +
+                                                        inst * 512)),
+          16, 0, 0);
+    }
+  };
+
+  const int ksteps = g.slab / FBK;
+  stage(0, p0);
+  if (1 < ksteps) stage(1, p0 + FBK);
+  const int a_row0 = wm * 128 + (lane & 15);
+  const int b_row0 = wn * 64 + (lane & 15);
+  const int ch_rd = (lane >> 4);  // 8-short k-chunk within the 32-k step
+
+  for (int step = 0; step < ksteps; ++step) {
+    const int buf = step % 3;
+    if (step + 1 < ksteps)
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    if (step + 2 < ksteps) stage((step + 2) % 3, p0 + (step + 2) * FBK);
+    s16x8 afrag[8], bfrag[4];
+#pragma unroll
+    for (int f = 0; f < 8; ++f) {
+      const int ra = a_row0 + f * 16;
+      afrag[f] = *(const s16x8*)(ldsA(buf) + ra * 32 + ch_rd * 8);
+    }
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      const int rb = b_row0 + f * 16;
+      bfrag[f] = *(const s16x8*)(ldsB(buf) + rb * 32 + ch_rd * 8);
+    }
+#pragma unroll
+    for (int mf = 0; mf < 8; ++mf)
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf)
+        acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag[mf], bfrag[nf], acc[mf][nf], 0, 0, 0);
+  }
+  // the last step's readers finish before the epilogue (no barrier
+  // needed: each wave only reads LDS it waited for, and the epilogue
+  // touches no LDS)
+
+#pragma unroll
+  for (int mf = 0; mf < 8; ++mf) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int kout = k0out + wm * 128 + mf * 16 + ((lane >> 4) << 2) + reg;
+      if (kout >= g.K) continue;
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        const int c = c0out + wn * 64 + nf * 16 + (lane & 15);
+        if (c < g.C) atomicAdd(&gw[(int64_t)kout * g.C + c], acc[mf][nf][reg]);
+      }
+    }
+  }
+}
+
 torch::Tensor pw_bwdw(torch::Tensor go, torch::Tensor x) {
   TORCH_CHECK(go.is_cuda() && go.is_contiguous() && x.is_contiguous());
   TORCH_CHECK(go.scalar_type() == torch::kBFloat16 &&
@@ -985,9 +1111,22 @@ torch::Tensor pw_bwdw(torch::Tensor go, torch::Tensor x) {
   TORCH_CHECK(blocks > 0 && blocks < (1LL << 31), "pw_bwdw grid");
   auto stream = at::cuda::getCurrentCUDAStream();
   if (big) {
-    hipLaunchKernelGGL(pw_bwdw256_kernel, dim3((uint32_t)blocks), dim3(512),
-                       0, stream.stream(), (const bf16*)go.data_ptr(),
-                       (const bf16*)x.data_ptr(), gw.data_ptr<float>(), g);
+    // measured (gpurun_out/bwdwp.log): the BK=32 3-buffer span is
+    // correct but 0.96x of the BK=64 2-buffer (halving the MFMA work
+    // per barrier cancels the span win). Kept for round-3 reference.
+    static const bool pipe = []() {
+      const char* e = getenv("MPI4DL_BWDW_PIPE");
+      return e && e[0] == '3';
+    }();
+    if (pipe)
+      hipLaunchKernelGGL(pw_bwdw256p_kernel, dim3((uint32_t)blocks),
+                         dim3(512), 0, stream.stream(),
+                         (const bf16*)go.data_ptr(),
+                         (const bf16*)x.data_ptr(), gw.data_ptr<float>(), g);
+    else
+      hipLaunchKernelGGL(pw_bwdw256_kernel, dim3((uint32_t)blocks), dim3(512),
+                         0, stream.stream(), (const bf16*)go.data_ptr(),
+                         (const bf16*)x.data_ptr(), gw.data_ptr<float>(), g);
     return gw;
   }
   static const int bufs = []() {

@@ -286,6 +286,7 @@ def test_native_conv_autograd_matches_torch():
         (292, 133, 64, 1, 1, 0),   # pw_fat with C and K tails
         (520, 264, 32, 1, 1, 0),   # pw_fat256 with C and K tails
         (512, 256, 64, 1, 1, 0),   # pw_fat256 exact tiles
+        (288, 272, 128, 1, 1, 0),  # pipelined bwdw256 (OHW>=16384)
         (16, 32, 64, 3, 2, 1),     # zero-stuff native s2 bwd-data
         (16, 32, 62, 3, 2, 1),     # odd output-padding case
     ],
